@@ -1,0 +1,32 @@
+"""Fused MI355X training/inference engines.
+
+``maybe_fused_trainer(module, device, cfg)`` returns a fused engine when the
+architecture is supported on GPU (FeedForwardNet → fused hourglass kernels),
+else None and the caller uses the eager path.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Optional
+
+import torch
+
+from . import is_available
+
+logger = logging.getLogger(__name__)
+
+
+def maybe_fused_trainer(module, device: torch.device, cfg) -> Optional[object]:
+    if device.type != "cuda" or not is_available():
+        return None
+    from ..model.torch_nets import FeedForwardNet
+
+    if isinstance(module, FeedForwardNet):
+        from .hourglass_engine import FusedHourglassTrainer
+
+        try:
+            return FusedHourglassTrainer(module, device, cfg)
+        except Exception as e:  # unsupported shape etc.
+            logger.warning("Fused hourglass engine rejected module: %s", e)
+            return None
+    return None
