@@ -1,0 +1,17 @@
+// Torch extension module registration (host-only TU; kernels + launchers live
+// in the .hip TUs so they get HIP device compilation).
+#include <torch/extension.h>
+
+#include "launchers.h"
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("hg_train_step", &hg_train_step_launch, "fused hourglass train step");
+    m.def("hg_adam", &hg_adam_launch, "slab-reduce + Adam + bf16 weight emit");
+    m.def("hg_bump", &hg_bump_launch, "advance step counters");
+    m.def("hg_forward", &hg_forward_launch, "fused serving forward (+anomaly)");
+    m.def("hg_grad_reduce", &hg_grad_reduce_launch, "slab reduce -> flat grad bucket");
+    m.def("hg_adam_flat", &hg_adam_flat_launch, "Adam from reduced flat grad");
+    m.attr("WMAX") = 64;
+    m.attr("MAXL") = 5;
+    m.attr("MT") = 64;
+}

@@ -1,0 +1,94 @@
+// Common device helpers for gordo_amd CDNA4 (gfx950) kernels.
+// Hand-written for MI355X: wave64, MFMA bf16 16x16x32, XOR-swizzled LDS tiles.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define GA_WAVE 64
+#define GA_WMAX 64   // padded layer width (max supported by fused path)
+#define GA_MT 64     // batch rows per block tile
+#define GA_MAXL 5    // max layers in fused path
+
+typedef __bf16 ga_bf16x8 __attribute__((ext_vector_type(8)));
+typedef float ga_f32x4 __attribute__((ext_vector_type(4)));
+typedef unsigned short ga_u16x8 __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ float ga_bf16_to_f32(unsigned short u) {
+    union { unsigned int i; float f; } c;
+    c.i = ((unsigned int)u) << 16;
+    return c.f;
+}
+
+__device__ __forceinline__ unsigned short ga_f32_to_bf16(float f) {
+    union { float f; unsigned int i; } c;
+    c.f = f;
+    // round-to-nearest-even
+    unsigned int lsb = (c.i >> 16) & 1u;
+    unsigned int rounded = c.i + 0x7FFFu + lsb;
+    if ((c.i & 0x7F800000u) == 0x7F800000u) rounded = c.i;  // inf/nan passthrough
+    return (unsigned short)(rounded >> 16);
+}
+
+// ---------------------------------------------------------------------------
+// LDS tile addressing: [64][64] bf16 tiles, row stride 128 B, XOR-swizzled so
+// that per-lane row-varying ds_read_b128 (MFMA fragment loads) spread over 8
+// bank slots instead of hitting one (guide T2: byte ^= (row&7)<<4).
+// Tile size: 64*64*2 = 8 KiB.
+// ---------------------------------------------------------------------------
+#define GA_TILE_BYTES (GA_WMAX * GA_WMAX * 2)
+
+__device__ __forceinline__ unsigned int ga_sw(int row, int col) {
+    // byte offset of element [row][col] in a swizzled [64][64] bf16 tile
+    unsigned int byte = (unsigned int)(col * 2) ^ (((unsigned int)row & 7u) << 4);
+    return ((unsigned int)row << 7) + byte;
+}
+
+// Load an 8-wide contiguous bf16 group [row][col0..col0+7] (col0 % 8 == 0)
+__device__ __forceinline__ ga_bf16x8 ga_tile_read8(const char* tile, int row, int col0) {
+    return *(const ga_bf16x8*)(tile + ga_sw(row, col0));
+}
+
+__device__ __forceinline__ void ga_tile_write8(char* tile, int row, int col0, ga_u16x8 v) {
+    *(ga_u16x8*)(tile + ga_sw(row, col0)) = v;
+}
+
+__device__ __forceinline__ unsigned short ga_tile_read1(const char* tile, int row, int col) {
+    return *(const unsigned short*)(tile + ga_sw(row, col));
+}
+
+__device__ __forceinline__ void ga_tile_write1(char* tile, int row, int col, unsigned short v) {
+    *(unsigned short*)(tile + ga_sw(row, col)) = v;
+}
+
+// ---------------------------------------------------------------------------
+// Activations (ids shared with the Python engine)
+// ---------------------------------------------------------------------------
+#define GA_ACT_LINEAR 0
+#define GA_ACT_TANH 1
+#define GA_ACT_RELU 2
+#define GA_ACT_SIGMOID 3
+
+__device__ __forceinline__ float ga_act(float x, int id) {
+    switch (id) {
+        case GA_ACT_TANH: return tanhf(x);
+        case GA_ACT_RELU: return x > 0.f ? x : 0.f;
+        case GA_ACT_SIGMOID: return 1.f / (1.f + __expf(-x));
+        default: return x;
+    }
+}
+
+// derivative expressed in terms of the activation OUTPUT h
+__device__ __forceinline__ float ga_act_grad_from_h(float h, int id) {
+    switch (id) {
+        case GA_ACT_TANH: return 1.f - h * h;
+        case GA_ACT_RELU: return h > 0.f ? 1.f : 0.f;
+        case GA_ACT_SIGMOID: return h * (1.f - h);
+        default: return 1.f;
+    }
+}
+
+// wave-wide f32 sum (64 lanes)
+__device__ __forceinline__ float ga_wave_sum(float v) {
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    return v;
+}

@@ -1,0 +1,33 @@
+// Host-callable launch functions implemented in the .hip TUs.
+#pragma once
+#include <torch/extension.h>
+
+#include <vector>
+
+void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
+                          torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
+                          torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
+                          torch::Tensor loss, int64_t B, int64_t F_out,
+                          std::vector<int64_t> acts, double gscale, bool y_is_x);
+
+void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                    torch::Tensor dw, torch::Tensor db, torch::Tensor w,
+                    torch::Tensor wt, torch::Tensor bias, torch::Tensor counters,
+                    int64_t n_slabs, int64_t L, double lr, double beta1, double beta2,
+                    double eps);
+
+void hg_bump_launch(torch::Tensor counters, int64_t n_batches);
+
+void hg_forward_launch(torch::Tensor X, torch::Tensor w, torch::Tensor bias,
+                       torch::Tensor xscale, torch::Tensor xshift, torch::Tensor yhat,
+                       c10::optional<torch::Tensor> err, c10::optional<torch::Tensor> total,
+                       c10::optional<torch::Tensor> inv_err_scale,
+                       int64_t F_in, int64_t F_out, std::vector<int64_t> acts);
+
+void hg_grad_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,
+                           int64_t n_slabs, int64_t L);
+
+void hg_adam_flat_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                         torch::Tensor g, torch::Tensor w, torch::Tensor wt,
+                         torch::Tensor bias, torch::Tensor counters, int64_t L,
+                         double lr, double beta1, double beta2, double eps);

@@ -1,0 +1,254 @@
+"""Fused hourglass training/inference engine (Python driver for the CDNA4
+kernels in hip/hourglass_kernels.hip).
+
+Training step = 3 device-side kernels (train_step → adam → bump) with zero
+host round-trips, captured into one hipGraph (torch.cuda.CUDAGraph) and
+replayed per step; loss is read back once per epoch.
+"""
+from __future__ import annotations
+
+import logging
+import math
+import time
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from . import ext
+from ..runtime.trainer import TrainHistory
+
+logger = logging.getLogger(__name__)
+
+WMAX = 64
+MAXL = 5
+ACT_IDS = {"linear": 0, None: 0, "tanh": 1, "relu": 2, "sigmoid": 3}
+
+
+def supported(module) -> Optional[str]:
+    """None if the fused path supports this FeedForwardNet, else the reason."""
+    widths = module.layer_widths
+    if max(widths) > WMAX:
+        return f"width {max(widths)} > {WMAX}"
+    L = len(widths) - 1
+    if L < 1 or L > MAXL:
+        return f"L={L} outside 1..{MAXL}"
+    for f in module.layer_funcs:
+        if f not in ACT_IDS:
+            return f"activation {f!r} not fused"
+    return None
+
+
+class _PackedParams:
+    """Padded device-side parameter state shared by train + inference."""
+
+    def __init__(self, module, device):
+        self.widths = module.layer_widths
+        self.L = len(self.widths) - 1
+        self.act_ids: List[int] = [ACT_IDS[f] for f in module.layer_funcs]
+        self.device = device
+        PW = self.L * WMAX * WMAX
+        self.PW = PW
+        self.P = PW + self.L * WMAX
+
+        master = torch.zeros(self.P, dtype=torch.float32, device=device)
+        linears = [m for m in module.net if isinstance(m, torch.nn.Linear)]
+        assert len(linears) == self.L
+        self.linears = linears
+        for i, lin in enumerate(linears):
+            out_w, in_w = lin.weight.shape
+            wpad = master[i * WMAX * WMAX : (i + 1) * WMAX * WMAX].view(WMAX, WMAX)
+            wpad[:out_w, :in_w] = lin.weight.detach().to(device, torch.float32)
+            bpad = master[PW + i * WMAX : PW + (i + 1) * WMAX]
+            bpad[:out_w] = lin.bias.detach().to(device, torch.float32)
+        self.master = master
+        self.bias = master[PW:].clone()  # fp32 bias view used by kernels
+        w = master[:PW].view(self.L, WMAX, WMAX)
+        self.w_bf16 = w.to(torch.bfloat16).contiguous()
+        self.wt_bf16 = w.transpose(1, 2).contiguous().to(torch.bfloat16).contiguous()
+
+    def sync_to_module(self):
+        """Copy trained padded weights back into the torch module (fp32)."""
+        w = self.master[: self.PW].view(self.L, WMAX, WMAX)
+        b = self.master[self.PW :].view(self.L, WMAX)
+        with torch.no_grad():
+            for i, lin in enumerate(self.linears):
+                out_w, in_w = lin.weight.shape
+                lin.weight.copy_(w[i, :out_w, :in_w].to(lin.weight.device))
+                lin.bias.copy_(b[i, :out_w].to(lin.bias.device))
+
+
+def pad_rows_bf16(X: np.ndarray, device) -> torch.Tensor:
+    """[N,F] fp32 → [N,64] bf16 device tensor, zero-padded."""
+    N, F = X.shape
+    t = torch.zeros(N, WMAX, dtype=torch.bfloat16, device=device)
+    t[:, :F] = torch.as_tensor(X, device=device, dtype=torch.float32).to(torch.bfloat16)
+    return t
+
+
+class FusedHourglassTrainer:
+    def __init__(self, module, device: torch.device, cfg):
+        reason = supported(module)
+        if reason:
+            raise ValueError(f"fused path unsupported: {reason}")
+        self.module = module
+        self.device = device
+        self.cfg = cfg
+        self.E = ext()
+        if self.E is None:
+            raise RuntimeError("HIP extension missing")
+
+    def fit(self, X: np.ndarray, y: np.ndarray, lookback: int = 0, lookahead: int = 0) -> TrainHistory:
+        if lookback:
+            raise ValueError("fused hourglass engine is for 2-D models")
+        cfg = self.cfg
+        device = self.device
+        if cfg.seed is not None:
+            torch.manual_seed(cfg.seed)
+        p = _PackedParams(self.module, device)
+        L, F_in, F_out = p.L, p.widths[0], p.widths[-1]
+
+        y_is_x = y is X or (X.shape == y.shape and np.shares_memory(X, y))
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), device)
+        Yd = Xd if y_is_x else pad_rows_bf16(np.asarray(y, np.float32), device)
+
+        N = Xd.shape[0]
+        B = max(64, min(cfg.batch_size, N))
+        n_tiles = (B + 63) // 64
+        n_batches = max(1, math.ceil(N / B))
+
+        dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=device)
+        db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=device)
+        loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
+        counters = torch.tensor([0, 1], dtype=torch.int32, device=device)
+        m = torch.zeros_like(p.master)
+        v = torch.zeros_like(p.master)
+        perm = torch.zeros(n_batches * B, dtype=torch.int32, device=device)
+
+        gscale = 2.0 / (B * F_out)
+        acts = [int(i) for i in p.act_ids]
+
+        def fill_perm():
+            idx = torch.randperm(N, device=device)
+            need = n_batches * B
+            if need > N:
+                extra = torch.randint(0, N, (need - N,), device=device)
+                idx = torch.cat([idx, extra])
+            perm.copy_(idx[:need].to(torch.int32))
+
+        def one_step():
+            self.E.hg_train_step(
+                Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                loss_acc, B, F_out, acts, gscale, bool(y_is_x),
+            )
+            self.E.hg_adam(
+                p.master, m, v, dw, db, p.w_bf16, p.wt_bf16, p.bias, counters,
+                n_tiles, L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+            )
+            self.E.hg_bump(counters, n_batches)
+
+        # ---- optional hipGraph capture of the whole step -----------------
+        graph = None
+        if getattr(cfg, "use_graph", True) and n_batches >= 1:
+            try:
+                fill_perm()
+                # warm-up on a side stream (allocations + kernel modules)
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    one_step()
+                torch.cuda.current_stream().wait_stream(s)
+                torch.cuda.synchronize()
+                # reset state disturbed by warm-up
+                counters.copy_(torch.tensor([0, 1], dtype=torch.int32, device=device))
+                loss_acc.zero_()
+                m.zero_(); v.zero_()
+                p2 = _PackedParams(self.module, device)
+                p.master.copy_(p2.master); p.bias.copy_(p2.bias)
+                p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    one_step()
+                # capture replays with pristine state below
+                counters.copy_(torch.tensor([0, 1], dtype=torch.int32, device=device))
+                loss_acc.zero_(); m.zero_(); v.zero_()
+                p.master.copy_(p2.master); p.bias.copy_(p2.bias)
+                p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)
+            except Exception as e:
+                logger.warning("hipGraph capture failed (%s); plain launches", e)
+                graph = None
+
+        history = TrainHistory(engine="fused_hip" + ("_graph" if graph else ""))
+        denom = float(n_batches) * B * F_out
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(cfg.epochs):
+            fill_perm()
+            if graph is not None:
+                for _ in range(n_batches):
+                    graph.replay()
+            else:
+                for _ in range(n_batches):
+                    one_step()
+            history.loss.append(float(loss_acc.item()) / denom)
+            loss_acc.zero_()
+        torch.cuda.synchronize()
+        history.duration_sec = time.perf_counter() - t0
+        history.epochs = cfg.epochs
+        history.samples_per_sec = (
+            cfg.epochs * n_batches * B / history.duration_sec if history.duration_sec else 0.0
+        )
+        p.sync_to_module()
+        return history
+
+
+class FusedHourglassForward:
+    """Serving/predict forward: optional fused input scaling and anomaly."""
+
+    def __init__(self, module, device: torch.device):
+        reason = supported(module)
+        if reason:
+            raise ValueError(f"fused path unsupported: {reason}")
+        self.E = ext()
+        if self.E is None:
+            raise RuntimeError("HIP extension missing")
+        self.device = device
+        self.p = _PackedParams(module, device)
+        self.F_in = self.p.widths[0]
+        self.F_out = self.p.widths[-1]
+        self.identity_scale = torch.zeros(WMAX, dtype=torch.float32, device=device)
+        self.identity_scale[: self.F_in] = 1.0
+        self.zero_shift = torch.zeros(WMAX, dtype=torch.float32, device=device)
+
+    def set_input_scaler(self, scale: np.ndarray, shift: np.ndarray):
+        """x_scaled = x*scale + shift (fused MinMaxScaler)."""
+        s = torch.zeros(WMAX, dtype=torch.float32, device=self.device)
+        t = torch.zeros(WMAX, dtype=torch.float32, device=self.device)
+        s[: len(scale)] = torch.as_tensor(scale, dtype=torch.float32, device=self.device)
+        t[: len(shift)] = torch.as_tensor(shift, dtype=torch.float32, device=self.device)
+        self.identity_scale = s
+        self.zero_shift = t
+
+    def forward(self, X: torch.Tensor) -> torch.Tensor:
+        X = X.to(self.device, torch.float32).contiguous()
+        yhat = torch.empty(X.shape[0], self.F_out, dtype=torch.float32, device=self.device)
+        self.E.hg_forward(
+            X, self.p.w_bf16, self.p.bias, self.identity_scale, self.zero_shift,
+            yhat, None, None, None, self.F_in, self.F_out, self.p.act_ids,
+        )
+        return yhat
+
+    def anomaly(self, X: torch.Tensor, inv_err_scale: np.ndarray):
+        """Fused forward + per-feature |err|*inv_scale + row L2 (y = scaled x)."""
+        X = X.to(self.device, torch.float32).contiguous()
+        B = X.shape[0]
+        yhat = torch.empty(B, self.F_out, dtype=torch.float32, device=self.device)
+        err = torch.empty(B, self.F_out, dtype=torch.float32, device=self.device)
+        total = torch.empty(B, dtype=torch.float32, device=self.device)
+        inv = torch.zeros(WMAX, dtype=torch.float32, device=self.device)
+        inv[: self.F_out] = torch.as_tensor(inv_err_scale, dtype=torch.float32, device=self.device)
+        self.E.hg_forward(
+            X, self.p.w_bf16, self.p.bias, self.identity_scale, self.zero_shift,
+            yhat, err, total, inv, self.F_in, self.F_out, self.p.act_ids,
+        )
+        return yhat, err, total

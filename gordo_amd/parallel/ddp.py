@@ -1,0 +1,235 @@
+"""Data-parallel training over RCCL/xGMI.
+
+The reference has no distributed compute (fleet parallelism only, SURVEY.md
+§2.4); this is the MI355X-native addition: one process per GPU, gradients
+exchanged as ONE flat fp32 bucket per step (the hourglass model is KB-scale,
+so the all-reduce is latency-bound — a single fused bucket is the right shape
+for xGMI, not many small ones).
+
+Step pipeline per rank (fused GPU path):
+    hg_train_step (gather+fwd+loss+bwd, per-block slabs)
+    hg_grad_reduce (slabs -> flat bucket)
+    all_reduce(bucket, AVG) over RCCL        <- the only comm
+    hg_adam_flat + hg_bump
+
+CPU path (gloo, used by the world_size=2 CPU tests here): eager torch autograd
+with the same single-bucket all-reduce semantics.
+"""
+from __future__ import annotations
+
+import logging
+import math
+import os
+import time
+from typing import Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+logger = logging.getLogger(__name__)
+
+
+class DPContext:
+    """Thin wrapper around torch.distributed process-group state."""
+
+    def __init__(self, backend: Optional[str] = None, device: Optional[torch.device] = None):
+        self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+        if device is None:
+            if torch.cuda.is_available():
+                device = torch.device(f"cuda:{self.local_rank}")
+            else:
+                device = torch.device("cpu")
+        self.device = device
+        self.enabled = self.world_size > 1
+        if self.enabled and not dist.is_initialized():
+            if backend is None:
+                backend = "nccl" if device.type == "cuda" else "gloo"
+            if device.type == "cuda":
+                torch.cuda.set_device(device)
+            dist.init_process_group(backend=backend)
+        self.backend = backend
+
+    def all_reduce_mean(self, t: torch.Tensor):
+        if self.enabled:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            t.div_(self.world_size)
+
+    def barrier(self):
+        if self.enabled:
+            if self.device.type == "cuda":
+                dist.barrier(device_ids=[self.device.index])
+            else:
+                dist.barrier()
+
+    def broadcast(self, t: torch.Tensor, src: int = 0):
+        if self.enabled:
+            dist.broadcast(t, src=src)
+
+
+class HourglassStepTrainer:
+    """Step-oriented trainer for the hourglass AE (bench + fleet building).
+
+    Exposes ``step()`` so callers own the loop (warmup/timed regions); weights
+    live on device; ``sync_to_module()`` copies them back.
+    """
+
+    def __init__(self, module, X: np.ndarray, batch_size: int, cfg, dp: Optional[DPContext] = None):
+        from ..runtime.trainer import TrainConfig  # noqa: F401 (type only)
+
+        self.dp = dp or DPContext()
+        self.device = self.dp.device
+        self.cfg = cfg
+        self.module = module
+        self.B = batch_size
+        self.X = X
+
+        self.fused = False
+        if self.device.type == "cuda":
+            from ..ops import is_available
+            from ..ops.hourglass_engine import supported
+
+            if is_available() and supported(module) is None:
+                self.fused = True
+        if self.fused:
+            self._init_fused()
+        else:
+            self._init_eager()
+
+    # ------------------------------------------------------------------
+    def _init_fused(self):
+        from ..ops import ext
+        from ..ops.hourglass_engine import WMAX, _PackedParams, pad_rows_bf16
+
+        self.E = ext()
+        device = self.device
+        p = _PackedParams(self.module, device)
+        # ranks must start from identical weights
+        self.dp.broadcast(p.master)
+        p.bias.copy_(p.master[p.PW:])
+        w = p.master[: p.PW].view(p.L, WMAX, WMAX)
+        p.w_bf16.copy_(w.to(torch.bfloat16))
+        p.wt_bf16.copy_(w.transpose(1, 2).to(torch.bfloat16))
+        self.p = p
+        self.F_out = p.widths[-1]
+        self.Xd = pad_rows_bf16(np.asarray(self.X, np.float32), device)
+        N = self.Xd.shape[0]
+        self.N = N
+        self.n_tiles = (self.B + 63) // 64
+        self.dw = torch.zeros(self.n_tiles, p.L, WMAX * WMAX, dtype=torch.float32, device=device)
+        self.db = torch.zeros(self.n_tiles, p.L, WMAX, dtype=torch.float32, device=device)
+        self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
+        self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
+        self.counters = torch.tensor([0, 1], dtype=torch.int32, device=device)
+        self.m = torch.zeros_like(p.master)
+        self.v = torch.zeros_like(p.master)
+        self.n_batches = max(1, math.ceil(N / self.B))
+        self.perm = torch.zeros(self.n_batches * self.B, dtype=torch.int32, device=device)
+        self._fill_perm()
+        self.gscale = 2.0 / (self.B * self.F_out)
+        self._graph = None
+
+    def _fill_perm(self):
+        N, need = self.N, self.perm.numel()
+        idx = torch.randperm(N, device=self.device)
+        if need > N:
+            idx = torch.cat([idx, torch.randint(0, N, (need - N,), device=self.device)])
+        self.perm.copy_(idx[:need].to(torch.int32))
+
+    def _fused_step(self):
+        p = self.p
+        cfg = self.cfg
+        self.E.hg_train_step(
+            self.Xd, self.Xd, self.perm, self.counters, p.w_bf16, p.wt_bf16, p.bias,
+            self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
+            self.gscale, True,
+        )
+        if self.dp.enabled:
+            self.E.hg_grad_reduce(self.dw, self.db, self.gflat, self.n_tiles, p.L)
+            self.dp.all_reduce_mean(self.gflat)
+            self.E.hg_adam_flat(
+                p.master, self.m, self.v, self.gflat, p.w_bf16, p.wt_bf16, p.bias,
+                self.counters, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+            )
+        else:
+            self.E.hg_adam(
+                p.master, self.m, self.v, self.dw, self.db, p.w_bf16, p.wt_bf16,
+                p.bias, self.counters, self.n_tiles, p.L,
+                cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+            )
+        self.E.hg_bump(self.counters, self.n_batches)
+        self._steps_since_perm = getattr(self, "_steps_since_perm", 0) + 1
+        if self._steps_since_perm >= self.n_batches:
+            self._fill_perm()
+            self._steps_since_perm = 0
+
+    # ------------------------------------------------------------------
+    def _init_eager(self):
+        device = self.device
+        self.module = self.module.to(device)
+        self.dp_params = list(self.module.parameters())
+        if self.dp.enabled:
+            flat = torch.nn.utils.parameters_to_vector(self.dp_params)
+            self.dp.broadcast(flat)
+            torch.nn.utils.vector_to_parameters(flat, self.dp_params)
+        self.Xt = torch.as_tensor(np.asarray(self.X, np.float32), device=device)
+        self.opt = torch.optim.Adam(
+            self.dp_params, lr=self.cfg.lr, betas=self.cfg.betas, eps=self.cfg.eps
+        )
+        self.loss_fn = torch.nn.MSELoss()
+        self._gen = torch.Generator(device="cpu")
+        self._gen.manual_seed(1234 + self.dp.rank)
+
+    def _eager_step(self):
+        idx = torch.randint(0, len(self.Xt), (self.B,), generator=self._gen).to(self.device)
+        xb = self.Xt[idx]
+        self.opt.zero_grad(set_to_none=True)
+        use_bf16 = self.device.type == "cuda" and self.cfg.dtype == "bf16"
+        if use_bf16:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                out = self.module(xb)
+                loss = self.loss_fn(out.float(), xb)
+        else:
+            loss = self.loss_fn(self.module(xb), xb)
+        loss.backward()
+        if self.dp.enabled:
+            grads = torch.nn.utils.parameters_to_vector(
+                [p.grad for p in self.dp_params]
+            )
+            self.dp.all_reduce_mean(grads)
+            offset = 0
+            for par in self.dp_params:
+                n = par.grad.numel()
+                par.grad.copy_(grads[offset : offset + n].view_as(par.grad))
+                offset += n
+        self.opt.step()
+        self._last_loss = loss
+
+    # ------------------------------------------------------------------
+    def step(self):
+        if self.fused:
+            self._fused_step()
+        else:
+            self._eager_step()
+
+    def synchronize(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+    def epoch_loss(self) -> float:
+        if self.fused:
+            val = float(self.loss_acc.item())
+            self.loss_acc.zero_()
+            return val
+        return float(self._last_loss.item()) if hasattr(self, "_last_loss") else float("nan")
+
+    def sync_to_module(self):
+        if self.fused:
+            self.p.sync_to_module()
+
+    def params_vector(self) -> torch.Tensor:
+        if self.fused:
+            return self.p.master.detach().clone()
+        return torch.nn.utils.parameters_to_vector(self.dp_params).detach().clone()

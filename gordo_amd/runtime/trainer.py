@@ -48,6 +48,7 @@ class TrainConfig:
     loss: str = "mse"
     verbose: int = 0
     seed: Optional[int] = None
+    use_graph: bool = True  # hipGraph-capture the fused train step
 
     @classmethod
     def from_kwargs(cls, kwargs: Dict) -> "TrainConfig":
