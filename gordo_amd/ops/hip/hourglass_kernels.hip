@@ -283,7 +283,13 @@ __global__ void hg_adam(
 
     float g = 0.f;
     if (idx < PW) {
-        for (int s = 0; s < n_slabs; ++s) g += dw[(size_t)s * PW + idx];
+        // master layout is [l][out][in]; the train kernel's slabs are
+        // [l][in][out] (the dW MFMA's C layout) -> swap in/out on read
+        int l = idx / (GA_WMAX * GA_WMAX);
+        int rem = idx - l * GA_WMAX * GA_WMAX;
+        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
+        size_t sidx = (size_t)l * GA_WMAX * GA_WMAX + (size_t)in * GA_WMAX + out;
+        for (int s = 0; s < n_slabs; ++s) g += dw[(size_t)s * PW + sidx];
     } else {
         int bidx = idx - PW;
         for (int s = 0; s < n_slabs; ++s) g += db[(size_t)s * (L * GA_WMAX) + bidx];
@@ -437,7 +443,13 @@ __global__ void hg_grad_reduce(const float* __restrict__ dw,
     if (idx >= P) return;
     float s = 0.f;
     if (idx < PW) {
-        for (int sl = 0; sl < n_slabs; ++sl) s += dw[(size_t)sl * PW + idx];
+        // emit the flat bucket in MASTER layout [l][out][in] (slabs are
+        // [l][in][out]) so hg_adam_flat can consume g[idx] directly
+        int l = idx / (GA_WMAX * GA_WMAX);
+        int rem = idx - l * GA_WMAX * GA_WMAX;
+        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
+        size_t sidx = (size_t)l * GA_WMAX * GA_WMAX + (size_t)in * GA_WMAX + out;
+        for (int sl = 0; sl < n_slabs; ++sl) s += dw[(size_t)sl * PW + sidx];
     } else {
         int bidx = idx - PW;
         for (int sl = 0; sl < n_slabs; ++sl) s += db[(size_t)sl * (L * GA_WMAX) + bidx];

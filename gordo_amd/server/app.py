@@ -1,0 +1,157 @@
+"""Model server.
+
+Parity: gordo_components/server/server.py + views — one server per model dir:
+
+    POST /prediction          X (JSON or parquet) → model-input/model-output
+    POST /anomaly/prediction  adds tag-anomaly-scores / total-anomaly-score
+    GET  /metadata            build metadata
+    GET  /download-model      serialized model bytes
+    GET  /healthcheck         liveness
+
+MI355X-native: FastAPI/uvicorn (async) instead of Flask/gunicorn; the model is
+loaded once; inference goes through ModelRuntime's fused HIP path on GPU.
+Request/response formats match the reference's capability: JSON records,
+nested dict, or parquet (?format=parquet or parquet content-type).
+"""
+from __future__ import annotations
+
+import io
+import logging
+import os
+from typing import Optional
+
+import numpy as np
+import pandas as pd
+
+from .. import __version__
+from ..serializer import dumps
+from .model_io import ModelRuntime
+from .utils import (
+    dataframe_from_dict,
+    dataframe_from_parquet_bytes,
+    dataframe_into_parquet_bytes,
+    dataframe_to_dict,
+)
+from ..model.utils import make_base_dataframe
+
+logger = logging.getLogger(__name__)
+
+
+def build_app(model_dir: str):
+    from fastapi import FastAPI, Request, Response
+    from fastapi.responses import JSONResponse
+
+    runtime = ModelRuntime(model_dir)
+    app = FastAPI(title="gordo-amd model server", version=__version__)
+    app.state.runtime = runtime
+
+    # ------------------------------------------------------------------
+    async def _extract_frames(request: Request):
+        ctype = request.headers.get("content-type", "")
+        body = await request.body()
+        if "octet-stream" in ctype or "parquet" in ctype:
+            X = dataframe_from_parquet_bytes(body)
+            y = None
+        else:
+            import json
+
+            payload = json.loads(body) if body else {}
+            if "X" not in payload and "data" not in payload:
+                return None, None, JSONResponse(
+                    {"error": "Body must contain 'X'"}, status_code=400
+                )
+            Xp = payload.get("X", payload)
+            X = dataframe_from_dict(Xp if isinstance(Xp, dict) else {"data": Xp})
+            y = None
+            if payload.get("y") is not None:
+                yp = payload["y"]
+                y = dataframe_from_dict(yp if isinstance(yp, dict) else {"data": yp})
+        # column sanity: when metadata lists tags and X has plain columns,
+        # reorder to the training tag order
+        tags = runtime.tag_names
+        if tags and not isinstance(X.columns, pd.MultiIndex):
+            missing = [t for t in tags if t not in X.columns]
+            if not missing and list(X.columns) != tags:
+                X = X[tags]
+            elif missing and X.shape[1] == len(tags):
+                X.columns = tags
+            elif missing:
+                return None, None, JSONResponse(
+                    {"error": f"Missing expected tag columns: {missing[:5]}"},
+                    status_code=400,
+                )
+        return X, y, None
+
+    def _respond(df: pd.DataFrame, request: Request) -> Response:
+        fmt = request.query_params.get("format", "json")
+        accept = request.headers.get("accept", "")
+        if fmt == "parquet" or "octet-stream" in accept:
+            return Response(
+                content=dataframe_into_parquet_bytes(df),
+                media_type="application/octet-stream",
+            )
+        return JSONResponse(dataframe_to_dict(df))
+
+    # ------------------------------------------------------------------
+    @app.get("/healthcheck")
+    async def healthcheck():
+        return {"gordo-server-version": __version__}
+
+    @app.get("/metadata")
+    async def metadata():
+        return {"version": __version__, "metadata": runtime.metadata}
+
+    @app.get("/download-model")
+    async def download_model():
+        return Response(content=dumps(runtime.model), media_type="application/octet-stream")
+
+    @app.post("/prediction")
+    async def prediction(request: Request):
+        X, _, err = await _extract_frames(request)
+        if err is not None:
+            return err
+        try:
+            out = runtime.predict(X)
+        except Exception as e:
+            logger.exception("prediction failed")
+            return JSONResponse({"error": str(e)}, status_code=422)
+        df = make_base_dataframe(
+            tags=list(X.columns),
+            model_input=X.values,
+            model_output=out,
+            index=np.asarray(X.index),
+        )
+        return _respond(df, request)
+
+    @app.post("/anomaly/prediction")
+    async def anomaly_prediction(request: Request):
+        if not runtime.is_anomaly:
+            return JSONResponse(
+                {"error": "Model is not an anomaly detector; use /prediction"},
+                status_code=422,
+            )
+        X, y, err = await _extract_frames(request)
+        if err is not None:
+            return err
+        freq = None
+        ds_meta = runtime.metadata.get("dataset", {})
+        if ds_meta.get("resolution"):
+            try:
+                freq = pd.Timedelta(ds_meta["resolution"])
+            except ValueError:
+                freq = None
+        try:
+            df = runtime.anomaly(X, y, frequency=freq)
+        except Exception as e:
+            logger.exception("anomaly failed")
+            return JSONResponse({"error": str(e)}, status_code=422)
+        return _respond(df, request)
+
+    return app
+
+
+def run_server(model_dir: str, host: str = "0.0.0.0", port: int = 5555, workers: int = 1):
+    import uvicorn
+
+    app = build_app(model_dir)
+    uvicorn.run(app, host=host, port=port, workers=workers, log_level="info")

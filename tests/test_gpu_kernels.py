@@ -112,7 +112,7 @@ def test_train_step_gradients_match_autograd():
     torch.cuda.synchronize()
 
     # --- torch reference on the same bf16-rounded input ------------------
-    Xref = Xd[:, :50].float()
+    Xref = Xd[:, :50].float().cpu()
     out = ref_net(Xref)
     loss_ref = torch.nn.functional.mse_loss(out, Xref)
     loss_ref.backward()
@@ -136,6 +136,31 @@ def test_train_step_gradients_match_autograd():
         # pad region produced zero grads
         assert dw_sum[i, ninp:, :].abs().max().item() == 0.0
         assert dw_sum[i, :, o:].abs().max().item() == 0.0
+
+    # --- Adam step: fused update must match a manual fp32 Adam -----------
+    lr, b1, b2, eps = 1e-3, 0.9, 0.999, 1e-8
+    master_before = p.master.clone()
+    m_opt = torch.zeros_like(p.master)
+    v_opt = torch.zeros_like(p.master)
+    E.hg_adam(p.master, m_opt, v_opt, dw, db, p.w_bf16, p.wt_bf16, p.bias,
+              counters, n_tiles, L, lr, b1, b2, eps)
+    torch.cuda.synchronize()
+
+    # manual reference on the kernel's own summed grads (layout-adjusted)
+    g = torch.zeros_like(master_before)
+    PW = L * WMAX * WMAX
+    g[:PW] = dw.sum(dim=0).view(L, WMAX, WMAX).transpose(1, 2).reshape(-1)
+    g[PW:] = db.sum(dim=0).reshape(-1)
+    m_ref = (1 - b1) * g
+    v_ref = (1 - b2) * g * g
+    upd = master_before - lr * (m_ref / (1 - b1)) / ((v_ref / (1 - b2)).sqrt() + eps)
+    diff = (p.master - upd).abs().max().item()
+    assert diff < 1e-5, f"adam mismatch {diff}"
+    # bf16 working copies match the new master
+    w_now = p.master[:PW].view(L, WMAX, WMAX)
+    assert torch.allclose(p.w_bf16.float(), w_now.to(torch.bfloat16).float())
+    assert torch.allclose(p.wt_bf16.float(),
+                          w_now.transpose(1, 2).contiguous().to(torch.bfloat16).float())
 
 
 def test_fused_trainer_learns():
