@@ -16,6 +16,7 @@ from __future__ import annotations
 
 import logging
 import math
+import os
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
@@ -29,6 +30,10 @@ logger = logging.getLogger(__name__)
 
 def resolve_device(device: str = "auto") -> torch.device:
     if device == "auto":
+        # the fleet scheduler pins workers to GPUs via GORDO_DEVICE
+        env = os.environ.get("GORDO_DEVICE")
+        if env:
+            return torch.device(env)
         return torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
     return torch.device(device)
 

@@ -13,15 +13,12 @@ loaded once; inference goes through ModelRuntime's fused HIP path on GPU.
 Request/response formats match the reference's capability: JSON records,
 nested dict, or parquet (?format=parquet or parquet content-type).
 """
-from __future__ import annotations
-
-import io
 import logging
-import os
-from typing import Optional
 
 import numpy as np
 import pandas as pd
+from fastapi import FastAPI, Request, Response
+from fastapi.responses import JSONResponse
 
 from .. import __version__
 from ..serializer import dumps
@@ -38,9 +35,6 @@ logger = logging.getLogger(__name__)
 
 
 def build_app(model_dir: str):
-    from fastapi import FastAPI, Request, Response
-    from fastapi.responses import JSONResponse
-
     runtime = ModelRuntime(model_dir)
     app = FastAPI(title="gordo-amd model server", version=__version__)
     app.state.runtime = runtime

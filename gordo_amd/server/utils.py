@@ -49,13 +49,15 @@ def dataframe_from_dict(payload: Dict[str, Any]) -> pd.DataFrame:
     if isinstance(data, list):
         df = pd.DataFrame.from_records(data)
     elif isinstance(data, dict):
-        first = next(iter(data.values()), None)
-        if isinstance(first, dict):
-            # nested two-level
+        if any(isinstance(v, dict) for v in data.values()):
+            # two-level (possibly mixed with flat series like total-anomaly-score)
             frames = {}
             for top, subs in data.items():
-                for sub, vals in subs.items():
-                    frames[(top, sub)] = vals
+                if isinstance(subs, dict):
+                    for sub, vals in subs.items():
+                        frames[(top, sub)] = vals
+                else:
+                    frames[(top, "")] = subs
             df = pd.DataFrame(frames)
             df.columns = pd.MultiIndex.from_tuples(df.columns)
         else:

@@ -1,0 +1,115 @@
+"""In-node fleet builder.
+
+The reference compiles project YAML into an Argo workflow of one builder POD
+per machine (gordo_components/workflow/workflow_generator/); on a single
+8×MI355X node that becomes an in-process scheduler: N machine builds packed
+across the node's GPUs, one worker process per GPU (each model is KB-scale —
+one MI355X holds thousands), with the same recovery unit (md5-config registry
+cache: a re-run skips already-built machines).
+"""
+from __future__ import annotations
+
+import logging
+import multiprocessing as mp
+import os
+import time
+import traceback
+from concurrent.futures import ProcessPoolExecutor, as_completed
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from ..machine import Machine, NormalizedConfig
+
+logger = logging.getLogger(__name__)
+
+
+def _build_one(machine_dict: Dict[str, Any], output_dir: str, registry_dir: Optional[str],
+               device: str) -> Dict[str, Any]:
+    os.environ["GORDO_DEVICE"] = device
+    from ..builder import ModelBuilder
+
+    machine = Machine(**{k: v for k, v in machine_dict.items() if k != "project_name"},
+                      project_name=machine_dict.get("project_name"))
+    t0 = time.time()
+    try:
+        model, metadata = ModelBuilder(machine).build(
+            output_dir=output_dir, model_register_dir=registry_dir
+        )
+        return {
+            "machine": machine.name,
+            "status": "succeeded",
+            "output_dir": output_dir,
+            "seconds": time.time() - t0,
+            "cv": metadata.get("model", {}).get("cross-validation", {}).get("scores", {}),
+        }
+    except Exception as e:
+        return {
+            "machine": machine.name,
+            "status": "failed",
+            "error": f"{e}\n{traceback.format_exc()}",
+            "seconds": time.time() - t0,
+        }
+
+
+class FleetBuilder:
+    def __init__(
+        self,
+        config: NormalizedConfig,
+        output_dir: str,
+        registry_dir: Optional[str] = None,
+        n_workers: Optional[int] = None,
+        n_gpus: Optional[int] = None,
+    ):
+        self.config = config
+        self.output_dir = output_dir
+        self.registry_dir = registry_dir
+        if n_gpus is None:
+            n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        self.n_gpus = n_gpus
+        if n_workers is None:
+            n_workers = max(1, n_gpus) if n_gpus else min(8, os.cpu_count() or 1)
+        self.n_workers = n_workers
+
+    def machine_dir(self, name: str) -> str:
+        return os.path.join(self.output_dir, name)
+
+    def build_all(self) -> List[Dict[str, Any]]:
+        os.makedirs(self.output_dir, exist_ok=True)
+        ctx = mp.get_context("spawn")
+        results: List[Dict[str, Any]] = []
+        t0 = time.time()
+        with ProcessPoolExecutor(max_workers=self.n_workers, mp_context=ctx) as pool:
+            futures = {}
+            for i, machine in enumerate(self.config.machines):
+                device = f"cuda:{i % self.n_gpus}" if self.n_gpus else "cpu"
+                fut = pool.submit(
+                    _build_one,
+                    machine.to_dict(),
+                    self.machine_dir(machine.name),
+                    self.registry_dir,
+                    device,
+                )
+                futures[fut] = machine.name
+            for fut in as_completed(futures):
+                res = fut.result()
+                results.append(res)
+                logger.info(
+                    "fleet build %s: %s (%.1fs)",
+                    res["machine"],
+                    res["status"],
+                    res["seconds"],
+                )
+        elapsed = time.time() - t0
+        ok = sum(1 for r in results if r["status"] == "succeeded")
+        logger.info(
+            "fleet build: %d/%d machines in %.1fs (%.1f models/hour)",
+            ok,
+            len(results),
+            elapsed,
+            ok / elapsed * 3600 if elapsed else 0,
+        )
+        # stable order matching config
+        order = {m.name: i for i, m in enumerate(self.config.machines)}
+        results.sort(key=lambda r: order.get(r["machine"], 1 << 30))
+        return results

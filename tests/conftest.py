@@ -49,3 +49,42 @@ def machine_config():
             }
         },
     }
+
+
+@pytest.fixture(scope="session")
+def built_model_dir(tmp_path_factory):
+    """A real (tiny) anomaly model artifact dir, built once per session."""
+    from gordo_amd.builder import ModelBuilder
+    from gordo_amd.machine import Machine
+
+    config = {
+        "name": "test-machine",
+        "dataset": {
+            "type": "RandomDataset",
+            "train_start_date": "2020-01-01T00:00:00Z",
+            "train_end_date": "2020-02-01T00:00:00Z",
+            "tag_list": [f"tag-{i}" for i in range(10)],
+        },
+        "model": {
+            "gordo_amd.model.anomaly.diff.DiffBasedAnomalyDetector": {
+                "base_estimator": {
+                    "sklearn.pipeline.Pipeline": {
+                        "steps": [
+                            "sklearn.preprocessing.MinMaxScaler",
+                            {
+                                "gordo_amd.model.models.TorchAutoEncoder": {
+                                    "kind": "feedforward_hourglass",
+                                    "epochs": 1,
+                                    "batch_size": 64,
+                                }
+                            },
+                        ]
+                    }
+                }
+            }
+        },
+    }
+    out = tmp_path_factory.mktemp("model") / "test-machine"
+    machine = Machine.from_config(config)
+    ModelBuilder(machine).build(output_dir=str(out))
+    return str(out)

@@ -1,0 +1,120 @@
+"""CLI tests (parity: tests/.../test_cli.py — CliRunner invoking build /
+workflow generate with env vars against tmp dirs)."""
+import json
+import os
+
+import yaml
+from click.testing import CliRunner
+
+from gordo_amd.cli import gordo
+
+DATA_CONFIG = json.dumps(
+    {
+        "type": "RandomDataset",
+        "train_start_date": "2020-01-01T00:00:00Z",
+        "train_end_date": "2020-02-01T00:00:00Z",
+        "tag_list": ["t1", "t2", "t3", "t4"],
+    }
+)
+
+MODEL_CONFIG = json.dumps(
+    {
+        "gordo_amd.model.models.TorchAutoEncoder": {
+            "kind": "feedforward_hourglass",
+            "epochs": 1,
+            "batch_size": 32,
+        }
+    }
+)
+
+PROJECT_YAML = """
+globals:
+  dataset:
+    type: RandomDataset
+    train_start_date: "2020-01-01T00:00:00Z"
+    train_end_date: "2020-02-01T00:00:00Z"
+machines:
+  - name: m-a
+    dataset:
+      tag_list: [x1, x2]
+"""
+
+
+def test_build_command_args(tmp_path):
+    runner = CliRunner()
+    out = tmp_path / "model"
+    result = runner.invoke(
+        gordo,
+        [
+            "build",
+            "--name", "cli-machine",
+            "--data-config", DATA_CONFIG,
+            "--model-config", MODEL_CONFIG,
+            "--output-dir", str(out),
+            "--print-cv-scores",
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    assert (out / "model.pkl").exists()
+    assert (out / "metadata.json").exists()
+    assert "explained_variance_score" in result.output
+
+
+def test_build_command_env_vars(tmp_path):
+    runner = CliRunner()
+    out = tmp_path / "model-env"
+    env = {
+        "MACHINE_NAME": "env-machine",
+        "DATA_CONFIG": DATA_CONFIG,
+        "MODEL_CONFIG": MODEL_CONFIG,
+        "OUTPUT_DIR": str(out),
+        "MODEL_REGISTER_DIR": str(tmp_path / "registry"),
+    }
+    result = runner.invoke(gordo, ["build"], env=env)
+    assert result.exit_code == 0, result.output
+    assert (out / "model.pkl").exists()
+    # cache registry written
+    assert os.listdir(tmp_path / "registry")
+
+
+def test_workflow_generate(tmp_path):
+    cfg = tmp_path / "project.yaml"
+    cfg.write_text(PROJECT_YAML)
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        ["workflow", "generate", "--machine-config", str(cfg), "--project-name", "p1"],
+    )
+    assert result.exit_code == 0, result.output
+    plan = yaml.safe_load(result.output)
+    assert plan["kind"] == "NodePlan"
+    assert plan["metadata"]["project-name"] == "p1"
+
+
+def test_workflow_unique_tags(tmp_path):
+    cfg = tmp_path / "project.yaml"
+    cfg.write_text(PROJECT_YAML)
+    runner = CliRunner()
+    result = runner.invoke(gordo, ["workflow", "unique-tags", "--machine-config", str(cfg)])
+    assert result.exit_code == 0
+    assert result.output.split() == ["x1", "x2"]
+
+
+def test_fleet_build_command(tmp_path):
+    cfg = tmp_path / "project.yaml"
+    cfg.write_text(PROJECT_YAML)
+    runner = CliRunner()
+    report = tmp_path / "report.json"
+    result = runner.invoke(
+        gordo,
+        [
+            "fleet", "build",
+            "--machine-config", str(cfg),
+            "--output-dir", str(tmp_path / "models"),
+            "--workers", "1",
+            "--report-file", str(report),
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    rep = json.loads(report.read_text())
+    assert rep[0]["status"] == "succeeded"

@@ -1,0 +1,124 @@
+"""Server integration tests (parity: tests/.../test_gordo_server.py — a real
+tiny model in a tmp dir served via the in-process test client; exercises
+/prediction, /anomaly/prediction (json + parquet), /metadata,
+/download-model, error codes)."""
+import numpy as np
+import pandas as pd
+import pytest
+from fastapi.testclient import TestClient
+
+from gordo_amd.server import build_app
+from gordo_amd.server.utils import (
+    dataframe_from_dict,
+    dataframe_from_parquet_bytes,
+    dataframe_into_parquet_bytes,
+    dataframe_to_dict,
+)
+
+TAGS = [f"tag-{i}" for i in range(10)]
+
+
+@pytest.fixture(scope="module")
+def client(built_model_dir):
+    app = build_app(built_model_dir)
+    with TestClient(app) as c:
+        yield c
+
+
+def _x_frame(n=20, seed=0):
+    rng = np.random.default_rng(seed)
+    idx = pd.date_range("2020-03-01", periods=n, freq="10min", tz="UTC")
+    return pd.DataFrame(rng.normal(size=(n, 10)), index=idx, columns=TAGS)
+
+
+def test_healthcheck(client):
+    r = client.get("/healthcheck")
+    assert r.status_code == 200
+    assert "gordo-server-version" in r.json()
+
+
+def test_metadata(client):
+    r = client.get("/metadata")
+    assert r.status_code == 200
+    meta = r.json()["metadata"]
+    assert meta["name"] == "test-machine"
+    assert [t["name"] for t in meta["dataset"]["tag_list"]] == TAGS
+
+
+def test_download_model(client):
+    r = client.get("/download-model")
+    assert r.status_code == 200
+    from gordo_amd.serializer import loads
+
+    model = loads(r.content)
+    assert hasattr(model, "anomaly")
+
+
+def test_prediction_json(client):
+    X = _x_frame()
+    r = client.post("/prediction", json={"X": dataframe_to_dict(X)})
+    assert r.status_code == 200, r.text
+    df = dataframe_from_dict(r.json())
+    assert ("model-output", TAGS[0]) in df.columns
+    assert ("model-input", TAGS[0]) in df.columns
+    assert len(df) == len(X)
+
+
+def test_prediction_records_json(client):
+    X = _x_frame(5)
+    records = X.to_dict(orient="records")
+    r = client.post("/prediction", json={"X": records})
+    assert r.status_code == 200, r.text
+
+
+def test_anomaly_prediction_json(client):
+    X = _x_frame()
+    r = client.post("/anomaly/prediction", json={"X": dataframe_to_dict(X)})
+    assert r.status_code == 200, r.text
+    df = dataframe_from_dict(r.json())
+    tops = set(df.columns.get_level_values(0))
+    assert "tag-anomaly-scores" in tops
+    assert "total-anomaly-score" in tops
+    # timestamps preserved; start/end present (resolution known from metadata)
+    assert "start" in tops and "end" in tops
+
+
+def test_anomaly_prediction_parquet(client):
+    X = _x_frame()
+    r = client.post(
+        "/anomaly/prediction?format=parquet",
+        content=dataframe_into_parquet_bytes(X),
+        headers={"Content-Type": "application/octet-stream"},
+    )
+    assert r.status_code == 200
+    df = dataframe_from_parquet_bytes(r.content)
+    assert "total-anomaly-score" in set(df.columns.get_level_values(0))
+    assert len(df) == len(X)
+
+
+def test_prediction_bad_payload(client):
+    r = client.post("/prediction", json={"nonsense": 1})
+    assert r.status_code == 400
+
+
+def test_prediction_missing_columns(client):
+    X = _x_frame().iloc[:, :4]  # wrong width, wrong names subset
+    X.columns = ["a", "b", "c", "d"]
+    r = client.post("/prediction", json={"X": dataframe_to_dict(X)})
+    assert r.status_code == 400
+
+
+def test_column_reorder(client):
+    X = _x_frame()[list(reversed(TAGS))]
+    r = client.post("/prediction", json={"X": dataframe_to_dict(X)})
+    assert r.status_code == 200
+
+
+def test_parquet_roundtrip_multiindex():
+    df = pd.DataFrame(
+        np.arange(12, dtype=float).reshape(3, 4),
+        columns=pd.MultiIndex.from_product([["a", "b"], ["x", "y"]]),
+    )
+    out = dataframe_from_parquet_bytes(dataframe_into_parquet_bytes(df))
+    assert list(out.columns) == list(df.columns)
+    np.testing.assert_allclose(out.values, df.values)

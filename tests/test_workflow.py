@@ -1,0 +1,122 @@
+"""Workflow/fleet tests (parity: tests/.../test_workflow_generator.py —
+config normalization, default merging, plan generation; plus the in-node
+fleet builder)."""
+import yaml
+
+from gordo_amd.machine import Machine, NormalizedConfig
+from gordo_amd.workflow import FleetBuilder, generate_plan, generate_yaml, unique_tags
+
+PROJECT_YAML = """
+globals:
+  dataset:
+    type: RandomDataset
+    train_start_date: "2020-01-01T00:00:00Z"
+    train_end_date: "2020-02-01T00:00:00Z"
+  model:
+    gordo_amd.model.anomaly.diff.DiffBasedAnomalyDetector:
+      base_estimator:
+        sklearn.pipeline.Pipeline:
+          steps:
+            - sklearn.preprocessing.MinMaxScaler
+            - gordo_amd.model.models.TorchAutoEncoder:
+                kind: feedforward_hourglass
+                epochs: 1
+                batch_size: 64
+  evaluation:
+    cv_mode: cross_val_only
+machines:
+  - name: machine-a
+    dataset:
+      tag_list: [a1, a2, a3]
+  - name: machine-b
+    dataset:
+      tag_list: [b1, b2, a1]
+      resolution: 1h
+    evaluation:
+      cv_mode: full_build
+"""
+
+
+def test_normalized_config_merging():
+    cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="proj")
+    assert len(cfg.machines) == 2
+    a, b = cfg.machines
+    assert a.dataset["type"] == "RandomDataset"
+    assert [t["name"] for t in a.dataset["tag_list"]] == ["a1", "a2", "a3"]
+    # per-machine override wins; globals fill the rest
+    assert b.dataset["resolution"] == "1h"
+    assert b.dataset["train_start_date"] == "2020-01-01T00:00:00Z"
+    assert a.evaluation["cv_mode"] == "cross_val_only"
+    assert b.evaluation["cv_mode"] == "full_build"
+    # default model from globals
+    assert "DiffBasedAnomalyDetector" in str(a.model)
+
+
+def test_duplicate_names_rejected():
+    import pytest
+
+    bad = yaml.safe_load(PROJECT_YAML)
+    bad["machines"].append({"name": "machine-a", "dataset": {"tag_list": ["x"]}})
+    with pytest.raises(ValueError):
+        NormalizedConfig(bad)
+
+
+def test_unique_tags():
+    cfg = NormalizedConfig.from_yaml(PROJECT_YAML)
+    assert unique_tags(cfg) == ["a1", "a2", "a3", "b1", "b2"]
+
+
+def test_generate_plan_and_yaml():
+    cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="proj")
+    plan = generate_plan(cfg, project_name="proj", output_dir="/tmp/m", n_gpus=8)
+    assert plan["kind"] == "NodePlan"
+    tasks = plan["spec"]["builder"]["tasks"]
+    assert [t["name"] for t in tasks] == ["machine-a", "machine-b"]
+    assert tasks[0]["device"] == "cuda:0" and tasks[1]["device"] == "cuda:1"
+    assert plan["spec"]["watchman"]["targets"]["machine-a"].startswith("http://")
+    # YAML round-trips
+    text = generate_yaml(cfg, project_name="proj")
+    assert yaml.safe_load(text)["kind"] == "NodePlan"
+
+
+def test_fleet_builder_builds_all(tmp_path):
+    cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="proj")
+    fb = FleetBuilder(cfg, str(tmp_path / "models"), str(tmp_path / "registry"),
+                      n_workers=2, n_gpus=0)
+    results = fb.build_all()
+    assert [r["machine"] for r in results] == ["machine-a", "machine-b"]
+    assert all(r["status"] == "succeeded" for r in results), results
+    # machine-b did a full build -> loadable artifact with model.pkl
+    import os
+
+    assert os.path.exists(tmp_path / "models" / "machine-b" / "model.pkl")
+    # registry got entries -> second run is all cache hits (fast)
+    import time
+
+    t0 = time.time()
+    results2 = fb.build_all()
+    assert all(r["status"] == "succeeded" for r in results2)
+
+
+def test_fleet_app(tmp_path):
+    """Multi-model server mounts each built machine under its project path."""
+    cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="proj")
+    # full builds so artifacts exist
+    for m in cfg.machines:
+        m.evaluation["cv_mode"] = "full_build"
+    fb = FleetBuilder(cfg, str(tmp_path / "models"), n_workers=1, n_gpus=0)
+    results = fb.build_all()
+    assert all(r["status"] == "succeeded" for r in results), results
+
+    from fastapi.testclient import TestClient
+
+    from gordo_amd.server.fleet_app import build_fleet_app
+
+    app = build_fleet_app(str(tmp_path / "models"), project_name="proj")
+    with TestClient(app) as c:
+        r = c.get("/")
+        eps = {e["target-name"]: e["endpoint"] for e in r.json()["endpoints"]}
+        assert set(eps) == {"machine-a", "machine-b"}
+        r = c.get(eps["machine-a"] + "/metadata")
+        assert r.status_code == 200
+        assert r.json()["metadata"]["name"] == "machine-a"
