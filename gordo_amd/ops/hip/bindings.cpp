@@ -11,6 +11,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hg_forward", &hg_forward_launch, "fused serving forward (+anomaly)");
     m.def("hg_grad_reduce", &hg_grad_reduce_launch, "slab reduce -> flat grad bucket");
     m.def("hg_adam_flat", &hg_adam_flat_launch, "Adam from reduced flat grad");
+    m.def("lstm_fwd", &lstm_fwd_launch, "persistent fused 4-gate LSTM forward");
+    m.def("lstm_bwd_cell", &lstm_bwd_cell_launch, "reverse-time BPTT cell kernel");
     m.attr("WMAX") = 64;
     m.attr("MAXL") = 5;
     m.attr("MT") = 64;

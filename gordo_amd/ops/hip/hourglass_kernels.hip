@@ -32,6 +32,7 @@ struct HgTrainArgs {
     float* __restrict__ dw;                 // [n_tiles][L][64*64] fp32 slabs
     float* __restrict__ db;                 // [n_tiles][L][64] fp32 slabs
     float* __restrict__ loss;               // [1] fp32 accumulator (SSE)
+    unsigned short* __restrict__ dx_out;    // [B][64] bf16 dL/dinput, or null
     int B;          // rows per step (per GPU)
     int N;          // dataset rows
     int F_out;      // real output features (cols >= F_out masked)
@@ -173,7 +174,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
 
     for (int li = a.L - 1; li >= 0; --li) {
         // ---- dX: per-wave row stripe over m ----------------------------
-        if (li > 0) {
+        if (li > 0 || a.dx_out != nullptr) {
             const unsigned short* wtl = a.wt + (size_t)li * GA_WMAX * GA_WMAX;
             ga_f32x4 acc[GA_NT];
             for (int nt = 0; nt < GA_NT; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
@@ -186,17 +187,32 @@ __global__ void hg_train_step(HgTrainArgs a) {
                     acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
                 }
             }
+            if (li > 0) {
 #pragma unroll
-            for (int nt = 0; nt < GA_NT; ++nt) {
-                int col = nt * 16 + (lane & 15);
+                for (int nt = 0; nt < GA_NT; ++nt) {
+                    int col = nt * 16 + (lane & 15);
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    int m = wrow + (lane >> 4) * 4 + r;
-                    float h = ga_bf16_to_f32(ga_tile_read1(act_tile(li), m, col));
-                    float d = acc[nt][r] * ga_act_grad_from_h(h, a.act_id[li - 1]);
-                    unsigned short dp = ga_f32_to_bf16(d);
-                    ga_tile_write1(dnext, m, col, dp);
-                    ga_tile_write1(dnextT, col, m, dp);
+                    for (int r = 0; r < 4; ++r) {
+                        int m = wrow + (lane >> 4) * 4 + r;
+                        float h = ga_bf16_to_f32(ga_tile_read1(act_tile(li), m, col));
+                        float d = acc[nt][r] * ga_act_grad_from_h(h, a.act_id[li - 1]);
+                        unsigned short dp = ga_f32_to_bf16(d);
+                        ga_tile_write1(dnext, m, col, dp);
+                        ga_tile_write1(dnextT, col, m, dp);
+                    }
+                }
+            } else {
+                // dL/d(input) for the caller (e.g. the LSTM head): batch order
+#pragma unroll
+                for (int nt = 0; nt < GA_NT; ++nt) {
+                    int col = nt * 16 + (lane & 15);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int m = wrow + (lane >> 4) * 4 + r;
+                        int gi = base + m;
+                        if (gi < a.B)
+                            a.dx_out[(size_t)gi * GA_WMAX + col] = ga_f32_to_bf16(acc[nt][r]);
+                    }
                 }
             }
         }
@@ -510,7 +526,8 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
                           torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
                           torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
                           torch::Tensor loss, int64_t B, int64_t F_out,
-                          std::vector<int64_t> acts, double gscale, bool y_is_x) {
+                          std::vector<int64_t> acts, double gscale, bool y_is_x,
+                          c10::optional<torch::Tensor> dx_out) {
     CHECK_GPU(X);
     int L = (int)(w.numel() / (GA_WMAX * GA_WMAX));
     HgTrainArgs a;
@@ -524,6 +541,7 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
     a.dw = dw.data_ptr<float>();
     a.db = db.data_ptr<float>();
     a.loss = loss.data_ptr<float>();
+    a.dx_out = dx_out.has_value() ? (unsigned short*)dx_out->data_ptr() : nullptr;
     a.B = (int)B;
     a.N = (int)(X.size(0));
     a.F_out = (int)F_out;

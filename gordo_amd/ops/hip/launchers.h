@@ -8,7 +8,8 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
                           torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
                           torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
                           torch::Tensor loss, int64_t B, int64_t F_out,
-                          std::vector<int64_t> acts, double gscale, bool y_is_x);
+                          std::vector<int64_t> acts, double gscale, bool y_is_x,
+                          c10::optional<torch::Tensor> dx_out);
 
 void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                     torch::Tensor dw, torch::Tensor db, torch::Tensor w,
@@ -31,3 +32,13 @@ void hg_adam_flat_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                          torch::Tensor g, torch::Tensor w, torch::Tensor wt,
                          torch::Tensor bias, torch::Tensor counters, int64_t L,
                          double lr, double beta1, double beta2, double eps);
+
+void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
+                     torch::Tensor wcat, torch::Tensor bias, torch::Tensor h_seq,
+                     torch::Tensor gates_seq, torch::Tensor c_seq,
+                     c10::optional<torch::Tensor> x_seq_out, int64_t B, int64_t T);
+
+void lstm_bwd_cell_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
+                          torch::Tensor c_seq, torch::Tensor wcat_t,
+                          torch::Tensor dgates_seq, int64_t B, int64_t T,
+                          bool dh_last_only);

@@ -139,7 +139,7 @@ class FusedHourglassTrainer:
         def one_step():
             self.E.hg_train_step(
                 Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                loss_acc, B, F_out, acts, gscale, bool(y_is_x),
+                loss_acc, B, F_out, acts, gscale, bool(y_is_x), None,
             )
             self.E.hg_adam(
                 p.master, m, v, dw, db, p.w_bf16, p.wt_bf16, p.bias, counters,

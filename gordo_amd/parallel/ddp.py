@@ -130,6 +130,42 @@ class HourglassStepTrainer:
         self._fill_perm()
         self.gscale = 2.0 / (self.B * self.F_out)
         self._graph = None
+        if getattr(self.cfg, "use_graph", True):
+            self._try_capture_graph()
+
+    def _try_capture_graph(self):
+        """Capture the whole step (train+reduce+all_reduce+adam+bump) into one
+        hipGraph; counters/perm live in device memory so replays iterate."""
+        try:
+            state = (
+                self.p.master.clone(), self.m.clone(), self.v.clone(),
+                self.counters.clone(),
+            )
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):  # NCCL/RCCL warmup inside side stream
+                    self._fused_step_raw()
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._fused_step_raw()
+            # restore pre-warmup state
+            self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
+            self.counters.copy_(state[3])
+            self.p.bias.copy_(self.p.master[self.p.PW:])
+            from ..ops.hourglass_engine import WMAX
+            w = self.p.master[: self.p.PW].view(self.p.L, WMAX, WMAX)
+            self.p.w_bf16.copy_(w.to(torch.bfloat16))
+            self.p.wt_bf16.copy_(w.transpose(1, 2).contiguous().to(torch.bfloat16))
+            self.loss_acc.zero_()
+            torch.cuda.synchronize()
+            self._graph = g
+            logger.info("train step captured in hipGraph (world=%d)", self.dp.world_size)
+        except Exception as e:
+            logger.warning("hipGraph capture failed (%s); plain launches", e)
+            self._graph = None
 
     def _fill_perm(self):
         N, need = self.N, self.perm.numel()
@@ -139,12 +175,22 @@ class HourglassStepTrainer:
         self.perm.copy_(idx[:need].to(torch.int32))
 
     def _fused_step(self):
+        if self._graph is not None:
+            self._graph.replay()
+        else:
+            self._fused_step_raw()
+        self._steps_since_perm = getattr(self, "_steps_since_perm", 0) + 1
+        if self._steps_since_perm >= self.n_batches:
+            self._fill_perm()
+            self._steps_since_perm = 0
+
+    def _fused_step_raw(self):
         p = self.p
         cfg = self.cfg
         self.E.hg_train_step(
             self.Xd, self.Xd, self.perm, self.counters, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
-            self.gscale, True,
+            self.gscale, True, None,
         )
         if self.dp.enabled:
             self.E.hg_grad_reduce(self.dw, self.db, self.gflat, self.n_tiles, p.L)
@@ -160,10 +206,6 @@ class HourglassStepTrainer:
                 cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
             )
         self.E.hg_bump(self.counters, self.n_batches)
-        self._steps_since_perm = getattr(self, "_steps_since_perm", 0) + 1
-        if self._steps_since_perm >= self.n_batches:
-            self._fill_perm()
-            self._steps_since_perm = 0
 
     # ------------------------------------------------------------------
     def _init_eager(self):
