@@ -43,8 +43,14 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=50)
-    ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
+    ap.add_argument("--batch", type=int, default=0)
     ap.add_argument("--tags", type=int, default=N_TAGS)
+    ap.add_argument(
+        "--model", choices=["hourglass", "lstm"], default="hourglass",
+        help="hourglass = BASELINE headline config; lstm = config #3 "
+        "(LSTM-AE, lookback 144, fused 4-gate cell)",
+    )
+    ap.add_argument("--lookback", type=int, default=144)
     args = ap.parse_args()
 
     from gordo_amd.model.factories import feedforward_hourglass
@@ -58,12 +64,33 @@ def main():
     on_gpu = device.type == "cuda"
 
     torch.manual_seed(42)  # identical init on every rank (then broadcast)
-    module = feedforward_hourglass(
-        args.tags, args.tags, encoding_layers=2, compression_factor=0.5
-    )
-    X = synthetic_sensor_data(DATASET_ROWS, args.tags, seed=1000 + rank)
-    cfg = TrainConfig(batch_size=args.batch, lr=1e-3, dtype="bf16")
-    trainer = HourglassStepTrainer(module, X, args.batch, cfg, dp=dp)
+    if args.model == "lstm":
+        from gordo_amd.model.factories.lstm_autoencoder import lstm_hourglass
+        from gordo_amd.ops.lstm_engine import LSTMStepTrainer
+
+        batch = args.batch or 16384
+        module = lstm_hourglass(
+            args.tags, args.tags, lookback_window=args.lookback,
+            encoding_layers=2, compression_factor=0.5,
+        )
+        rows = max(DATASET_ROWS // 4, batch + args.lookback + 1)
+        X = synthetic_sensor_data(rows, args.tags, seed=1000 + rank)
+        cfg = TrainConfig(batch_size=batch, lr=1e-3, dtype="bf16")
+        trainer = LSTMStepTrainer(module, X, batch, cfg, dp=dp)
+        model_name = "lstm_hourglass-50x" + "x".join(str(d) for d in module.dims)
+        seq_len = args.lookback
+        engine = "fused_hip_lstm"
+    else:
+        batch = args.batch or PER_GPU_BATCH
+        module = feedforward_hourglass(
+            args.tags, args.tags, encoding_layers=2, compression_factor=0.5
+        )
+        X = synthetic_sensor_data(DATASET_ROWS, args.tags, seed=1000 + rank)
+        cfg = TrainConfig(batch_size=batch, lr=1e-3, dtype="bf16")
+        trainer = HourglassStepTrainer(module, X, batch, cfg, dp=dp)
+        model_name = "feedforward_hourglass-50x38x25x38x50"
+        seq_len = 1
+        engine = "fused_hip" if trainer.fused else "eager"
 
     # ---- warmup -----------------------------------------------------------
     for _ in range(args.warmup):
@@ -91,7 +118,7 @@ def main():
     elapsed = float(t.item())
 
     if rank == 0:
-        value = world * args.batch * args.steps / elapsed
+        value = world * batch * args.steps / elapsed
         out = {
             "metric": "samples_per_sec",
             "value": value,
@@ -106,11 +133,11 @@ def main():
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "feedforward_hourglass-50x38x25x38x50",
-                "global_batch": world * args.batch,
-                "seq_len": 1,
+                "model": model_name,
+                "global_batch": world * batch,
+                "seq_len": seq_len,
                 "parallelism": f"dp{world}",
-                "engine": "fused_hip" if trainer.fused else "eager",
+                "engine": engine,
             },
         }
         print(json.dumps(out))

@@ -186,6 +186,13 @@ class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
                 f"Need at least lookback_window={lookback} rows, got {len(X)}"
             )
         device = resolve_device(self.kwargs.get("device", "auto"))
+        if device.type == "cuda":
+            from ..ops import is_available
+            from ..ops.lstm_engine import FusedLSTMForward, lstm_supported
+
+            if is_available() and lstm_supported(self.model) is None:
+                fwd = FusedLSTMForward(self.model, device)
+                return fwd.predict(X, lookahead=self.lookahead)
         model = self.model.to(device)
         model.eval()
         Xt = torch.as_tensor(X, device=device)

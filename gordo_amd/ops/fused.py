@@ -19,7 +19,7 @@ logger = logging.getLogger(__name__)
 def maybe_fused_trainer(module, device: torch.device, cfg) -> Optional[object]:
     if device.type != "cuda" or not is_available():
         return None
-    from ..model.torch_nets import FeedForwardNet
+    from ..model.torch_nets import FeedForwardNet, LSTMNet
 
     if isinstance(module, FeedForwardNet):
         from .hourglass_engine import FusedHourglassTrainer
@@ -28,5 +28,13 @@ def maybe_fused_trainer(module, device: torch.device, cfg) -> Optional[object]:
             return FusedHourglassTrainer(module, device, cfg)
         except Exception as e:  # unsupported shape etc.
             logger.warning("Fused hourglass engine rejected module: %s", e)
+            return None
+    if isinstance(module, LSTMNet):
+        from .lstm_engine import FusedLSTMTrainer
+
+        try:
+            return FusedLSTMTrainer(module, device, cfg)
+        except Exception as e:
+            logger.warning("Fused LSTM engine rejected module: %s", e)
             return None
     return None

@@ -1,0 +1,415 @@
+"""Fused LSTM training engine (Python driver for the CDNA4 persistent-cell
+kernels in hip/lstm_kernels.hip — SURVEY.md §2.3 K4/K5/K6).
+
+The reference trains Keras stacked LSTMs over rolling windows
+(gordo_components/model/factories/lstm_autoencoder.py +
+model/utils.py::create_keras_timeseriesgenerator). Here each training step is:
+
+  per layer l:        lstm_fwd      — ONE kernel loops t=0..T-1 on-chip
+                                      (layer 0 gathers windows straight from
+                                      the resident dataset — no materialized
+                                      [B,T,F] batch in HBM traffic terms)
+  head fwd+bwd+loss:  hg_train_step — 1-layer dense head on h_{T-1}, MSE,
+                                      dx_out = dL/dh_{T-1}
+  per layer l (rev):  lstm_bwd_cell — reverse-time BPTT (dgates), then the
+                      non-sequential bulk GEMMs (dW = in^T x dgates over
+                      K = B*T, dx_seq = dgates x W_ih) via rocBLAS bf16
+  optimizer:          fp32-master Adam (torch foreach) + bf16 weight re-emit
+
+Weight layout: per layer one packed [4H=256][128] bf16 "wcat" (cols 0..63 =
+W_ih zero-padded, 64..127 = W_hh; rows gate-major i,f,g,o each padded to 64 —
+same gate order as torch.nn.LSTM) plus a fused fp32 bias b_ih+b_hh [256].
+Zero padding is gradient-invariant: padded h stays exactly 0 through the cell
+(weights/bias 0 -> g=tanh(0)=0 -> c=0 -> h=0) and padded grads are 0.
+"""
+from __future__ import annotations
+
+import logging
+import math
+import time
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from . import ext
+from .hourglass_engine import ACT_IDS, WMAX, pad_rows_bf16
+from ..runtime.trainer import TrainHistory
+
+logger = logging.getLogger(__name__)
+
+H4 = 4 * WMAX  # 256
+KCAT = 2 * WMAX  # 128
+
+
+def lstm_supported(module) -> Optional[str]:
+    """None if the fused path supports this LSTMNet, else the reason."""
+    from ..model.torch_nets import LSTMNet
+
+    if not isinstance(module, LSTMNet):
+        return f"{type(module).__name__} is not LSTMNet"
+    if module.n_features > WMAX:
+        return f"n_features {module.n_features} > {WMAX}"
+    if module.n_features_out > WMAX:
+        return f"n_features_out {module.n_features_out} > {WMAX}"
+    for cell in module.cells:
+        if cell.hidden_size > WMAX or cell.input_size > WMAX:
+            return f"LSTM dims {cell.input_size}->{cell.hidden_size} > {WMAX}"
+        if cell.num_layers != 1 or cell.bidirectional:
+            return "only stacked unidirectional single-layer cells"
+    if module.out_func not in ACT_IDS:
+        return f"out_func {module.out_func!r} not fused"
+    return None
+
+
+class _LSTMPackedParams:
+    """Padded fp32 masters + bf16 kernel weights for the stacked LSTM + head."""
+
+    def __init__(self, module, device):
+        self.module = module
+        self.device = device
+        self.Lc = len(module.cells)
+        self.wcat_master: List[torch.Tensor] = []
+        self.bias_master: List[torch.Tensor] = []
+        self.wcat_bf16: List[torch.Tensor] = []
+        self.wcat_t_bf16: List[torch.Tensor] = []
+        self.h_real: List[int] = []
+        for cell in module.cells:
+            H, I = cell.hidden_size, cell.input_size
+            self.h_real.append(H)
+            wcat = torch.zeros(H4, KCAT, dtype=torch.float32, device=device)
+            bias = torch.zeros(H4, dtype=torch.float32, device=device)
+            w_ih = cell.weight_ih_l0.detach().to(device, torch.float32)  # [4H, I]
+            w_hh = cell.weight_hh_l0.detach().to(device, torch.float32)  # [4H, H]
+            b = (cell.bias_ih_l0 + cell.bias_hh_l0).detach().to(device, torch.float32)
+            for g in range(4):
+                wcat[g * WMAX : g * WMAX + H, :I] = w_ih[g * H : (g + 1) * H]
+                wcat[g * WMAX : g * WMAX + H, WMAX : WMAX + H] = w_hh[g * H : (g + 1) * H]
+                bias[g * WMAX : g * WMAX + H] = b[g * H : (g + 1) * H]
+            self.wcat_master.append(wcat)
+            self.bias_master.append(bias)
+            self.wcat_bf16.append(wcat.to(torch.bfloat16).contiguous())
+            self.wcat_t_bf16.append(wcat.t().contiguous().to(torch.bfloat16))
+        # dense head: bf16 w [1][64][64] (out,in), wt (in,out), fp32 bias [64]
+        self.head_wm = torch.zeros(WMAX, WMAX, dtype=torch.float32, device=device)
+        self.head_bm = torch.zeros(WMAX, dtype=torch.float32, device=device)
+        out_w, in_w = module.head.weight.shape
+        self.head_wm[:out_w, :in_w] = module.head.weight.detach().to(device, torch.float32)
+        self.head_bm[:out_w] = module.head.bias.detach().to(device, torch.float32)
+        self.head_w = self.head_wm.to(torch.bfloat16).contiguous()
+        self.head_wt = self.head_wm.t().contiguous().to(torch.bfloat16)
+        self.head_bias = self.head_bm.clone()
+
+    def refresh_bf16(self):
+        for l in range(self.Lc):
+            self.wcat_bf16[l].copy_(self.wcat_master[l].to(torch.bfloat16))
+            self.wcat_t_bf16[l].copy_(self.wcat_master[l].t().to(torch.bfloat16))
+        self.head_w.copy_(self.head_wm.to(torch.bfloat16))
+        self.head_wt.copy_(self.head_wm.t().to(torch.bfloat16))
+        self.head_bias.copy_(self.head_bm)
+
+    def masters(self) -> List[torch.Tensor]:
+        return [*self.wcat_master, *self.bias_master, self.head_wm, self.head_bm]
+
+    def sync_to_module(self):
+        with torch.no_grad():
+            for l, cell in enumerate(self.module.cells):
+                H, I = cell.hidden_size, cell.input_size
+                wcat = self.wcat_master[l]
+                bias = self.bias_master[l]
+                for g in range(4):
+                    cell.weight_ih_l0[g * H : (g + 1) * H].copy_(
+                        wcat[g * WMAX : g * WMAX + H, :I].to(cell.weight_ih_l0.device)
+                    )
+                    cell.weight_hh_l0[g * H : (g + 1) * H].copy_(
+                        wcat[g * WMAX : g * WMAX + H, WMAX : WMAX + H].to(cell.weight_hh_l0.device)
+                    )
+                    # split the fused bias evenly (torch keeps two bias vectors)
+                    half = bias[g * WMAX : g * WMAX + H].to(cell.bias_ih_l0.device) * 0.5
+                    cell.bias_ih_l0[g * H : (g + 1) * H].copy_(half)
+                    cell.bias_hh_l0[g * H : (g + 1) * H].copy_(half)
+            out_w, in_w = self.module.head.weight.shape
+            self.module.head.weight.copy_(self.head_wm[:out_w, :in_w].to(self.module.head.weight.device))
+            self.module.head.bias.copy_(self.head_bm[:out_w].to(self.module.head.bias.device))
+
+
+class FusedLSTMTrainer:
+    """One-GPU fused trainer for LSTMNet (KerasLSTM{AutoEncoder,Forecast} parity)."""
+
+    def __init__(self, module, device: torch.device, cfg):
+        reason = lstm_supported(module)
+        if reason:
+            raise ValueError(f"fused LSTM path unsupported: {reason}")
+        self.module = module
+        self.device = device
+        self.cfg = cfg
+        self.E = ext()
+        if self.E is None:
+            raise RuntimeError("HIP extension missing")
+
+    # ------------------------------------------------------------------
+    def fit(self, X: np.ndarray, y: np.ndarray, lookback: int = 144, lookahead: int = 0) -> TrainHistory:
+        cfg = self.cfg
+        device = self.device
+        if cfg.seed is not None:
+            torch.manual_seed(cfg.seed)
+        T = int(lookback)
+        E = self.E
+        p = _LSTMPackedParams(self.module, device)
+        self.p = p
+        Lc = p.Lc
+        F_out = self.module.n_features_out
+
+        y_is_x = y is X or (X.shape == y.shape and np.shares_memory(X, y))
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), device)
+        Yd = Xd if y_is_x else pad_rows_bf16(np.asarray(y, np.float32), device)
+        N = Xd.shape[0]
+        n_windows = N - T + 1 - lookahead
+        if n_windows <= 0:
+            raise ValueError(f"need > {T + lookahead} rows for lookback={T}")
+
+        B = max(64, min(cfg.batch_size, n_windows))
+        n_batches = max(1, math.ceil(n_windows / B))
+        st = self._alloc_step_state(Xd, Yd, B, T, n_batches, n_windows, lookahead, F_out)
+
+        opt = torch.optim.Adam(
+            p.masters(), lr=cfg.lr, betas=cfg.betas, eps=cfg.eps, foreach=True
+        )
+
+        history = TrainHistory(engine="fused_hip_lstm")
+        denom = float(n_batches) * B * F_out
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(cfg.epochs):
+            self._fill_perm(st)
+            for s in range(n_batches):
+                self._step(st, s, opt)
+            history.loss.append(float(st["loss_acc"].item()) / denom)
+            st["loss_acc"].zero_()
+        torch.cuda.synchronize()
+        history.duration_sec = time.perf_counter() - t0
+        history.epochs = cfg.epochs
+        history.samples_per_sec = (
+            cfg.epochs * n_batches * B / history.duration_sec if history.duration_sec else 0.0
+        )
+        p.sync_to_module()
+        return history
+
+    # ------------------------------------------------------------------
+    def _alloc_step_state(self, Xd, Yd, B, T, n_batches, n_windows, lookahead, F_out):
+        device = self.device
+        Lc = self.p.Lc
+        bf16 = torch.bfloat16
+        st = {
+            "Xd": Xd, "Yd": Yd, "B": B, "T": T, "n_batches": n_batches,
+            "n_windows": n_windows, "lookahead": lookahead, "F_out": F_out,
+            "perm": torch.zeros(n_batches * B, dtype=torch.int32, device=device),
+            "h_seq": [torch.zeros(B, T, WMAX, dtype=bf16, device=device) for _ in range(Lc)],
+            "c_seq": [torch.zeros(B, T, WMAX, dtype=bf16, device=device) for _ in range(Lc)],
+            "gates_seq": [torch.zeros(B, T, H4, dtype=bf16, device=device) for _ in range(Lc)],
+            "x_seq0": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
+            "dgates": torch.zeros(B, T, H4, dtype=bf16, device=device),
+            "dh_seq": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
+            "hprev": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
+            "h_last": torch.zeros(B, WMAX, dtype=bf16, device=device),
+            "y_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
+            "dx_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
+            "perm_id": torch.arange(B, dtype=torch.int32, device=device),
+            "counters0": torch.tensor([0, 1], dtype=torch.int32, device=device),
+            "head_dw": torch.zeros((B + 63) // 64, 1, WMAX * WMAX, dtype=torch.float32, device=device),
+            "head_db": torch.zeros((B + 63) // 64, 1, WMAX, dtype=torch.float32, device=device),
+            "loss_acc": torch.zeros(1, dtype=torch.float32, device=device),
+            "gscale": 2.0 / (B * F_out),
+            "act_out": [ACT_IDS[self.module.out_func]],
+        }
+        return st
+
+    def _fill_perm(self, st):
+        nw, need = st["n_windows"], st["perm"].numel()
+        idx = torch.randperm(nw, device=self.device)
+        if need > nw:
+            idx = torch.cat([idx, torch.randint(0, nw, (need - nw,), device=self.device)])
+        st["perm"].copy_(idx[:need].to(torch.int32))
+
+    # ------------------------------------------------------------------
+    def _forward(self, st, win: torch.Tensor):
+        """Run the stacked fused forward for window starts ``win`` [B] int32."""
+        E, p = self.E, self.p
+        B, T = st["B"], st["T"]
+        for l in range(p.Lc):
+            src = st["Xd"] if l == 0 else st["h_seq"][l - 1]
+            E.lstm_fwd(
+                src, win if l == 0 else None, p.wcat_bf16[l], p.bias_master[l],
+                st["h_seq"][l], st["gates_seq"][l], st["c_seq"][l],
+                st["x_seq0"] if l == 0 else None, B, T,
+            )
+        st["h_last"].copy_(st["h_seq"][p.Lc - 1][:, T - 1, :])
+
+    def _step(self, st, s: int, opt):
+        win = st["perm"][s * st["B"] : (s + 1) * st["B"]]
+        self._compute_grads(st, win)
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        self.p.refresh_bf16()
+
+    def _compute_grads(self, st, win: torch.Tensor):
+        """Forward + head + BPTT; leaves fp32 grads on every master tensor."""
+        E, p = self.E, self.p
+        B, T = st["B"], st["T"]
+        self._forward(st, win)
+
+        # ---- head: dense fwd + MSE + bwd (dx_out = dL/dh_{T-1}) ----------
+        tgt = win.long() + (T - 1 + st["lookahead"])
+        torch.index_select(st["Yd"], 0, tgt, out=st["y_head"])
+        E.hg_train_step(
+            st["h_last"], st["y_head"], st["perm_id"], st["counters0"],
+            p.head_w, p.head_wt, p.head_bias, st["head_dw"], st["head_db"],
+            st["loss_acc"], B, st["F_out"], st["act_out"], st["gscale"], False,
+            st["dx_head"],
+        )
+
+        # ---- BPTT + bulk grad GEMMs, top layer down ----------------------
+        dg2 = st["dgates"].view(-1, H4)
+        for l in reversed(range(p.Lc)):
+            dh = st["dx_head"] if l == p.Lc - 1 else st["dh_seq"]
+            E.lstm_bwd_cell(
+                dh, st["gates_seq"][l], st["c_seq"][l], p.wcat_t_bf16[l],
+                st["dgates"], B, T, l == p.Lc - 1,
+            )
+            x_seq = st["x_seq0"] if l == 0 else st["h_seq"][l - 1]
+            st["hprev"][:, 0].zero_()
+            st["hprev"][:, 1:].copy_(st["h_seq"][l][:, : T - 1])
+            # dWcat = dgates^T x [x_seq | hprev]  (K = B*T, rocBLAS bf16 TN)
+            gw = torch.zeros_like(p.wcat_master[l])
+            gw[:, :WMAX] = torch.matmul(dg2.t(), x_seq.view(-1, WMAX)).float()
+            gw[:, WMAX:] = torch.matmul(dg2.t(), st["hprev"].view(-1, WMAX)).float()
+            p.wcat_master[l].grad = gw
+            p.bias_master[l].grad = dg2.sum(0, dtype=torch.float32)
+            if l > 0:
+                # dx for the layer below: dgates x W_ih  [B*T, 64]
+                torch.matmul(dg2, p.wcat_bf16[l][:, :WMAX], out=st["dh_seq"].view(-1, WMAX))
+
+        # ---- head grads from the hg slabs (slab layout is [in][out]) -----
+        p.head_wm.grad = st["head_dw"].sum(0).view(WMAX, WMAX).t().contiguous()
+        p.head_bm.grad = st["head_db"].sum(0).view(WMAX)
+
+
+class LSTMStepTrainer:
+    """Step-oriented fused LSTM trainer (bench + fleet building; DP-capable).
+
+    DP exchanges ONE flat fp32 grad bucket per step over RCCL (the whole
+    stacked model is ~0.5 MB — latency-bound on xGMI, so a single fused
+    all-reduce, mirroring parallel/ddp.py's hourglass trainer).
+    """
+
+    def __init__(self, module, X: np.ndarray, batch_size: int, cfg, dp=None, lookahead: int = 0):
+        from ..parallel.ddp import DPContext
+
+        self.dp = dp or DPContext()
+        self.device = self.dp.device
+        self.cfg = cfg
+        self.module = module
+        self.tr = FusedLSTMTrainer(module, self.device, cfg)
+        p = _LSTMPackedParams(module, self.device)
+        self.tr.p = p
+        self.p = p
+        # ranks start from identical weights
+        if self.dp.enabled:
+            for t in p.masters():
+                self.dp.broadcast(t)
+            p.refresh_bf16()
+        T = module.lookback_window
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device)
+        n_windows = Xd.shape[0] - T + 1 - lookahead
+        B = max(64, min(batch_size, n_windows))
+        self.B = B
+        n_batches = max(1, math.ceil(n_windows / B))
+        self.st = self.tr._alloc_step_state(
+            Xd, Xd, B, T, n_batches, n_windows, lookahead, module.n_features_out
+        )
+        self.tr._fill_perm(self.st)
+        self.opt = torch.optim.Adam(
+            p.masters(), lr=cfg.lr, betas=cfg.betas, eps=cfg.eps, foreach=True
+        )
+        self._s = 0
+        if self.dp.enabled:
+            self._flat = torch.zeros(
+                sum(t.numel() for t in p.masters()), dtype=torch.float32, device=self.device
+            )
+
+    def step(self):
+        st = self.st
+        win = st["perm"][self._s * self.B : (self._s + 1) * self.B]
+        self.tr._compute_grads(st, win)
+        if self.dp.enabled:
+            ms = self.p.masters()
+            off = 0
+            for t in ms:
+                n = t.numel()
+                self._flat[off : off + n].copy_(t.grad.view(-1))
+                off += n
+            self.dp.all_reduce_mean(self._flat)
+            off = 0
+            for t in ms:
+                n = t.numel()
+                t.grad.view(-1).copy_(self._flat[off : off + n])
+                off += n
+        self.opt.step()
+        self.opt.zero_grad(set_to_none=True)
+        self.p.refresh_bf16()
+        self._s += 1
+        if self._s >= st["n_batches"]:
+            self.tr._fill_perm(st)
+            self._s = 0
+
+    def synchronize(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+
+class FusedLSTMForward:
+    """Serving/predict forward through the fused cell kernels."""
+
+    def __init__(self, module, device: torch.device):
+        reason = lstm_supported(module)
+        if reason:
+            raise ValueError(f"fused LSTM path unsupported: {reason}")
+        self.E = ext()
+        self.device = device
+        self.module = module
+        self.p = _LSTMPackedParams(module, device)
+
+    def predict(self, X: np.ndarray, lookahead: int = 0, batch_size: int = 16384) -> np.ndarray:
+        """X [N, F] -> predictions for every window, [N - T + 1, F_out]."""
+        p = self.p
+        T = self.module.lookback_window
+        F_out = self.module.n_features_out
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device)
+        N = Xd.shape[0]
+        n_windows = N - T + 1
+        if n_windows <= 0:
+            raise ValueError(f"need >= {T} rows for lookback={T}")
+        outs = []
+        bf16 = torch.bfloat16
+        B = min(batch_size, n_windows)
+        h_seq = [torch.zeros(B, T, WMAX, dtype=bf16, device=self.device) for _ in range(p.Lc)]
+        c_seq = torch.zeros(B, T, WMAX, dtype=bf16, device=self.device)
+        gates = torch.zeros(B, T, H4, dtype=bf16, device=self.device)
+        for s in range(0, n_windows, B):
+            n = min(B, n_windows - s)
+            win = torch.arange(s, s + n, dtype=torch.int32, device=self.device)
+            if n < B:
+                win = torch.cat([win, torch.zeros(B - n, dtype=torch.int32, device=self.device)])
+            for l in range(p.Lc):
+                src = Xd if l == 0 else h_seq[l - 1]
+                self.E.lstm_fwd(
+                    src, win if l == 0 else None, p.wcat_bf16[l], p.bias_master[l],
+                    h_seq[l], gates, c_seq, None, B, T,
+                )
+            h_last = h_seq[p.Lc - 1][:, T - 1, :].float()
+            yhat = torch.addmm(p.head_bm, h_last, p.head_wm.t())[:n, :F_out]
+            af = self.module.out_func
+            if af and af != "linear":
+                yhat = {"tanh": torch.tanh, "relu": torch.relu, "sigmoid": torch.sigmoid}[af](yhat)
+            outs.append(yhat.cpu().numpy())
+        return np.concatenate(outs, axis=0)

@@ -108,7 +108,7 @@ def test_train_step_gradients_match_autograd():
     perm = torch.arange(N, dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
     E.hg_train_step(Xd, Xd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                    loss, B, F_out, p.act_ids, gscale, True)
+                    loss, B, F_out, p.act_ids, gscale, True, None)
     torch.cuda.synchronize()
 
     # --- torch reference on the same bf16-rounded input ------------------

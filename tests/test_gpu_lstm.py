@@ -1,0 +1,153 @@
+"""GPU numerics tests for the fused LSTM cell kernels (K4/K5/K6) vs plain
+PyTorch fp32 references. All @pytest.mark.gpu."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from gordo_amd.model.torch_nets import LSTMNet
+    from gordo_amd.ops.lstm_engine import (
+        H4,
+        WMAX,
+        FusedLSTMForward,
+        FusedLSTMTrainer,
+        _LSTMPackedParams,
+    )
+    from gordo_amd.runtime.trainer import TrainConfig
+else:
+    pytest.skip("no GPU", allow_module_level=True)
+
+DEV = torch.device("cuda:0")
+
+
+def make_lstm(n_features=20, dims=(32,), T=12, seed=0, out=None):
+    torch.manual_seed(seed)
+    return LSTMNet(
+        n_features, out or n_features, list(dims), [None] * len(dims),
+        out_func="linear", lookback_window=T,
+    )
+
+
+def ref_forward(net, X, T):
+    """fp32 torch forward over all rolling windows of bf16-rounded X."""
+    Xb = torch.as_tensor(X).to(torch.bfloat16).float()
+    n = len(Xb) - T + 1
+    w = torch.stack([Xb[i : i + T] for i in range(n)])
+    with torch.no_grad():
+        return net(w).numpy()
+
+
+def test_lstm_fused_forward_matches_torch():
+    T = 12
+    net = make_lstm(T=T, seed=1)
+    X = np.random.default_rng(0).normal(size=(200, 20)).astype(np.float32)
+    ref = ref_forward(net, X, T)
+    out = FusedLSTMForward(net, DEV).predict(X)
+    assert out.shape == ref.shape
+    np.testing.assert_allclose(out, ref, atol=4e-2, rtol=4e-2)
+
+
+def test_lstm_fused_forward_stacked_long():
+    """2-layer stack, lookback 144 (the headline config's shape)."""
+    T = 144
+    net = make_lstm(n_features=50, dims=(48, 24), T=T, seed=2)
+    X = np.random.default_rng(1).normal(size=(400, 50)).astype(np.float32)
+    ref = ref_forward(net, X, T)
+    out = FusedLSTMForward(net, DEV).predict(X)
+    assert out.shape == ref.shape
+    # longer bf16 recurrence accumulates more rounding
+    np.testing.assert_allclose(out, ref, atol=8e-2, rtol=8e-2)
+
+
+def test_lstm_grads_match_autograd():
+    """One fused step's wcat/bias/head grads vs fp32 torch autograd."""
+    T, B, F = 12, 256, 20
+    net = make_lstm(n_features=F, dims=(32,), T=T, seed=3)
+    ref_net = make_lstm(n_features=F, dims=(32,), T=T, seed=3)
+    X = np.random.default_rng(2).normal(size=(B + T + 2, F)).astype(np.float32)
+
+    cfg = TrainConfig(epochs=1, batch_size=B)
+    tr = FusedLSTMTrainer(net, DEV, cfg)
+    p = _LSTMPackedParams(net, DEV)
+    tr.p = p
+    from gordo_amd.ops.hourglass_engine import pad_rows_bf16
+
+    Xd = pad_rows_bf16(X, DEV)
+    st = tr._alloc_step_state(Xd, Xd, B, T, 1, B, 0, F)
+    win = torch.arange(B, dtype=torch.int32, device=DEV)
+    tr._compute_grads(st, win)
+    torch.cuda.synchronize()
+
+    # fp32 reference on the same bf16-rounded inputs
+    Xb = torch.as_tensor(X).to(torch.bfloat16).float()
+    wins = torch.stack([Xb[i : i + T] for i in range(B)])
+    tgt = torch.stack([Xb[i + T - 1] for i in range(B)])
+    out = ref_net(wins)
+    loss = torch.nn.functional.mse_loss(out, tgt)
+    loss.backward()
+
+    # loss agreement
+    kern_loss = float(st["loss_acc"].item()) / (B * F)
+    assert abs(kern_loss - float(loss)) / max(float(loss), 1e-9) < 6e-2
+
+    cell = ref_net.cells[0]
+    H, I = cell.hidden_size, cell.input_size
+    gw = p.wcat_master[0].grad.cpu()
+    gb = p.bias_master[0].grad.cpu()
+    for g in range(4):
+        ref_ih = cell.weight_ih_l0.grad[g * H : (g + 1) * H]
+        ref_hh = cell.weight_hh_l0.grad[g * H : (g + 1) * H]
+        # torch's two bias vectors get IDENTICAL grads (both add to the same
+        # pre-activation); the fused combined bias's grad equals either one.
+        ref_b = cell.bias_ih_l0.grad[g * H : (g + 1) * H]
+        got_ih = gw[g * WMAX : g * WMAX + H, :I]
+        got_hh = gw[g * WMAX : g * WMAX + H, WMAX : WMAX + H]
+        got_b = gb[g * WMAX : g * WMAX + H]
+        s = ref_ih.abs().max().item() or 1.0
+        assert (got_ih - ref_ih).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} dW_ih"
+        s = ref_hh.abs().max().item() or 1.0
+        assert (got_hh - ref_hh).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} dW_hh"
+        s = ref_b.abs().max().item() or 1.0
+        assert (got_b - ref_b).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} db"
+        # padded rows/cols carry zero grad
+        assert gw[g * WMAX + H : (g + 1) * WMAX].abs().max().item() == 0.0
+    # head grads
+    ref_hw = ref_net.head.weight.grad
+    got_hw = p.head_wm.grad.cpu()[: ref_hw.shape[0], : ref_hw.shape[1]]
+    s = ref_hw.abs().max().item() or 1.0
+    assert (got_hw - ref_hw).abs().max().item() < 0.08 * s + 2e-4, "head dW"
+
+
+def test_lstm_estimator_learns_fused():
+    from gordo_amd.model.models import TorchLSTMAutoEncoder
+
+    rng = np.random.default_rng(4)
+    t = np.arange(3000)
+    base = np.stack(
+        [np.sin(t / (8 + i)) + 0.05 * rng.standard_normal(len(t)) for i in range(20)],
+        axis=1,
+    ).astype(np.float32)
+    model = TorchLSTMAutoEncoder(
+        kind="lstm_hourglass", lookback_window=16, epochs=5, batch_size=512, lr=3e-3
+    )
+    model.fit(base)
+    assert model.history["engine"] == "fused_hip_lstm", model.history
+    assert model.history["loss"][-1] < model.history["loss"][0], model.history["loss"]
+    out = model.predict(base[:200])
+    assert out.shape == (200 - 16 + 1, 20)
+    assert np.isfinite(out).all()
+
+
+def test_lstm_forecast_fused():
+    from gordo_amd.model.models import TorchLSTMForecast
+
+    X = np.random.default_rng(5).normal(size=(1500, 10)).astype(np.float32)
+    model = TorchLSTMForecast(
+        kind="lstm_symmetric", dims=[32], lookback_window=12, epochs=2, batch_size=256
+    )
+    model.fit(X)
+    assert model.history["engine"] == "fused_hip_lstm", model.history
+    score = model.score(X)
+    assert np.isfinite(score)
