@@ -11,8 +11,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hg_forward", &hg_forward_launch, "fused serving forward (+anomaly)");
     m.def("hg_grad_reduce", &hg_grad_reduce_launch, "slab reduce -> flat grad bucket");
     m.def("hg_adam_flat", &hg_adam_flat_launch, "Adam from reduced flat grad");
+    m.def("hg_slab_reduce", &hg_slab_reduce_launch, "stage-A slab reduce (+counter bump)");
+    m.def("hg_adam_p", &hg_adam_p_launch, "stage-B Adam over G partials");
+    m.def("hg_partial_flat", &hg_partial_flat_launch, "partials -> flat DP bucket");
     m.def("lstm_fwd", &lstm_fwd_launch, "persistent fused 4-gate LSTM forward");
-    m.def("lstm_bwd_cell", &lstm_bwd_cell_launch, "reverse-time BPTT cell kernel");
+    m.def("lstm_bwd_cell", &lstm_bwd_cell_launch, "reverse-time BPTT cell kernel (+fused dx)");
+    m.def("lstm_dw", &lstm_dw_launch, "split-K dWcat + dbias from one dgates pass");
     m.attr("WMAX") = 64;
     m.attr("MAXL") = 5;
     m.attr("MT") = 64;

@@ -340,6 +340,91 @@ __global__ void hg_bump(int* counters, int n_batches) {
 }
 
 // ===========================================================================
+// Two-stage slab reduction (the single-stage hg_adam loop over 128 slabs with
+// only P/256 blocks was 51% of the hourglass step): stage A fans the slab sum
+// over G groups (G x P/256 blocks, ~7x more parallelism), folding the counter
+// bump in (block (0,0) — no other block touches counters, and every consumer
+// of counters runs later in the stream); stage B is Adam over the G partials.
+// Layout conversion slab[l][in][out] -> master[l][out][in] happens in stage A.
+// ===========================================================================
+__launch_bounds__(256)
+__global__ void hg_slab_reduce(const float* __restrict__ dw,
+                               const float* __restrict__ db,
+                               float* __restrict__ partial,  // [G][P]
+                               int* __restrict__ counters,
+                               int n_slabs, int G, int L, int n_batches) {
+    const int PW = L * GA_WMAX * GA_WMAX;
+    const int P = PW + L * GA_WMAX;
+    int g = blockIdx.y;
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g == 0 && idx == 0 && counters != nullptr) {
+        counters[0] = (counters[0] + 1) % n_batches;
+        counters[1] += 1;
+    }
+    if (idx >= P) return;
+    int per = (n_slabs + G - 1) / G;
+    int s0 = g * per, s1 = min(n_slabs, s0 + per);
+    float s = 0.f;
+    if (idx < PW) {
+        int l = idx / (GA_WMAX * GA_WMAX);
+        int rem = idx - l * GA_WMAX * GA_WMAX;
+        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
+        size_t sidx = (size_t)l * GA_WMAX * GA_WMAX + (size_t)in * GA_WMAX + out;
+        for (int sl = s0; sl < s1; ++sl) s += dw[(size_t)sl * PW + sidx];
+    } else {
+        int bidx = idx - PW;
+        for (int sl = s0; sl < s1; ++sl) s += db[(size_t)sl * (L * GA_WMAX) + bidx];
+    }
+    partial[(size_t)g * P + idx] = s;
+}
+
+__launch_bounds__(256)
+__global__ void hg_adam_p(
+    float* __restrict__ master, float* __restrict__ m_buf, float* __restrict__ v_buf,
+    const float* __restrict__ partial, unsigned short* __restrict__ w_bf16,
+    unsigned short* __restrict__ wt_bf16, float* __restrict__ bias,
+    const int* __restrict__ counters, int G, int L,
+    float lr, float beta1, float beta2, float eps) {
+    const int PW = L * GA_WMAX * GA_WMAX;
+    const int P = PW + L * GA_WMAX;
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= P) return;
+    float g = 0.f;
+    for (int i = 0; i < G; ++i) g += partial[(size_t)i * P + idx];
+    int t = counters[1];
+    float m = beta1 * m_buf[idx] + (1.f - beta1) * g;
+    float v = beta2 * v_buf[idx] + (1.f - beta2) * g * g;
+    m_buf[idx] = m;
+    v_buf[idx] = v;
+    float mhat = m / (1.f - __powf(beta1, (float)t));
+    float vhat = v / (1.f - __powf(beta2, (float)t));
+    float p = master[idx] - lr * mhat / (__fsqrt_rn(vhat) + eps);
+    master[idx] = p;
+    if (idx < PW) {
+        unsigned short pb = ga_f32_to_bf16(p);
+        w_bf16[idx] = pb;
+        int l = idx / (GA_WMAX * GA_WMAX);
+        int rem = idx - l * GA_WMAX * GA_WMAX;
+        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
+        wt_bf16[(size_t)l * GA_WMAX * GA_WMAX + in * GA_WMAX + out] = pb;
+    } else {
+        bias[idx - PW] = p;
+    }
+}
+
+// partial [G][P] -> flat grad bucket [P] (DP all-reduce input)
+__launch_bounds__(256)
+__global__ void hg_partial_flat(const float* __restrict__ partial,
+                                float* __restrict__ g_out, int G, int L) {
+    const int P = L * GA_WMAX * GA_WMAX + L * GA_WMAX;
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= P) return;
+    float s = 0.f;
+    for (int i = 0; i < G; ++i) s += partial[(size_t)i * P + idx];
+    g_out[idx] = s;
+}
+
+// ===========================================================================
 // Serving forward (fused K1+K7+K8): raw fp32 input -> min-max scale -> L-layer
 // forward -> yhat (scaled space); optionally per-feature scaled |err| and row
 // L2 total (autoencoder anomaly: y = scaled x). hipGraph-capturable.
@@ -606,6 +691,42 @@ void hg_forward_launch(torch::Tensor X, torch::Tensor w, torch::Tensor bias,
     size_t smem = 3 * GA_TILE_BYTES + 4 * GA_WMAX * 4;
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(hg_forward, dim3(n_tiles), dim3(256), smem, stream.stream(), a);
+}
+
+void hg_slab_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor partial,
+                           c10::optional<torch::Tensor> counters, int64_t n_slabs,
+                           int64_t G, int64_t L, int64_t n_batches) {
+    int P = (int)(L * GA_WMAX * GA_WMAX + L * GA_WMAX);
+    dim3 grid((P + 255) / 256, (int)G);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(hg_slab_reduce, grid, dim3(256), 0, stream.stream(),
+                       dw.data_ptr<float>(), db.data_ptr<float>(),
+                       partial.data_ptr<float>(),
+                       counters.has_value() ? counters->data_ptr<int>() : nullptr,
+                       (int)n_slabs, (int)G, (int)L, (int)n_batches);
+}
+
+void hg_adam_p_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                      torch::Tensor partial, torch::Tensor w, torch::Tensor wt,
+                      torch::Tensor bias, torch::Tensor counters, int64_t G,
+                      int64_t L, double lr, double beta1, double beta2, double eps) {
+    int P = (int)(L * GA_WMAX * GA_WMAX + L * GA_WMAX);
+    int blocks = (P + 255) / 256;
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(hg_adam_p, dim3(blocks), dim3(256), 0, stream.stream(),
+                       master.data_ptr<float>(), m.data_ptr<float>(), v.data_ptr<float>(),
+                       partial.data_ptr<float>(), (unsigned short*)w.data_ptr(),
+                       (unsigned short*)wt.data_ptr(), bias.data_ptr<float>(),
+                       counters.data_ptr<int>(), (int)G, (int)L, (float)lr,
+                       (float)beta1, (float)beta2, (float)eps);
+}
+
+void hg_partial_flat_launch(torch::Tensor partial, torch::Tensor g, int64_t G, int64_t L) {
+    int P = (int)(L * GA_WMAX * GA_WMAX + L * GA_WMAX);
+    int blocks = (P + 255) / 256;
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(hg_partial_flat, dim3(blocks), dim3(256), 0, stream.stream(),
+                       partial.data_ptr<float>(), g.data_ptr<float>(), (int)G, (int)L);
 }
 
 void hg_grad_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,

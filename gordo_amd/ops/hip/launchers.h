@@ -1,7 +1,7 @@
-// Host-callable launch functions implemented in the .hip TUs.
+// Host launch-function declarations shared by bindings.cpp (host-only TU) and
+// the .hip TUs that define them.
 #pragma once
 #include <torch/extension.h>
-
 #include <vector>
 
 void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
@@ -33,6 +33,17 @@ void hg_adam_flat_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                          torch::Tensor bias, torch::Tensor counters, int64_t L,
                          double lr, double beta1, double beta2, double eps);
 
+void hg_slab_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor partial,
+                           c10::optional<torch::Tensor> counters, int64_t n_slabs,
+                           int64_t G, int64_t L, int64_t n_batches);
+
+void hg_adam_p_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                      torch::Tensor partial, torch::Tensor w, torch::Tensor wt,
+                      torch::Tensor bias, torch::Tensor counters, int64_t G,
+                      int64_t L, double lr, double beta1, double beta2, double eps);
+
+void hg_partial_flat_launch(torch::Tensor partial, torch::Tensor g, int64_t G, int64_t L);
+
 void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
                      torch::Tensor wcat, torch::Tensor bias, torch::Tensor h_seq,
                      torch::Tensor gates_seq, torch::Tensor c_seq,
@@ -40,5 +51,9 @@ void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
 
 void lstm_bwd_cell_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
                           torch::Tensor c_seq, torch::Tensor wcat_t,
-                          torch::Tensor dgates_seq, int64_t B, int64_t T,
-                          bool dh_last_only);
+                          torch::Tensor dgates_seq, c10::optional<torch::Tensor> dx_seq,
+                          int64_t B, int64_t T, bool dh_last_only);
+
+void lstm_dw_launch(torch::Tensor dgates, torch::Tensor x_seq, torch::Tensor h_seq,
+                    torch::Tensor partial, torch::Tensor db_partial,
+                    int64_t BT, int64_t T);

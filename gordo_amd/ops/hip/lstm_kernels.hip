@@ -41,6 +41,10 @@ struct LstmFwdArgs {
 
 extern __shared__ __attribute__((aligned(16))) char ga_smem_lstm[];
 
+// LDS plan (57.25 KiB -> 2 blocks/CU): x, h, c + 4 gate tiles + bias.
+// The cell epilogue lands MFMA-layout results in LDS tiles; a vectorized
+// writer phase then streams them to HBM as 16-B stores (the original direct
+// per-element u16 stores — 96 scalar stores/lane/t — were the fwd bottleneck).
 __launch_bounds__(256)
 __global__ void lstm_fwd(LstmFwdArgs a) {
     const int tid = threadIdx.x;
@@ -49,9 +53,11 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
     const int base = blockIdx.x * GA_MT;
     const int wrow = wid * 16;
 
-    char* x_tile = ga_smem_lstm;                    // [64][64] swizzled bf16
-    char* h_tile = ga_smem_lstm + GA_TILE_BYTES;    // [64][64] swizzled bf16
-    float* s_bias = (float*)(ga_smem_lstm + 2 * GA_TILE_BYTES);  // [256]
+    char* x_tile = ga_smem_lstm;                     // [64][64] swizzled bf16
+    char* h_tile = ga_smem_lstm + GA_TILE_BYTES;     // [64][64]
+    char* c_tile = ga_smem_lstm + 2 * GA_TILE_BYTES; // [64][64]
+    char* g_tiles = ga_smem_lstm + 3 * GA_TILE_BYTES;  // 4 x [64][64] i,f,g,o
+    float* s_bias = (float*)(ga_smem_lstm + 7 * GA_TILE_BYTES);  // [256]
 
     for (int i = tid; i < LSTM_4H; i += 256) s_bias[i] = a.bias[i];
 
@@ -67,6 +73,11 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
     for (int jt = 0; jt < 4; ++jt)
 #pragma unroll
         for (int r = 0; r < 4; ++r) c_reg[jt][r] = 0.f;
+
+    // writer-phase mapping: row m = tid>>2, 16 cols starting at (tid&3)*16
+    const int wm = tid >> 2;
+    const int wc0 = (tid & 3) * 16;
+    const int wgi = base + wm;
 
     __syncthreads();
 
@@ -107,7 +118,7 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
         }
         __syncthreads();  // x_tile/h_tile reads done; epilogue rewrites h_tile
 
-        // ---- cell update (lane-local: gates i,f,g,o of one (m,j) live in
+        // ---- cell update -> LDS tiles (gates i,f,g,o of one (m,j) live in
         //      acc[jt], acc[jt+4], acc[jt+8], acc[jt+12]) -------------------
 #pragma unroll
         for (int jt = 0; jt < 4; ++jt) {
@@ -115,7 +126,6 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int m = wrow + (lane >> 4) * 4 + r;
-                int gi = base + m;
                 float i_g = 1.f / (1.f + __expf(-(acc[jt][r] + s_bias[col])));
                 float f_g = 1.f / (1.f + __expf(-(acc[jt + 4][r] + s_bias[GA_WMAX + col])));
                 float g_g = tanhf(acc[jt + 8][r] + s_bias[2 * GA_WMAX + col]);
@@ -123,20 +133,33 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
                 float c_new = f_g * c_reg[jt][r] + i_g * g_g;
                 c_reg[jt][r] = c_new;
                 float h_new = o_g * tanhf(c_new);
-                unsigned short hb = ga_f32_to_bf16(h_new);
-                ga_tile_write1(h_tile, m, col, hb);
-                if (gi < a.B) {
-                    size_t bt = (size_t)gi * a.T + t;
-                    a.h_seq[bt * GA_WMAX + col] = hb;
-                    a.c_seq[bt * GA_WMAX + col] = ga_f32_to_bf16(c_new);
-                    a.gates_seq[bt * LSTM_4H + col] = ga_f32_to_bf16(i_g);
-                    a.gates_seq[bt * LSTM_4H + GA_WMAX + col] = ga_f32_to_bf16(f_g);
-                    a.gates_seq[bt * LSTM_4H + 2 * GA_WMAX + col] = ga_f32_to_bf16(g_g);
-                    a.gates_seq[bt * LSTM_4H + 3 * GA_WMAX + col] = ga_f32_to_bf16(o_g);
-                }
+                ga_tile_write1(h_tile, m, col, ga_f32_to_bf16(h_new));
+                ga_tile_write1(c_tile, m, col, ga_f32_to_bf16(c_new));
+                ga_tile_write1(g_tiles, m, col, ga_f32_to_bf16(i_g));
+                ga_tile_write1(g_tiles + GA_TILE_BYTES, m, col, ga_f32_to_bf16(f_g));
+                ga_tile_write1(g_tiles + 2 * GA_TILE_BYTES, m, col, ga_f32_to_bf16(g_g));
+                ga_tile_write1(g_tiles + 3 * GA_TILE_BYTES, m, col, ga_f32_to_bf16(o_g));
             }
         }
         __syncthreads();
+
+        // ---- vectorized writer: 16-B stores of h, c, gates ---------------
+        if (wgi < a.B) {
+            size_t bt = (size_t)wgi * a.T + t;
+#pragma unroll
+            for (int half = 0; half < 2; ++half) {
+                int c0 = wc0 + half * 8;
+                *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + c0) =
+                    *(ga_u16x8*)&((char*)h_tile)[ga_sw(wm, c0)];
+                *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + c0) =
+                    *(ga_u16x8*)&((char*)c_tile)[ga_sw(wm, c0)];
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + g * GA_WMAX + c0) =
+                        *(ga_u16x8*)&((char*)g_tiles)[g * GA_TILE_BYTES + ga_sw(wm, c0)];
+            }
+        }
+        __syncthreads();  // writer reads done before next epilogue overwrites
     }
 }
 
@@ -151,6 +174,10 @@ struct LstmBwdArgs {
     const unsigned short* __restrict__ c_seq;      // [B][T][64]
     const unsigned short* __restrict__ wcat_t;     // [128][256] bf16 (WcatT)
     unsigned short* __restrict__ dgates_seq;       // [B][T][256] out
+    unsigned short* __restrict__ dx_seq;           // [B][T][64] out, or null:
+                                                   // dx = dgates x W_ih fused
+                                                   // (skips a rocBLAS pass
+                                                   // re-reading dgates)
     int B, T;
     int dh_last_only;  // 1: dh_out is [B][64] at the final step only
 };
@@ -173,90 +200,254 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
     }
     // dc state registers: elementwise thread mapping m = tid>>2, 16 cols each
     const int em = tid >> 2;          // 0..63 row
-    const int eg = tid & 3;           // col group: cols eg*16 + e
+    const int eg = tid & 3;           // col group: cols eg*16 .. eg*16+15
+    const int c0 = eg * 16;
     float dc_reg[16];
 #pragma unroll
     for (int e = 0; e < 16; ++e) dc_reg[e] = 0.f;
     __syncthreads();
 
     for (int t = a.T - 1; t >= 0; --t) {
-        // ---- elementwise BPTT state update + dgates -> LDS + global -------
+        // ---- elementwise BPTT: all global IO as 16-B vectors --------------
         {
-            int gi = base + em;
-            size_t bt = (size_t)gi * a.T + t;
+            const int gi = base + em;
+            const size_t bt = (size_t)gi * a.T + t;
+            const bool live = gi < a.B;
+            ga_u16x8 vi[2], vf[2], vg[2], vo[2], vc[2], vcp[2], vdh[2];
+            ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
-            for (int e = 0; e < 16; ++e) {
-                int col = eg * 16 + e;
-                float dh = 0.f, i_g = 0.f, f_g = 0.f, g_g = 0.f, o_g = 0.f;
-                float c_t = 0.f, c_prev = 0.f;
-                if (gi < a.B) {
-                    if (a.dh_last_only) {
-                        dh = (t == a.T - 1)
-                                 ? ga_bf16_to_f32(a.dh_out[(size_t)gi * GA_WMAX + col])
-                                 : 0.f;
-                    } else {
-                        dh = ga_bf16_to_f32(a.dh_out[bt * GA_WMAX + col]);
-                    }
-                    i_g = ga_bf16_to_f32(a.gates_seq[bt * LSTM_4H + col]);
-                    f_g = ga_bf16_to_f32(a.gates_seq[bt * LSTM_4H + GA_WMAX + col]);
-                    g_g = ga_bf16_to_f32(a.gates_seq[bt * LSTM_4H + 2 * GA_WMAX + col]);
-                    o_g = ga_bf16_to_f32(a.gates_seq[bt * LSTM_4H + 3 * GA_WMAX + col]);
-                    c_t = ga_bf16_to_f32(a.c_seq[bt * GA_WMAX + col]);
-                    c_prev = (t > 0) ? ga_bf16_to_f32(a.c_seq[(bt - 1) * GA_WMAX + col]) : 0.f;
+            for (int hh = 0; hh < 2; ++hh) {
+                int cc = c0 + hh * 8;
+                if (live) {
+                    vi[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc);
+                    vf[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
+                    vg[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
+                    vo[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
+                    vc[hh] = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
+                    vcp[hh] = (t > 0) ? *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc) : z;
+                    if (a.dh_last_only)
+                        vdh[hh] = (t == a.T - 1)
+                                      ? *(const ga_u16x8*)(a.dh_out + (size_t)gi * GA_WMAX + cc)
+                                      : z;
+                    else
+                        vdh[hh] = *(const ga_u16x8*)(a.dh_out + bt * GA_WMAX + cc);
+                } else {
+                    vi[hh] = vf[hh] = vg[hh] = vo[hh] = vc[hh] = vcp[hh] = vdh[hh] = z;
                 }
-                dh += ga_bf16_to_f32(ga_tile_read1(dh_tile, em, col));
-                float tanhc = tanhf(c_t);
-                float dc = dc_reg[e] + dh * o_g * (1.f - tanhc * tanhc);
-                float do_pre = dh * tanhc * o_g * (1.f - o_g);
-                float di_pre = dc * g_g * i_g * (1.f - i_g);
-                float df_pre = dc * c_prev * f_g * (1.f - f_g);
-                float dg_pre = dc * i_g * (1.f - g_g * g_g);
-                dc_reg[e] = dc * f_g;  // carry to t-1
-                unsigned short di_b = ga_f32_to_bf16(di_pre);
-                unsigned short df_b = ga_f32_to_bf16(df_pre);
-                unsigned short dg_b = ga_f32_to_bf16(dg_pre);
-                unsigned short do_b = ga_f32_to_bf16(do_pre);
-                ga_tile_write1(dg_tiles, em, col, di_b);
-                ga_tile_write1(dg_tiles + GA_TILE_BYTES, em, col, df_b);
-                ga_tile_write1(dg_tiles + 2 * GA_TILE_BYTES, em, col, dg_b);
-                ga_tile_write1(dg_tiles + 3 * GA_TILE_BYTES, em, col, do_b);
-                if (gi < a.B) {
-                    a.dgates_seq[bt * LSTM_4H + col] = di_b;
-                    a.dgates_seq[bt * LSTM_4H + GA_WMAX + col] = df_b;
-                    a.dgates_seq[bt * LSTM_4H + 2 * GA_WMAX + col] = dg_b;
-                    a.dgates_seq[bt * LSTM_4H + 3 * GA_WMAX + col] = do_b;
+            }
+            ga_u16x8 odi[2], odf[2], odg[2], odo[2];
+#pragma unroll
+            for (int hh = 0; hh < 2; ++hh) {
+                ga_u16x8 vrec = *(ga_u16x8*)&dh_tile[ga_sw(em, c0 + hh * 8)];
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    float dh = ga_bf16_to_f32(vdh[hh][e]) + ga_bf16_to_f32(vrec[e]);
+                    float i_g = ga_bf16_to_f32(vi[hh][e]);
+                    float f_g = ga_bf16_to_f32(vf[hh][e]);
+                    float g_g = ga_bf16_to_f32(vg[hh][e]);
+                    float o_g = ga_bf16_to_f32(vo[hh][e]);
+                    float c_t = ga_bf16_to_f32(vc[hh][e]);
+                    float c_prev = ga_bf16_to_f32(vcp[hh][e]);
+                    float tanhc = tanhf(c_t);
+                    int ei = hh * 8 + e;
+                    float dc = dc_reg[ei] + dh * o_g * (1.f - tanhc * tanhc);
+                    odo[hh][e] = ga_f32_to_bf16(dh * tanhc * o_g * (1.f - o_g));
+                    odi[hh][e] = ga_f32_to_bf16(dc * g_g * i_g * (1.f - i_g));
+                    odf[hh][e] = ga_f32_to_bf16(dc * c_prev * f_g * (1.f - f_g));
+                    odg[hh][e] = ga_f32_to_bf16(dc * i_g * (1.f - g_g * g_g));
+                    dc_reg[ei] = dc * f_g;  // carry to t-1
+                }
+                int cc = c0 + hh * 8;
+                *(ga_u16x8*)&dg_tiles[ga_sw(em, cc)] = odi[hh];
+                *(ga_u16x8*)&dg_tiles[GA_TILE_BYTES + ga_sw(em, cc)] = odf[hh];
+                *(ga_u16x8*)&dg_tiles[2 * GA_TILE_BYTES + ga_sw(em, cc)] = odg[hh];
+                *(ga_u16x8*)&dg_tiles[3 * GA_TILE_BYTES + ga_sw(em, cc)] = odo[hh];
+                if (live) {
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + cc) = odi[hh];
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + GA_WMAX + cc) = odf[hh];
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc) = odg[hh];
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc) = odo[hh];
                 }
             }
         }
         __syncthreads();
 
-        if (t == 0) break;  // no recurrent grad into h_{-1}
+        const bool need_rec = (t > 0);
+        const bool need_dx = (a.dx_seq != nullptr);
+        if (!need_rec && !need_dx) break;
 
-        // ---- dh_rec[m][k<H] = dgates[m][256] x W_hh[256->k] ---------------
-        // B[k4h][j=k] = Wcat[k4h][64+j] = WcatT[64+j][k4h]: row-read of WcatT
-        ga_f32x4 acc[4];
+        // ---- shared-A GEMMs over dgates tiles -----------------------------
+        //   dh_rec[m][j<H] = dgates[m][256] x W_hh  (B row-read of WcatT 64+j)
+        //   dx[m][j<64]    = dgates[m][256] x W_ih  (B row-read of WcatT j)
+        ga_f32x4 acc[4], accx[4];
 #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int nt = 0; nt < 4; ++nt) {
+            acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            accx[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+        }
         for (int kk = 0; kk < LSTM_4H; kk += 32) {
             const char* dgt = dg_tiles + (kk >> 6) * GA_TILE_BYTES;
             ga_bf16x8 afrag = ga_tile_read8(dgt, wrow + (lane & 15), (kk & 63) + (lane >> 4) * 8);
+            if (need_rec)
 #pragma unroll
-            for (int nt = 0; nt < 4; ++nt) {
-                ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t + (size_t)(GA_WMAX + nt * 16 + (lane & 15)) * LSTM_4H + kk + (lane >> 4) * 8);
-                acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
-            }
+                for (int nt = 0; nt < 4; ++nt) {
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t + (size_t)(GA_WMAX + nt * 16 + (lane & 15)) * LSTM_4H + kk + (lane >> 4) * 8);
+                    acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
+                }
+            if (need_dx)
+#pragma unroll
+                for (int nt = 0; nt < 4; ++nt) {
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t + (size_t)(nt * 16 + (lane & 15)) * LSTM_4H + kk + (lane >> 4) * 8);
+                    accx[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, accx[nt], 0, 0, 0);
+                }
         }
         __syncthreads();  // dg_tiles reads done before next iteration rewrites
+        if (need_rec)
 #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
-            int col = nt * 16 + (lane & 15);
+            for (int nt = 0; nt < 4; ++nt) {
+                int col = nt * 16 + (lane & 15);
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int m = wrow + (lane >> 4) * 4 + r;
-                ga_tile_write1(dh_tile, m, col, ga_f32_to_bf16(acc[nt][r]));
+                for (int r = 0; r < 4; ++r) {
+                    int m = wrow + (lane >> 4) * 4 + r;
+                    ga_tile_write1(dh_tile, m, col, ga_f32_to_bf16(acc[nt][r]));
+                }
+            }
+        if (need_dx)
+#pragma unroll
+            for (int nt = 0; nt < 4; ++nt) {
+                int col = nt * 16 + (lane & 15);
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int m = wrow + (lane >> 4) * 4 + r;
+                    int gi = base + m;
+                    if (gi < a.B)
+                        a.dx_seq[((size_t)gi * a.T + t) * GA_WMAX + col] = ga_f32_to_bf16(accx[nt][r]);
+                }
+            }
+        __syncthreads();
+        if (!need_rec) break;
+    }
+}
+
+// ===========================================================================
+// lstm_dw: split-K weight-gradient kernel.
+//   dWcat[4H=256][K=128] = dgates^T x [x_seq | h_{t-1}],  K-split over B*T
+//   dbias[256]           = column sums of dgates        (same dgates pass)
+// rocBLAS is hopeless at this shape (C is ONE 256x64 tile -> 4 workgroups for
+// a K=B*T~2.4M reduction, measured 2.8 ms each); here NB blocks each reduce a
+// grid-strided set of 32-row chunks with MFMA (operands staged k-contiguous
+// in LDS) and write per-block partials [NB][256*128] summed by torch.sum —
+// one pass over dgates replaces TWO rocBLAS GEMMs + a bf16 bias reduce + the
+// hprev materialization (h_{t-1} is read with an in-kernel t-1 shift).
+// ===========================================================================
+#define DW_KC 32                // k-rows per staged chunk
+#define DW_STR 40               // LDS row stride (u16) — 80 B, 16B-aligned,
+                                // odd multiple of 16B so successive k-rows
+                                // spread over LDS banks
+
+struct LstmDwArgs {
+    const unsigned short* __restrict__ dgates;  // [BT][256]
+    const unsigned short* __restrict__ x_seq;   // [BT][64]
+    const unsigned short* __restrict__ h_seq;   // [BT][64] (read at t-1)
+    float* __restrict__ partial;                // [NB][256*128]
+    float* __restrict__ db_partial;             // [NB][256]
+    int BT, T, NB;
+};
+
+__launch_bounds__(256)
+__global__ void lstm_dw(LstmDwArgs a) {
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;  // wave = gate block (rows wid*64..wid*64+63)
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    unsigned short* dgT = (unsigned short*)smem;                 // [256][DW_STR]
+    unsigned short* icT = (unsigned short*)(smem + 256 * DW_STR * 2);  // [128][DW_STR]
+    float* s_db = (float*)(smem + (256 + 128) * DW_STR * 2);     // [8][256]
+
+    ga_f32x4 acc[4][8];  // [jt][ct] — C tile [64 rows of gate wid][128 cols]
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt)
+#pragma unroll
+        for (int ct = 0; ct < 8; ++ct) acc[jt][ct] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+    float db_reg[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) db_reg[e] = 0.f;
+
+    const int n_chunks = (a.BT + DW_KC - 1) / DW_KC;
+    for (int c = blockIdx.x; c < n_chunks; c += a.NB) {
+        const int r0 = c * DW_KC;
+        // ---- stage dgates chunk transposed (k-contiguous) + db partial ----
+        // granule g: row = g>>5 (0..31), col0 = (g&31)*8; thread's col0 fixed
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            int g = p * 256 + tid;
+            int row = g >> 5, col0 = (g & 31) * 8;
+            int r = r0 + row;
+            ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (r < a.BT) v = *(const ga_u16x8*)(a.dgates + (size_t)r * LSTM_4H + col0);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                dgT[(size_t)(col0 + e) * DW_STR + row] = v[e];
+                db_reg[e] += ga_bf16_to_f32(v[e]);
+            }
+        }
+        // ---- stage [x | h_{t-1}] transposed -------------------------------
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            int g = p * 256 + tid;
+            int row = g >> 4, col0 = (g & 15) * 8;
+            int r = r0 + row;
+            ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (r < a.BT) {
+                if (col0 < GA_WMAX) {
+                    v = *(const ga_u16x8*)(a.x_seq + (size_t)r * GA_WMAX + col0);
+                } else {
+                    int t = r % a.T;
+                    if (t > 0)
+                        v = *(const ga_u16x8*)(a.h_seq + (size_t)(r - 1) * GA_WMAX + col0 - GA_WMAX);
+                }
+            }
+#pragma unroll
+            for (int e = 0; e < 8; ++e) icT[(size_t)(col0 + e) * DW_STR + row] = v[e];
+        }
+        __syncthreads();
+
+        // ---- 32 MFMAs: C[64][128] += dgT_chunk x icT_chunk ----------------
+#pragma unroll
+        for (int jt = 0; jt < 4; ++jt) {
+            ga_bf16x8 afrag = *(const ga_bf16x8*)(dgT + (size_t)(wid * 64 + jt * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+#pragma unroll
+            for (int ct = 0; ct < 8; ++ct) {
+                ga_bf16x8 bfrag = *(const ga_bf16x8*)(icT + (size_t)(ct * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+                acc[jt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[jt][ct], 0, 0, 0);
             }
         }
         __syncthreads();
+    }
+
+    // ---- emit per-block partials ------------------------------------------
+    float* out = a.partial + (size_t)blockIdx.x * LSTM_4H * LSTM_K;
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt)
+#pragma unroll
+        for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int j = wid * 64 + jt * 16 + (lane >> 4) * 4 + r;
+                int col = ct * 16 + (lane & 15);
+                out[(size_t)j * LSTM_K + col] = acc[jt][ct][r];
+            }
+    // db: threads with the same (tid&31) cover the same 8 cols; 8 row-groups
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+        s_db[(size_t)(tid >> 5) * LSTM_4H + (tid & 31) * 8 + e] = db_reg[e];
+    __syncthreads();
+    if (tid < LSTM_4H) {
+        float s = 0.f;
+#pragma unroll
+        for (int gr = 0; gr < 8; ++gr) s += s_db[(size_t)gr * LSTM_4H + tid];
+        a.db_partial[(size_t)blockIdx.x * LSTM_4H + tid] = s;
     }
 }
 
@@ -283,21 +474,22 @@ void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
     a.T = (int)T;
     a.N = (int)X.size(0);
     int n_tiles = (int)((B + GA_MT - 1) / GA_MT);
-    size_t smem = 2 * GA_TILE_BYTES + LSTM_4H * sizeof(float);
+    size_t smem = 7 * GA_TILE_BYTES + LSTM_4H * sizeof(float);
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(lstm_fwd, dim3(n_tiles), dim3(256), smem, stream.stream(), a);
 }
 
 void lstm_bwd_cell_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
                           torch::Tensor c_seq, torch::Tensor wcat_t,
-                          torch::Tensor dgates_seq, int64_t B, int64_t T,
-                          bool dh_last_only) {
+                          torch::Tensor dgates_seq, c10::optional<torch::Tensor> dx_seq,
+                          int64_t B, int64_t T, bool dh_last_only) {
     LstmBwdArgs a;
     a.dh_out = (const unsigned short*)dh_out.data_ptr();
     a.gates_seq = (const unsigned short*)gates_seq.data_ptr();
     a.c_seq = (const unsigned short*)c_seq.data_ptr();
     a.wcat_t = (const unsigned short*)wcat_t.data_ptr();
     a.dgates_seq = (unsigned short*)dgates_seq.data_ptr();
+    a.dx_seq = dx_seq.has_value() ? (unsigned short*)dx_seq->data_ptr() : nullptr;
     a.B = (int)B;
     a.T = (int)T;
     a.dh_last_only = dh_last_only ? 1 : 0;
@@ -305,4 +497,22 @@ void lstm_bwd_cell_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
     size_t smem = 5 * GA_TILE_BYTES;
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(lstm_bwd_cell, dim3(n_tiles), dim3(256), smem, stream.stream(), a);
+}
+
+void lstm_dw_launch(torch::Tensor dgates, torch::Tensor x_seq, torch::Tensor h_seq,
+                    torch::Tensor partial, torch::Tensor db_partial,
+                    int64_t BT, int64_t T) {
+    LstmDwArgs a;
+    a.dgates = (const unsigned short*)dgates.data_ptr();
+    a.x_seq = (const unsigned short*)x_seq.data_ptr();
+    a.h_seq = (const unsigned short*)h_seq.data_ptr();
+    a.partial = partial.data_ptr<float>();
+    a.db_partial = db_partial.data_ptr<float>();
+    a.BT = (int)BT;
+    a.T = (int)T;
+    a.NB = (int)partial.size(0);
+    TORCH_CHECK(db_partial.size(0) == partial.size(0), "partial slab counts differ");
+    size_t smem = (size_t)(256 + 128) * DW_STR * 2 + 8 * LSTM_4H * sizeof(float);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(lstm_dw, dim3(a.NB), dim3(256), smem, stream.stream(), a);
 }

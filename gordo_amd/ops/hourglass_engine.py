@@ -120,7 +120,10 @@ class FusedHourglassTrainer:
         dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=device)
         db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=device)
         loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        counters = torch.tensor([0, 1], dtype=torch.int32, device=device)
+        # counters [batch idx, adam t]: hg_slab_reduce pre-bumps, so t starts 0
+        counters = torch.tensor([0, 0], dtype=torch.int32, device=device)
+        G = min(16, n_tiles)
+        partial = torch.zeros(G, p.P, dtype=torch.float32, device=device)
         m = torch.zeros_like(p.master)
         v = torch.zeros_like(p.master)
         perm = torch.zeros(n_batches * B, dtype=torch.int32, device=device)
@@ -141,11 +144,12 @@ class FusedHourglassTrainer:
                 Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
                 loss_acc, B, F_out, acts, gscale, bool(y_is_x), None,
             )
-            self.E.hg_adam(
-                p.master, m, v, dw, db, p.w_bf16, p.wt_bf16, p.bias, counters,
-                n_tiles, L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+            # two-stage slab reduce (counter bump folded into stage A)
+            self.E.hg_slab_reduce(dw, db, partial, counters, n_tiles, G, L, n_batches)
+            self.E.hg_adam_p(
+                p.master, m, v, partial, p.w_bf16, p.wt_bf16, p.bias, counters,
+                G, L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
             )
-            self.E.hg_bump(counters, n_batches)
 
         # ---- optional hipGraph capture of the whole step -----------------
         graph = None
@@ -160,7 +164,7 @@ class FusedHourglassTrainer:
                 torch.cuda.current_stream().wait_stream(s)
                 torch.cuda.synchronize()
                 # reset state disturbed by warm-up
-                counters.copy_(torch.tensor([0, 1], dtype=torch.int32, device=device))
+                counters.copy_(torch.tensor([0, 0], dtype=torch.int32, device=device))
                 loss_acc.zero_()
                 m.zero_(); v.zero_()
                 p2 = _PackedParams(self.module, device)
@@ -170,7 +174,7 @@ class FusedHourglassTrainer:
                 with torch.cuda.graph(graph):
                     one_step()
                 # capture replays with pristine state below
-                counters.copy_(torch.tensor([0, 1], dtype=torch.int32, device=device))
+                counters.copy_(torch.tensor([0, 0], dtype=torch.int32, device=device))
                 loss_acc.zero_(); m.zero_(); v.zero_()
                 p.master.copy_(p2.master); p.bias.copy_(p2.bias)
                 p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)

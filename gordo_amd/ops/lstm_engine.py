@@ -11,9 +11,12 @@ model/utils.py::create_keras_timeseriesgenerator). Here each training step is:
                                       [B,T,F] batch in HBM traffic terms)
   head fwd+bwd+loss:  hg_train_step — 1-layer dense head on h_{T-1}, MSE,
                                       dx_out = dL/dh_{T-1}
-  per layer l (rev):  lstm_bwd_cell — reverse-time BPTT (dgates), then the
-                      non-sequential bulk GEMMs (dW = in^T x dgates over
-                      K = B*T, dx_seq = dgates x W_ih) via rocBLAS bf16
+  per layer l (rev):  lstm_bwd_cell — reverse-time BPTT (dgates) with the
+                      dx_seq = dgates x W_ih GEMM fused into the same pass,
+                      then lstm_dw — split-K MFMA kernel producing
+                      dWcat = dgates^T x [x | h_{t-1}] AND dbias from ONE
+                      pass over dgates (rocBLAS is ~20x slower at this
+                      single-tile, K=B*T shape)
   optimizer:          fp32-master Adam (torch foreach) + bf16 weight re-emit
 
 Weight layout: per layer one packed [4H=256][128] bf16 "wcat" (cols 0..63 =
@@ -210,7 +213,11 @@ class FusedLSTMTrainer:
             "x_seq0": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
             "dgates": torch.zeros(B, T, H4, dtype=bf16, device=device),
             "dh_seq": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
-            "hprev": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
+            # split-K dW partials: NB blocks sized to fill 256 CUs
+            "dw_part": torch.zeros(256, H4 * KCAT, dtype=torch.float32, device=device),
+            "db_part": torch.zeros(256, H4, dtype=torch.float32, device=device),
+            "gw": [torch.zeros(H4 * KCAT, dtype=torch.float32, device=device) for _ in range(Lc)],
+            "gb": [torch.zeros(H4, dtype=torch.float32, device=device) for _ in range(Lc)],
             "h_last": torch.zeros(B, WMAX, dtype=bf16, device=device),
             "y_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
             "dx_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
@@ -268,26 +275,26 @@ class FusedLSTMTrainer:
             st["dx_head"],
         )
 
-        # ---- BPTT + bulk grad GEMMs, top layer down ----------------------
-        dg2 = st["dgates"].view(-1, H4)
+        # ---- BPTT (fused dx) + split-K dW kernel, top layer down ---------
+        # dh_seq doubles as the dx output: within a block, dh_out[gi][t] is
+        # read (elementwise phase) before dx_seq[gi][t] is written (GEMM
+        # phase of the same t), and blocks own disjoint gi rows.
         for l in reversed(range(p.Lc)):
             dh = st["dx_head"] if l == p.Lc - 1 else st["dh_seq"]
             E.lstm_bwd_cell(
                 dh, st["gates_seq"][l], st["c_seq"][l], p.wcat_t_bf16[l],
-                st["dgates"], B, T, l == p.Lc - 1,
+                st["dgates"], st["dh_seq"] if l > 0 else None, B, T,
+                l == p.Lc - 1,
             )
             x_seq = st["x_seq0"] if l == 0 else st["h_seq"][l - 1]
-            st["hprev"][:, 0].zero_()
-            st["hprev"][:, 1:].copy_(st["h_seq"][l][:, : T - 1])
-            # dWcat = dgates^T x [x_seq | hprev]  (K = B*T, rocBLAS bf16 TN)
-            gw = torch.zeros_like(p.wcat_master[l])
-            gw[:, :WMAX] = torch.matmul(dg2.t(), x_seq.view(-1, WMAX)).float()
-            gw[:, WMAX:] = torch.matmul(dg2.t(), st["hprev"].view(-1, WMAX)).float()
-            p.wcat_master[l].grad = gw
-            p.bias_master[l].grad = dg2.sum(0, dtype=torch.float32)
-            if l > 0:
-                # dx for the layer below: dgates x W_ih  [B*T, 64]
-                torch.matmul(dg2, p.wcat_bf16[l][:, :WMAX], out=st["dh_seq"].view(-1, WMAX))
+            E.lstm_dw(
+                st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
+                B * T, T,
+            )
+            torch.sum(st["dw_part"], dim=0, out=st["gw"][l])
+            torch.sum(st["db_part"], dim=0, out=st["gb"][l])
+            p.wcat_master[l].grad = st["gw"][l].view(H4, KCAT)
+            p.bias_master[l].grad = st["gb"][l]
 
         # ---- head grads from the hg slabs (slab layout is [in][out]) -----
         p.head_wm.grad = st["head_dw"].sum(0).view(WMAX, WMAX).t().contiguous()

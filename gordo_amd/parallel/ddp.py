@@ -122,7 +122,10 @@ class HourglassStepTrainer:
         self.db = torch.zeros(self.n_tiles, p.L, WMAX, dtype=torch.float32, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        self.counters = torch.tensor([0, 1], dtype=torch.int32, device=device)
+        # counters [batch idx, adam t]: hg_slab_reduce pre-bumps, so t starts 0
+        self.counters = torch.tensor([0, 0], dtype=torch.int32, device=device)
+        self.G = min(16, self.n_tiles)
+        self.partial = torch.zeros(self.G, p.P, dtype=torch.float32, device=device)
         self.m = torch.zeros_like(p.master)
         self.v = torch.zeros_like(p.master)
         self.n_batches = max(1, math.ceil(N / self.B))
@@ -192,20 +195,24 @@ class HourglassStepTrainer:
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
             self.gscale, True, None,
         )
+        # two-stage slab reduce; stage A also bumps the counters
+        self.E.hg_slab_reduce(
+            self.dw, self.db, self.partial, self.counters, self.n_tiles,
+            self.G, p.L, self.n_batches,
+        )
         if self.dp.enabled:
-            self.E.hg_grad_reduce(self.dw, self.db, self.gflat, self.n_tiles, p.L)
+            self.E.hg_partial_flat(self.partial, self.gflat, self.G, p.L)
             self.dp.all_reduce_mean(self.gflat)
             self.E.hg_adam_flat(
                 p.master, self.m, self.v, self.gflat, p.w_bf16, p.wt_bf16, p.bias,
                 self.counters, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
             )
         else:
-            self.E.hg_adam(
-                p.master, self.m, self.v, self.dw, self.db, p.w_bf16, p.wt_bf16,
-                p.bias, self.counters, self.n_tiles, p.L,
+            self.E.hg_adam_p(
+                p.master, self.m, self.v, self.partial, p.w_bf16, p.wt_bf16,
+                p.bias, self.counters, self.G, p.L,
                 cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
             )
-        self.E.hg_bump(self.counters, self.n_batches)
 
     # ------------------------------------------------------------------
     def _init_eager(self):
