@@ -68,12 +68,13 @@ def main():
         from gordo_amd.model.factories.lstm_autoencoder import lstm_hourglass
         from gordo_amd.ops.lstm_engine import LSTMStepTrainer
 
-        batch = args.batch or 16384
+        # 32768 windows/GPU -> 512 blocks = 2 blocks/CU in the cell kernels
+        batch = args.batch or 32768
         module = lstm_hourglass(
             args.tags, args.tags, lookback_window=args.lookback,
             encoding_layers=2, compression_factor=0.5,
         )
-        rows = max(DATASET_ROWS // 4, batch + args.lookback + 1)
+        rows = max(DATASET_ROWS // 2, batch + args.lookback + 1)
         X = synthetic_sensor_data(rows, args.tags, seed=1000 + rank)
         cfg = TrainConfig(batch_size=batch, lr=1e-3, dtype="bf16")
         trainer = LSTMStepTrainer(module, X, batch, cfg, dp=dp)

@@ -45,7 +45,7 @@ extern __shared__ __attribute__((aligned(16))) char ga_smem_lstm[];
 // The cell epilogue lands MFMA-layout results in LDS tiles; a vectorized
 // writer phase then streams them to HBM as 16-B stores (the original direct
 // per-element u16 stores — 96 scalar stores/lane/t — were the fwd bottleneck).
-__launch_bounds__(256)
+__launch_bounds__(256, 2)
 __global__ void lstm_fwd(LstmFwdArgs a) {
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -101,45 +101,56 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
         }
         __syncthreads();
 
-        // ---- gates GEMM: [64][256] = cat(x,h)[64][128] x Wcat^T ----------
-        ga_f32x4 acc[LSTM_NT];
-#pragma unroll
-        for (int nt = 0; nt < LSTM_NT; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
-        for (int kk = 0; kk < LSTM_K; kk += 32) {
-            const char* src = (kk < 64) ? x_tile : h_tile;
-            int klocal = kk & 63;
-            ga_bf16x8 afrag = ga_tile_read8(src, wrow + (lane & 15), klocal + (lane >> 4) * 8);
-#pragma unroll
-            for (int nt = 0; nt < LSTM_NT; ++nt) {
-                // B[k][j] = Wcat[j][k] row-read (16B, L2-resident)
-                ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat + (size_t)(nt * 16 + (lane & 15)) * LSTM_K + kk + (lane >> 4) * 8);
-                acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
-            }
-        }
-        __syncthreads();  // x_tile/h_tile reads done; epilogue rewrites h_tile
-
-        // ---- cell update -> LDS tiles (gates i,f,g,o of one (m,j) live in
-        //      acc[jt], acc[jt+4], acc[jt+8], acc[jt+12]) -------------------
+        unsigned short h_stage[4][4];  // deferred h_tile update (see below)
+        // ---- gates GEMM + cell update, one 16-col tile at a time ---------
+        // (per-jt: only the 4 gate acc tiles are live -> 16 f32/lane, which
+        // keeps the kernel under the 128-VGPR budget for 2 waves/SIMD; the
+        // x/h afrag re-reads per jt are cheap LDS hits)
 #pragma unroll
         for (int jt = 0; jt < 4; ++jt) {
             int col = jt * 16 + (lane & 15);
+            ga_f32x4 acc4[4];  // i, f, g, o pre-activations for col tile jt
+#pragma unroll
+            for (int g = 0; g < 4; ++g) acc4[g] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            for (int kk = 0; kk < LSTM_K; kk += 32) {
+                const char* src = (kk < 64) ? x_tile : h_tile;
+                int klocal = kk & 63;
+                ga_bf16x8 afrag = ga_tile_read8(src, wrow + (lane & 15), klocal + (lane >> 4) * 8);
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+                    // B[k][j] = Wcat[j][k] row-read (16B, L2-resident)
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat + (size_t)(g * GA_WMAX + jt * 16 + (lane & 15)) * LSTM_K + kk + (lane >> 4) * 8);
+                    acc4[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc4[g], 0, 0, 0);
+                }
+            }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int m = wrow + (lane >> 4) * 4 + r;
-                float i_g = 1.f / (1.f + __expf(-(acc[jt][r] + s_bias[col])));
-                float f_g = 1.f / (1.f + __expf(-(acc[jt + 4][r] + s_bias[GA_WMAX + col])));
-                float g_g = tanhf(acc[jt + 8][r] + s_bias[2 * GA_WMAX + col]);
-                float o_g = 1.f / (1.f + __expf(-(acc[jt + 12][r] + s_bias[3 * GA_WMAX + col])));
+                float i_g = 1.f / (1.f + __expf(-(acc4[0][r] + s_bias[col])));
+                float f_g = 1.f / (1.f + __expf(-(acc4[1][r] + s_bias[GA_WMAX + col])));
+                float g_g = ga_tanh(acc4[2][r] + s_bias[2 * GA_WMAX + col]);
+                float o_g = 1.f / (1.f + __expf(-(acc4[3][r] + s_bias[3 * GA_WMAX + col])));
                 float c_new = f_g * c_reg[jt][r] + i_g * g_g;
                 c_reg[jt][r] = c_new;
-                float h_new = o_g * tanhf(c_new);
-                ga_tile_write1(h_tile, m, col, ga_f32_to_bf16(h_new));
+                float h_new = o_g * ga_tanh(c_new);
+                // NOTE: h_tile writes must not race the GEMM's h reads of
+                // OTHER waves -> write to c/g tiles now, h to a temp column
+                // in c-order; h_tile is updated after the barrier below.
                 ga_tile_write1(c_tile, m, col, ga_f32_to_bf16(c_new));
                 ga_tile_write1(g_tiles, m, col, ga_f32_to_bf16(i_g));
                 ga_tile_write1(g_tiles + GA_TILE_BYTES, m, col, ga_f32_to_bf16(f_g));
                 ga_tile_write1(g_tiles + 2 * GA_TILE_BYTES, m, col, ga_f32_to_bf16(g_g));
                 ga_tile_write1(g_tiles + 3 * GA_TILE_BYTES, m, col, ga_f32_to_bf16(o_g));
+                h_stage[jt][r] = ga_f32_to_bf16(h_new);
             }
+        }
+        __syncthreads();  // all waves done reading x/h for this t
+#pragma unroll
+        for (int jt = 0; jt < 4; ++jt) {
+            int col = jt * 16 + (lane & 15);
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                ga_tile_write1(h_tile, wrow + (lane >> 4) * 4 + r, col, h_stage[jt][r]);
         }
         __syncthreads();
 
@@ -182,7 +193,7 @@ struct LstmBwdArgs {
     int dh_last_only;  // 1: dh_out is [B][64] at the final step only
 };
 
-__launch_bounds__(256)
+__launch_bounds__(256, 2)
 __global__ void lstm_bwd_cell(LstmBwdArgs a) {
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -213,60 +224,56 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
             const int gi = base + em;
             const size_t bt = (size_t)gi * a.T + t;
             const bool live = gi < a.B;
-            ga_u16x8 vi[2], vf[2], vg[2], vo[2], vc[2], vcp[2], vdh[2];
-            ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+            const ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+            // one 8-col half at a time: keeps only ~7 input vectors + 4
+            // output vectors live (fits the 2-waves/SIMD VGPR budget)
 #pragma unroll
             for (int hh = 0; hh < 2; ++hh) {
                 int cc = c0 + hh * 8;
+                ga_u16x8 vi = z, vf = z, vg = z, vo = z, vc = z, vcp = z, vdh = z;
                 if (live) {
-                    vi[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc);
-                    vf[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
-                    vg[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
-                    vo[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
-                    vc[hh] = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
-                    vcp[hh] = (t > 0) ? *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc) : z;
-                    if (a.dh_last_only)
-                        vdh[hh] = (t == a.T - 1)
-                                      ? *(const ga_u16x8*)(a.dh_out + (size_t)gi * GA_WMAX + cc)
-                                      : z;
-                    else
-                        vdh[hh] = *(const ga_u16x8*)(a.dh_out + bt * GA_WMAX + cc);
-                } else {
-                    vi[hh] = vf[hh] = vg[hh] = vo[hh] = vc[hh] = vcp[hh] = vdh[hh] = z;
+                    vi = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc);
+                    vf = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
+                    vg = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
+                    vo = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
+                    vc = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
+                    if (t > 0) vcp = *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc);
+                    if (a.dh_last_only) {
+                        if (t == a.T - 1)
+                            vdh = *(const ga_u16x8*)(a.dh_out + (size_t)gi * GA_WMAX + cc);
+                    } else {
+                        vdh = *(const ga_u16x8*)(a.dh_out + bt * GA_WMAX + cc);
+                    }
                 }
-            }
-            ga_u16x8 odi[2], odf[2], odg[2], odo[2];
-#pragma unroll
-            for (int hh = 0; hh < 2; ++hh) {
-                ga_u16x8 vrec = *(ga_u16x8*)&dh_tile[ga_sw(em, c0 + hh * 8)];
+                ga_u16x8 vrec = *(ga_u16x8*)&dh_tile[ga_sw(em, cc)];
+                ga_u16x8 odi, odf, odg, odo;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
-                    float dh = ga_bf16_to_f32(vdh[hh][e]) + ga_bf16_to_f32(vrec[e]);
-                    float i_g = ga_bf16_to_f32(vi[hh][e]);
-                    float f_g = ga_bf16_to_f32(vf[hh][e]);
-                    float g_g = ga_bf16_to_f32(vg[hh][e]);
-                    float o_g = ga_bf16_to_f32(vo[hh][e]);
-                    float c_t = ga_bf16_to_f32(vc[hh][e]);
-                    float c_prev = ga_bf16_to_f32(vcp[hh][e]);
-                    float tanhc = tanhf(c_t);
+                    float dh = ga_bf16_to_f32(vdh[e]) + ga_bf16_to_f32(vrec[e]);
+                    float i_g = ga_bf16_to_f32(vi[e]);
+                    float f_g = ga_bf16_to_f32(vf[e]);
+                    float g_g = ga_bf16_to_f32(vg[e]);
+                    float o_g = ga_bf16_to_f32(vo[e]);
+                    float c_t = ga_bf16_to_f32(vc[e]);
+                    float c_prev = ga_bf16_to_f32(vcp[e]);
+                    float tanhc = ga_tanh(c_t);
                     int ei = hh * 8 + e;
                     float dc = dc_reg[ei] + dh * o_g * (1.f - tanhc * tanhc);
-                    odo[hh][e] = ga_f32_to_bf16(dh * tanhc * o_g * (1.f - o_g));
-                    odi[hh][e] = ga_f32_to_bf16(dc * g_g * i_g * (1.f - i_g));
-                    odf[hh][e] = ga_f32_to_bf16(dc * c_prev * f_g * (1.f - f_g));
-                    odg[hh][e] = ga_f32_to_bf16(dc * i_g * (1.f - g_g * g_g));
+                    odo[e] = ga_f32_to_bf16(dh * tanhc * o_g * (1.f - o_g));
+                    odi[e] = ga_f32_to_bf16(dc * g_g * i_g * (1.f - i_g));
+                    odf[e] = ga_f32_to_bf16(dc * c_prev * f_g * (1.f - f_g));
+                    odg[e] = ga_f32_to_bf16(dc * i_g * (1.f - g_g * g_g));
                     dc_reg[ei] = dc * f_g;  // carry to t-1
                 }
-                int cc = c0 + hh * 8;
-                *(ga_u16x8*)&dg_tiles[ga_sw(em, cc)] = odi[hh];
-                *(ga_u16x8*)&dg_tiles[GA_TILE_BYTES + ga_sw(em, cc)] = odf[hh];
-                *(ga_u16x8*)&dg_tiles[2 * GA_TILE_BYTES + ga_sw(em, cc)] = odg[hh];
-                *(ga_u16x8*)&dg_tiles[3 * GA_TILE_BYTES + ga_sw(em, cc)] = odo[hh];
+                *(ga_u16x8*)&dg_tiles[ga_sw(em, cc)] = odi;
+                *(ga_u16x8*)&dg_tiles[GA_TILE_BYTES + ga_sw(em, cc)] = odf;
+                *(ga_u16x8*)&dg_tiles[2 * GA_TILE_BYTES + ga_sw(em, cc)] = odg;
+                *(ga_u16x8*)&dg_tiles[3 * GA_TILE_BYTES + ga_sw(em, cc)] = odo;
                 if (live) {
-                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + cc) = odi[hh];
-                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + GA_WMAX + cc) = odf[hh];
-                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc) = odg[hh];
-                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc) = odo[hh];
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + cc) = odi;
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + GA_WMAX + cc) = odf;
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc) = odg;
+                    *(ga_u16x8*)(a.dgates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc) = odo;
                 }
             }
         }

@@ -213,9 +213,9 @@ class FusedLSTMTrainer:
             "x_seq0": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
             "dgates": torch.zeros(B, T, H4, dtype=bf16, device=device),
             "dh_seq": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
-            # split-K dW partials: NB blocks sized to fill 256 CUs
-            "dw_part": torch.zeros(256, H4 * KCAT, dtype=torch.float32, device=device),
-            "db_part": torch.zeros(256, H4, dtype=torch.float32, device=device),
+            # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
+            "dw_part": torch.zeros(512, H4 * KCAT, dtype=torch.float32, device=device),
+            "db_part": torch.zeros(512, H4, dtype=torch.float32, device=device),
             "gw": [torch.zeros(H4 * KCAT, dtype=torch.float32, device=device) for _ in range(Lc)],
             "gb": [torch.zeros(H4, dtype=torch.float32, device=device) for _ in range(Lc)],
             "h_last": torch.zeros(B, WMAX, dtype=bf16, device=device),
