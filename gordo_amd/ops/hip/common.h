@@ -19,14 +19,14 @@ __device__ __forceinline__ float ga_bf16_to_f32(unsigned short u) {
     return c.f;
 }
 
+// single v_cvt_pk_bf16_f32 (HW round-to-nearest-even) — the manual shift/add
+// RTNE emulation costs ~6 VALU ops per convert and showed up as the top VALU
+// consumer in the LSTM cell kernels (≈100 converts/thread/timestep).
 __device__ __forceinline__ unsigned short ga_f32_to_bf16(float f) {
-    union { float f; unsigned int i; } c;
-    c.f = f;
-    // round-to-nearest-even
-    unsigned int lsb = (c.i >> 16) & 1u;
-    unsigned int rounded = c.i + 0x7FFFu + lsb;
-    if ((c.i & 0x7F800000u) == 0x7F800000u) rounded = c.i;  // inf/nan passthrough
-    return (unsigned short)(rounded >> 16);
+    __bf16 h = (__bf16)f;
+    union { __bf16 h; unsigned short u; } c;
+    c.h = h;
+    return c.u;
 }
 
 // ---------------------------------------------------------------------------

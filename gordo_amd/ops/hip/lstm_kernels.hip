@@ -99,6 +99,24 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
             }
             ga_tile_write8(x_tile, m, col0, v);
         }
+        // ---- deferred writer for t-1: 16-B stores of h, c, gates ---------
+        // (overlapped with the x stage: both only read tiles the other
+        // doesn't touch; the barrier below fences phase2's c/g overwrite)
+        if (t > 0 && wgi < a.B) {
+            size_t bt = (size_t)wgi * a.T + (t - 1);
+#pragma unroll
+            for (int half = 0; half < 2; ++half) {
+                int c0 = wc0 + half * 8;
+                *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + c0) =
+                    *(ga_u16x8*)&h_tile[ga_sw(wm, c0)];
+                *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + c0) =
+                    *(ga_u16x8*)&c_tile[ga_sw(wm, c0)];
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + g * GA_WMAX + c0) =
+                        *(ga_u16x8*)&g_tiles[g * GA_TILE_BYTES + ga_sw(wm, c0)];
+            }
+        }
         __syncthreads();
 
         unsigned short h_stage[4][4];  // deferred h_tile update (see below)
@@ -153,24 +171,23 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
                 ga_tile_write1(h_tile, wrow + (lane >> 4) * 4 + r, col, h_stage[jt][r]);
         }
         __syncthreads();
+    }
 
-        // ---- vectorized writer: 16-B stores of h, c, gates ---------------
-        if (wgi < a.B) {
-            size_t bt = (size_t)wgi * a.T + t;
+    // trailing writer for t = T-1
+    if (wgi < a.B) {
+        size_t bt = (size_t)wgi * a.T + (a.T - 1);
 #pragma unroll
-            for (int half = 0; half < 2; ++half) {
-                int c0 = wc0 + half * 8;
-                *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + c0) =
-                    *(ga_u16x8*)&((char*)h_tile)[ga_sw(wm, c0)];
-                *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + c0) =
-                    *(ga_u16x8*)&((char*)c_tile)[ga_sw(wm, c0)];
+        for (int half = 0; half < 2; ++half) {
+            int c0 = wc0 + half * 8;
+            *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + c0) =
+                *(ga_u16x8*)&h_tile[ga_sw(wm, c0)];
+            *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + c0) =
+                *(ga_u16x8*)&c_tile[ga_sw(wm, c0)];
 #pragma unroll
-                for (int g = 0; g < 4; ++g)
-                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + g * GA_WMAX + c0) =
-                        *(ga_u16x8*)&((char*)g_tiles)[g * GA_TILE_BYTES + ga_sw(wm, c0)];
-            }
+            for (int g = 0; g < 4; ++g)
+                *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + g * GA_WMAX + c0) =
+                    *(ga_u16x8*)&g_tiles[g * GA_TILE_BYTES + ga_sw(wm, c0)];
         }
-        __syncthreads();  // writer reads done before next epilogue overwrites
     }
 }
 
@@ -308,7 +325,8 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
                     accx[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, accx[nt], 0, 0, 0);
                 }
         }
-        __syncthreads();  // dg_tiles reads done before next iteration rewrites
+        // (no barrier needed here: nothing below touches dg_tiles, and the
+        // end-of-iteration barrier fences the next elementwise rewrite)
         if (need_rec)
 #pragma unroll
             for (int nt = 0; nt < 4; ++nt) {
@@ -368,9 +386,14 @@ __global__ void lstm_dw(LstmDwArgs a) {
     const int wid = tid >> 6;  // wave = gate block (rows wid*64..wid*64+63)
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    unsigned short* dgT = (unsigned short*)smem;                 // [256][DW_STR]
-    unsigned short* icT = (unsigned short*)(smem + 256 * DW_STR * 2);  // [128][DW_STR]
-    float* s_db = (float*)(smem + (256 + 128) * DW_STR * 2);     // [8][256]
+    // double-buffered: stage chunk c+1 into buf[~c&1] WHILE MFMAing buf[c&1]
+    // — one barrier per chunk and the staging global-load latency hides
+    // behind the MFMA phase instead of being exposed serially.
+#define DW_BUF_U16 ((256 + 128) * DW_STR)
+    unsigned short* dgT[2] = {(unsigned short*)smem,
+                              (unsigned short*)smem + DW_BUF_U16};
+    unsigned short* icT[2] = {dgT[0] + 256 * DW_STR, dgT[1] + 256 * DW_STR};
+    float* s_db = (float*)((unsigned short*)smem + 2 * DW_BUF_U16);  // [8][256]
 
     ga_f32x4 acc[4][8];  // [jt][ct] — C tile [64 rows of gate wid][128 cols]
 #pragma unroll
@@ -381,55 +404,63 @@ __global__ void lstm_dw(LstmDwArgs a) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) db_reg[e] = 0.f;
 
+    // thread-fixed staging coordinates
+    const int dg_row0 = tid >> 5, dg_col0 = (tid & 31) * 8;   // +8 rows per pass
+    const int ic_row0 = tid >> 4, ic_col0 = (tid & 15) * 8;   // +16 rows per pass
+
     const int n_chunks = (a.BT + DW_KC - 1) / DW_KC;
-    for (int c = blockIdx.x; c < n_chunks; c += a.NB) {
+
+    auto stage = [&](int c, int buf) {
         const int r0 = c * DW_KC;
-        // ---- stage dgates chunk transposed (k-contiguous) + db partial ----
-        // granule g: row = g>>5 (0..31), col0 = (g&31)*8; thread's col0 fixed
+        // dgates chunk transposed (k-contiguous) + db partial accumulation
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
-            int g = p * 256 + tid;
-            int row = g >> 5, col0 = (g & 31) * 8;
+            int row = p * 8 + dg_row0;
             int r = r0 + row;
             ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (r < a.BT) v = *(const ga_u16x8*)(a.dgates + (size_t)r * LSTM_4H + col0);
+            if (r < a.BT) v = *(const ga_u16x8*)(a.dgates + (size_t)r * LSTM_4H + dg_col0);
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
-                dgT[(size_t)(col0 + e) * DW_STR + row] = v[e];
+                dgT[buf][(size_t)(dg_col0 + e) * DW_STR + row] = v[e];
                 db_reg[e] += ga_bf16_to_f32(v[e]);
             }
         }
-        // ---- stage [x | h_{t-1}] transposed -------------------------------
+        // [x | h_{t-1}] transposed
 #pragma unroll
         for (int p = 0; p < 2; ++p) {
-            int g = p * 256 + tid;
-            int row = g >> 4, col0 = (g & 15) * 8;
+            int row = p * 16 + ic_row0;
             int r = r0 + row;
             ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
             if (r < a.BT) {
-                if (col0 < GA_WMAX) {
-                    v = *(const ga_u16x8*)(a.x_seq + (size_t)r * GA_WMAX + col0);
+                if (ic_col0 < GA_WMAX) {
+                    v = *(const ga_u16x8*)(a.x_seq + (size_t)r * GA_WMAX + ic_col0);
                 } else {
                     int t = r % a.T;
                     if (t > 0)
-                        v = *(const ga_u16x8*)(a.h_seq + (size_t)(r - 1) * GA_WMAX + col0 - GA_WMAX);
+                        v = *(const ga_u16x8*)(a.h_seq + (size_t)(r - 1) * GA_WMAX + ic_col0 - GA_WMAX);
                 }
             }
 #pragma unroll
-            for (int e = 0; e < 8; ++e) icT[(size_t)(col0 + e) * DW_STR + row] = v[e];
+            for (int e = 0; e < 8; ++e) icT[buf][(size_t)(ic_col0 + e) * DW_STR + row] = v[e];
         }
-        __syncthreads();
+    };
 
+    int nb = 0;
+    stage(blockIdx.x, 0);
+    __syncthreads();
+    for (int c = blockIdx.x; c < n_chunks; c += a.NB) {
+        if (c + a.NB < n_chunks) stage(c + a.NB, nb ^ 1);
         // ---- 32 MFMAs: C[64][128] += dgT_chunk x icT_chunk ----------------
 #pragma unroll
         for (int jt = 0; jt < 4; ++jt) {
-            ga_bf16x8 afrag = *(const ga_bf16x8*)(dgT + (size_t)(wid * 64 + jt * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+            ga_bf16x8 afrag = *(const ga_bf16x8*)(dgT[nb] + (size_t)(wid * 64 + jt * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
 #pragma unroll
             for (int ct = 0; ct < 8; ++ct) {
-                ga_bf16x8 bfrag = *(const ga_bf16x8*)(icT + (size_t)(ct * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+                ga_bf16x8 bfrag = *(const ga_bf16x8*)(icT[nb] + (size_t)(ct * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
                 acc[jt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[jt][ct], 0, 0, 0);
             }
         }
+        nb ^= 1;
         __syncthreads();
     }
 
@@ -519,7 +550,7 @@ void lstm_dw_launch(torch::Tensor dgates, torch::Tensor x_seq, torch::Tensor h_s
     a.T = (int)T;
     a.NB = (int)partial.size(0);
     TORCH_CHECK(db_partial.size(0) == partial.size(0), "partial slab counts differ");
-    size_t smem = (size_t)(256 + 128) * DW_STR * 2 + 8 * LSTM_4H * sizeof(float);
+    size_t smem = (size_t)2 * DW_BUF_U16 * 2 + 8 * LSTM_4H * sizeof(float);
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(lstm_dw, dim3(a.NB), dim3(256), smem, stream.stream(), a);
 }
