@@ -61,6 +61,19 @@ __device__ __forceinline__ void ga_tile_write1(char* tile, int row, int col, uns
 }
 
 // ---------------------------------------------------------------------------
+// Swizzled LDS image for weight matrices read as MFMA B-fragments: 16-lane
+// groups read 16 consecutive ROWS at one column offset; a plain row-major
+// image puts all 16 on one bank (row stride is a multiple of 256 B). XORing
+// bits 4..7 of the in-row byte offset with (row&15) permutes the 16-B slots
+// so the 16 rows land on 16 distinct slots. rowbytes must be a multiple of
+// 256 and the in-row offset a multiple of 16.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ unsigned int ga_swW(int row, int colbyte, int rowbytes) {
+    return (unsigned int)row * (unsigned int)rowbytes +
+           ((unsigned int)colbyte ^ (((unsigned int)row & 15u) << 4));
+}
+
+// ---------------------------------------------------------------------------
 // Activations (ids shared with the Python engine)
 // ---------------------------------------------------------------------------
 #define GA_ACT_LINEAR 0
@@ -68,19 +81,28 @@ __device__ __forceinline__ void ga_tile_write1(char* tile, int row, int col, uns
 #define GA_ACT_RELU 2
 #define GA_ACT_SIGMOID 3
 
+// v_rcp_f32 (1 ULP) — plain `1.f/x` and even __fdividef lower to the IEEE
+// div_scale/div_fmas/div_fixup sequence (~8 instrs, serially dependent),
+// which was the top instruction cluster in the LSTM cell epilogue.
+__device__ __forceinline__ float ga_rcp(float x) { return __builtin_amdgcn_rcpf(x); }
+
 // fast tanh via the hardware-exp path (libm tanhf is an ocml call, ~10x the
 // cost; rel err ~1e-7, far below bf16 resolution). Clamp keeps exp finite.
 __device__ __forceinline__ float ga_tanh(float x) {
     float xc = fminf(fmaxf(x, -15.f), 15.f);
     float e = __expf(2.f * xc);
-    return __fdividef(e - 1.f, e + 1.f);
+    return (e - 1.f) * ga_rcp(e + 1.f);
+}
+
+__device__ __forceinline__ float ga_sigmoid(float x) {
+    return ga_rcp(1.f + __expf(-x));
 }
 
 __device__ __forceinline__ float ga_act(float x, int id) {
     switch (id) {
         case GA_ACT_TANH: return ga_tanh(x);
         case GA_ACT_RELU: return x > 0.f ? x : 0.f;
-        case GA_ACT_SIGMOID: return 1.f / (1.f + __expf(-x));
+        case GA_ACT_SIGMOID: return ga_sigmoid(x);
         default: return x;
     }
 }

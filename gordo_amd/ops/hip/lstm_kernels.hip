@@ -41,11 +41,11 @@ struct LstmFwdArgs {
 
 extern __shared__ __attribute__((aligned(16))) char ga_smem_lstm[];
 
-// LDS plan (57.25 KiB -> 2 blocks/CU): x, h, c + 4 gate tiles + bias.
-// The cell epilogue lands MFMA-layout results in LDS tiles; a vectorized
-// writer phase then streams them to HBM as 16-B stores (the original direct
-// per-element u16 stores — 96 scalar stores/lane/t — were the fwd bottleneck).
-__launch_bounds__(256, 2)
+// LDS plan (~121.25 KiB -> 1 block/CU): x, h, c + 4 gate tiles + bias + the
+// whole 64-KiB Wcat (staged ONCE, swizzled; ablation showed the per-t L2
+// B-fragment loads were 74% of the kernel). The per-t x tile is prefetched
+// into registers during the previous GEMM so its HBM latency is hidden.
+__launch_bounds__(256)
 __global__ void lstm_fwd(LstmFwdArgs a) {
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -58,8 +58,16 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
     char* c_tile = ga_smem_lstm + 2 * GA_TILE_BYTES; // [64][64]
     char* g_tiles = ga_smem_lstm + 3 * GA_TILE_BYTES;  // 4 x [64][64] i,f,g,o
     float* s_bias = (float*)(ga_smem_lstm + 7 * GA_TILE_BYTES);  // [256]
+    char* w_lds = ga_smem_lstm + 7 * GA_TILE_BYTES + 1024;  // [256][128] swizzled
 
     for (int i = tid; i < LSTM_4H; i += 256) s_bias[i] = a.bias[i];
+
+    // stage the whole Wcat into LDS once (re-read 4*T times by the GEMM)
+    for (int g = tid; g < LSTM_4H * 16; g += 256) {
+        int row = g >> 4, col0 = (g & 15) * 8;
+        ga_u16x8 v = *(const ga_u16x8*)(a.wcat + (size_t)row * LSTM_K + col0);
+        *(ga_u16x8*)&w_lds[ga_swW(row, col0 * 2, 256)] = v;
+    }
 
     // zero h_tile (h_{-1} = 0)
     for (int c = tid; c < GA_MT * 8; c += 256) {
@@ -79,25 +87,38 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
     const int wc0 = (tid & 3) * 16;
     const int wgi = base + wm;
 
-    __syncthreads();
-
-    for (int t = 0; t < a.T; ++t) {
-        // ---- stage x_t tile ------------------------------------------------
-        for (int c = tid; c < GA_MT * 8; c += 256) {
-            int m = c >> 3, col0 = (c & 7) * 8;
-            int gi = base + m;
+    // x prefetch: thread owns two fixed (row, col) granules of the x tile
+    const int px_m[2] = {tid >> 3, (tid + 256) >> 3};
+    const int px_c[2] = {(tid & 7) * 8, (tid & 7) * 8};
+    ga_u16x8 px[2];
+    auto px_load = [&](int t) {
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            int gi = base + px_m[i];
             ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
             if (gi < a.B) {
                 if (a.win != nullptr) {
                     int row = a.win[gi] + t;
-                    v = *(const ga_u16x8*)(a.X + (size_t)row * GA_WMAX + col0);
+                    v = *(const ga_u16x8*)(a.X + (size_t)row * GA_WMAX + px_c[i]);
                 } else {
-                    v = *(const ga_u16x8*)(a.X + ((size_t)gi * a.T + t) * GA_WMAX + col0);
+                    v = *(const ga_u16x8*)(a.X + ((size_t)gi * a.T + t) * GA_WMAX + px_c[i]);
                 }
-                if (a.x_seq_out != nullptr)
-                    *(ga_u16x8*)(a.x_seq_out + ((size_t)gi * a.T + t) * GA_WMAX + col0) = v;
             }
-            ga_tile_write8(x_tile, m, col0, v);
+            px[i] = v;
+        }
+    };
+    px_load(0);
+
+    __syncthreads();
+
+    for (int t = 0; t < a.T; ++t) {
+        // ---- stage x_t tile from the prefetch registers -------------------
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            int gi = base + px_m[i];
+            if (gi < a.B && a.x_seq_out != nullptr)
+                *(ga_u16x8*)(a.x_seq_out + ((size_t)gi * a.T + t) * GA_WMAX + px_c[i]) = px[i];
+            ga_tile_write8(x_tile, px_m[i], px_c[i], px[i]);
         }
         // ---- deferred writer for t-1: 16-B stores of h, c, gates ---------
         // (overlapped with the x stage: both only read tiles the other
@@ -118,6 +139,7 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
             }
         }
         __syncthreads();
+        if (t + 1 < a.T) px_load(t + 1);  // overlap next x load with the GEMM
 
         unsigned short h_stage[4][4];  // deferred h_tile update (see below)
         // ---- gates GEMM + cell update, one 16-col tile at a time ---------
@@ -136,18 +158,19 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
                 ga_bf16x8 afrag = ga_tile_read8(src, wrow + (lane & 15), klocal + (lane >> 4) * 8);
 #pragma unroll
                 for (int g = 0; g < 4; ++g) {
-                    // B[k][j] = Wcat[j][k] row-read (16B, L2-resident)
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat + (size_t)(g * GA_WMAX + jt * 16 + (lane & 15)) * LSTM_K + kk + (lane >> 4) * 8);
+                    // B[k][j] = Wcat[j][k] row-read from the swizzled LDS image
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)&w_lds[ga_swW(
+                        g * GA_WMAX + jt * 16 + (lane & 15), (kk + (lane >> 4) * 8) * 2, 256)];
                     acc4[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc4[g], 0, 0, 0);
                 }
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int m = wrow + (lane >> 4) * 4 + r;
-                float i_g = 1.f / (1.f + __expf(-(acc4[0][r] + s_bias[col])));
-                float f_g = 1.f / (1.f + __expf(-(acc4[1][r] + s_bias[GA_WMAX + col])));
+                float i_g = ga_sigmoid(acc4[0][r] + s_bias[col]);
+                float f_g = ga_sigmoid(acc4[1][r] + s_bias[GA_WMAX + col]);
                 float g_g = ga_tanh(acc4[2][r] + s_bias[2 * GA_WMAX + col]);
-                float o_g = 1.f / (1.f + __expf(-(acc4[3][r] + s_bias[3 * GA_WMAX + col])));
+                float o_g = ga_sigmoid(acc4[3][r] + s_bias[3 * GA_WMAX + col]);
                 float c_new = f_g * c_reg[jt][r] + i_g * g_g;
                 c_reg[jt][r] = c_new;
                 float h_new = o_g * ga_tanh(c_new);
@@ -210,7 +233,10 @@ struct LstmBwdArgs {
     int dh_last_only;  // 1: dh_out is [B][64] at the final step only
 };
 
-__launch_bounds__(256, 2)
+// LDS: dh tile + 4 dgates tiles + the whole 64-KiB WcatT (staged once,
+// swizzled — same L2-latency fix as lstm_fwd). Elementwise inputs for step
+// t-1 are prefetched into registers during step t's GEMM.
+__launch_bounds__(256)
 __global__ void lstm_bwd_cell(LstmBwdArgs a) {
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -233,35 +259,57 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
     float dc_reg[16];
 #pragma unroll
     for (int e = 0; e < 16; ++e) dc_reg[e] = 0.f;
+
+    // stage WcatT into LDS once (rows 512 B -> swizzled 16-B slots)
+    char* wT_lds = ga_smem_lstm + 5 * GA_TILE_BYTES;  // [128][256]
+    for (int g = tid; g < LSTM_K * 32; g += 256) {
+        int row = g >> 5, col0 = (g & 31) * 8;
+        ga_u16x8 v = *(const ga_u16x8*)(a.wcat_t + (size_t)row * LSTM_4H + col0);
+        *(ga_u16x8*)&wT_lds[ga_swW(row, col0 * 2, 512)] = v;
+    }
+
+    // register prefetch of the elementwise inputs (issued during the GEMM of
+    // the following timestep so the HBM latency is off the critical path)
+    const int pgi = base + em;
+    const bool plive = pgi < a.B;
+    ga_u16x8 pvi[2], pvf[2], pvg[2], pvo[2], pvc[2], pvcp[2], pvdh[2];
+    auto preload = [&](int t) {
+        const size_t bt = (size_t)pgi * a.T + t;
+        const ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) {
+            int cc = c0 + hh * 8;
+            pvi[hh] = pvf[hh] = pvg[hh] = pvo[hh] = pvc[hh] = pvcp[hh] = pvdh[hh] = z;
+            if (plive) {
+                pvi[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc);
+                pvf[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
+                pvg[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
+                pvo[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
+                pvc[hh] = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
+                if (t > 0) pvcp[hh] = *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc);
+                if (a.dh_last_only) {
+                    if (t == a.T - 1)
+                        pvdh[hh] = *(const ga_u16x8*)(a.dh_out + (size_t)pgi * GA_WMAX + cc);
+                } else {
+                    pvdh[hh] = *(const ga_u16x8*)(a.dh_out + bt * GA_WMAX + cc);
+                }
+            }
+        }
+    };
+    preload(a.T - 1);
     __syncthreads();
 
     for (int t = a.T - 1; t >= 0; --t) {
-        // ---- elementwise BPTT: all global IO as 16-B vectors --------------
+        // ---- elementwise BPTT on the prefetched registers -----------------
         {
-            const int gi = base + em;
+            const int gi = pgi;
             const size_t bt = (size_t)gi * a.T + t;
-            const bool live = gi < a.B;
-            const ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
-            // one 8-col half at a time: keeps only ~7 input vectors + 4
-            // output vectors live (fits the 2-waves/SIMD VGPR budget)
+            const bool live = plive;
 #pragma unroll
             for (int hh = 0; hh < 2; ++hh) {
                 int cc = c0 + hh * 8;
-                ga_u16x8 vi = z, vf = z, vg = z, vo = z, vc = z, vcp = z, vdh = z;
-                if (live) {
-                    vi = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc);
-                    vf = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
-                    vg = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
-                    vo = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
-                    vc = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
-                    if (t > 0) vcp = *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc);
-                    if (a.dh_last_only) {
-                        if (t == a.T - 1)
-                            vdh = *(const ga_u16x8*)(a.dh_out + (size_t)gi * GA_WMAX + cc);
-                    } else {
-                        vdh = *(const ga_u16x8*)(a.dh_out + bt * GA_WMAX + cc);
-                    }
-                }
+                ga_u16x8 vi = pvi[hh], vf = pvf[hh], vg = pvg[hh], vo = pvo[hh],
+                         vc = pvc[hh], vcp = pvcp[hh], vdh = pvdh[hh];
                 ga_u16x8 vrec = *(ga_u16x8*)&dh_tile[ga_sw(em, cc)];
                 ga_u16x8 odi, odf, odg, odo;
 #pragma unroll
@@ -299,6 +347,7 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
         const bool need_rec = (t > 0);
         const bool need_dx = (a.dx_seq != nullptr);
         if (!need_rec && !need_dx) break;
+        if (t > 0) preload(t - 1);  // overlap next loads with the GEMM
 
         // ---- shared-A GEMMs over dgates tiles -----------------------------
         //   dh_rec[m][j<H] = dgates[m][256] x W_hh  (B row-read of WcatT 64+j)
@@ -315,13 +364,15 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
             if (need_rec)
 #pragma unroll
                 for (int nt = 0; nt < 4; ++nt) {
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t + (size_t)(GA_WMAX + nt * 16 + (lane & 15)) * LSTM_4H + kk + (lane >> 4) * 8);
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)&wT_lds[ga_swW(
+                        GA_WMAX + nt * 16 + (lane & 15), (kk + (lane >> 4) * 8) * 2, 512)];
                     acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
                 }
             if (need_dx)
 #pragma unroll
                 for (int nt = 0; nt < 4; ++nt) {
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t + (size_t)(nt * 16 + (lane & 15)) * LSTM_4H + kk + (lane >> 4) * 8);
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)&wT_lds[ga_swW(
+                        nt * 16 + (lane & 15), (kk + (lane >> 4) * 8) * 2, 512)];
                     accx[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, accx[nt], 0, 0, 0);
                 }
         }
@@ -390,10 +441,10 @@ __global__ void lstm_dw(LstmDwArgs a) {
     // — one barrier per chunk and the staging global-load latency hides
     // behind the MFMA phase instead of being exposed serially.
 #define DW_BUF_U16 ((256 + 128) * DW_STR)
-    unsigned short* dgT[2] = {(unsigned short*)smem,
-                              (unsigned short*)smem + DW_BUF_U16};
-    unsigned short* icT[2] = {dgT[0] + 256 * DW_STR, dgT[1] + 256 * DW_STR};
-    float* s_db = (float*)((unsigned short*)smem + 2 * DW_BUF_U16);  // [8][256]
+    unsigned short* const smem_u16 = (unsigned short*)smem;
+    float* s_db = (float*)(smem_u16 + 2 * DW_BUF_U16);  // [8][256]
+#define DW_DGT(buf) (smem_u16 + (buf) * DW_BUF_U16)
+#define DW_ICT(buf) (smem_u16 + (buf) * DW_BUF_U16 + 256 * DW_STR)
 
     ga_f32x4 acc[4][8];  // [jt][ct] — C tile [64 rows of gate wid][128 cols]
 #pragma unroll
@@ -421,7 +472,7 @@ __global__ void lstm_dw(LstmDwArgs a) {
             if (r < a.BT) v = *(const ga_u16x8*)(a.dgates + (size_t)r * LSTM_4H + dg_col0);
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
-                dgT[buf][(size_t)(dg_col0 + e) * DW_STR + row] = v[e];
+                DW_DGT(buf)[(size_t)(dg_col0 + e) * DW_STR + row] = v[e];
                 db_reg[e] += ga_bf16_to_f32(v[e]);
             }
         }
@@ -441,7 +492,7 @@ __global__ void lstm_dw(LstmDwArgs a) {
                 }
             }
 #pragma unroll
-            for (int e = 0; e < 8; ++e) icT[buf][(size_t)(ic_col0 + e) * DW_STR + row] = v[e];
+            for (int e = 0; e < 8; ++e) DW_ICT(buf)[(size_t)(ic_col0 + e) * DW_STR + row] = v[e];
         }
     };
 
@@ -453,10 +504,10 @@ __global__ void lstm_dw(LstmDwArgs a) {
         // ---- 32 MFMAs: C[64][128] += dgT_chunk x icT_chunk ----------------
 #pragma unroll
         for (int jt = 0; jt < 4; ++jt) {
-            ga_bf16x8 afrag = *(const ga_bf16x8*)(dgT[nb] + (size_t)(wid * 64 + jt * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+            ga_bf16x8 afrag = *(const ga_bf16x8*)(DW_DGT(nb) + (size_t)(wid * 64 + jt * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
 #pragma unroll
             for (int ct = 0; ct < 8; ++ct) {
-                ga_bf16x8 bfrag = *(const ga_bf16x8*)(icT[nb] + (size_t)(ct * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+                ga_bf16x8 bfrag = *(const ga_bf16x8*)(DW_ICT(nb) + (size_t)(ct * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
                 acc[jt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[jt][ct], 0, 0, 0);
             }
         }
@@ -512,7 +563,13 @@ void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
     a.T = (int)T;
     a.N = (int)X.size(0);
     int n_tiles = (int)((B + GA_MT - 1) / GA_MT);
-    size_t smem = 7 * GA_TILE_BYTES + LSTM_4H * sizeof(float);
+    size_t smem = 7 * GA_TILE_BYTES + LSTM_4H * sizeof(float) + LSTM_4H * LSTM_K * 2;
+    static bool smem_cfg = [] {  // >64 KiB dynamic LDS needs an explicit opt-in
+        (void)hipFuncSetAttribute((const void*)lstm_fwd,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+        return true;
+    }();
+    (void)smem_cfg;
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(lstm_fwd, dim3(n_tiles), dim3(256), smem, stream.stream(), a);
 }
@@ -532,7 +589,13 @@ void lstm_bwd_cell_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
     a.T = (int)T;
     a.dh_last_only = dh_last_only ? 1 : 0;
     int n_tiles = (int)((B + GA_MT - 1) / GA_MT);
-    size_t smem = 5 * GA_TILE_BYTES;
+    size_t smem = 5 * GA_TILE_BYTES + LSTM_K * LSTM_4H * 2;
+    static bool smem_cfg = [] {
+        (void)hipFuncSetAttribute((const void*)lstm_bwd_cell,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+        return true;
+    }();
+    (void)smem_cfg;
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(lstm_bwd_cell, dim3(n_tiles), dim3(256), smem, stream.stream(), a);
 }
