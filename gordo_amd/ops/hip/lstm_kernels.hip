@@ -417,9 +417,13 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
 // hprev materialization (h_{t-1} is read with an in-kernel t-1 shift).
 // ===========================================================================
 #define DW_KC 32                // k-rows per staged chunk
-#define DW_STR 40               // LDS row stride (u16) — 80 B, 16B-aligned,
-                                // odd multiple of 16B so successive k-rows
-                                // spread over LDS banks
+// Row-major chunk images with padded strides: staging is all 16-B vector
+// writes (the transposed scalar writes were 32-way bank-conflicted: col
+// stride 80 B puts every lane of a half-wave on one bank — measured 80% of
+// the kernel); the transpose happens at MFMA-fragment load time as scalar
+// reads, whose bank pattern the pad spreads.
+#define DW_DGS 264              // dg chunk row stride (u16): [32][264]
+#define DW_ICS 136              // ic chunk row stride (u16): [32][136]
 
 struct LstmDwArgs {
     const unsigned short* __restrict__ dgates;  // [BT][256]
@@ -440,11 +444,11 @@ __global__ void lstm_dw(LstmDwArgs a) {
     // double-buffered: stage chunk c+1 into buf[~c&1] WHILE MFMAing buf[c&1]
     // — one barrier per chunk and the staging global-load latency hides
     // behind the MFMA phase instead of being exposed serially.
-#define DW_BUF_U16 ((256 + 128) * DW_STR)
+#define DW_BUF_U16 (DW_KC * (DW_DGS + DW_ICS))
     unsigned short* const smem_u16 = (unsigned short*)smem;
     float* s_db = (float*)(smem_u16 + 2 * DW_BUF_U16);  // [8][256]
 #define DW_DGT(buf) (smem_u16 + (buf) * DW_BUF_U16)
-#define DW_ICT(buf) (smem_u16 + (buf) * DW_BUF_U16 + 256 * DW_STR)
+#define DW_ICT(buf) (smem_u16 + (buf) * DW_BUF_U16 + DW_KC * DW_DGS)
 
     ga_f32x4 acc[4][8];  // [jt][ct] — C tile [64 rows of gate wid][128 cols]
 #pragma unroll
@@ -461,22 +465,16 @@ __global__ void lstm_dw(LstmDwArgs a) {
 
     const int n_chunks = (a.BT + DW_KC - 1) / DW_KC;
 
-    auto stage = [&](int c, int buf) {
+    auto load_chunk = [&](int c, ga_u16x8* vdg, ga_u16x8* vic) {
         const int r0 = c * DW_KC;
-        // dgates chunk transposed (k-contiguous) + db partial accumulation
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
             int row = p * 8 + dg_row0;
             int r = r0 + row;
             ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
             if (r < a.BT) v = *(const ga_u16x8*)(a.dgates + (size_t)r * LSTM_4H + dg_col0);
-#pragma unroll
-            for (int e = 0; e < 8; ++e) {
-                DW_DGT(buf)[(size_t)(dg_col0 + e) * DW_STR + row] = v[e];
-                db_reg[e] += ga_bf16_to_f32(v[e]);
-            }
+            vdg[p] = v;
         }
-        // [x | h_{t-1}] transposed
 #pragma unroll
         for (int p = 0; p < 2; ++p) {
             int row = p * 16 + ic_row0;
@@ -491,26 +489,60 @@ __global__ void lstm_dw(LstmDwArgs a) {
                         v = *(const ga_u16x8*)(a.h_seq + (size_t)(r - 1) * GA_WMAX + ic_col0 - GA_WMAX);
                 }
             }
+            vic[p] = v;
+        }
+    };
+    auto write_chunk = [&](int buf, const ga_u16x8* vdg, const ga_u16x8* vic) {
 #pragma unroll
-            for (int e = 0; e < 8; ++e) DW_ICT(buf)[(size_t)(ic_col0 + e) * DW_STR + row] = v[e];
+        for (int p = 0; p < 4; ++p) {
+            int row = p * 8 + dg_row0;
+            *(ga_u16x8*)(DW_DGT(buf) + (size_t)row * DW_DGS + dg_col0) = vdg[p];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) db_reg[e] += ga_bf16_to_f32(vdg[p][e]);
+        }
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            int row = p * 16 + ic_row0;
+            *(ga_u16x8*)(DW_ICT(buf) + (size_t)row * DW_ICS + ic_col0) = vic[p];
         }
     };
 
     int nb = 0;
-    stage(blockIdx.x, 0);
+    ga_u16x8 vdg[4], vic[2];
+    load_chunk(blockIdx.x, vdg, vic);
+    write_chunk(0, vdg, vic);
     __syncthreads();
     for (int c = blockIdx.x; c < n_chunks; c += a.NB) {
-        if (c + a.NB < n_chunks) stage(c + a.NB, nb ^ 1);
-        // ---- 32 MFMAs: C[64][128] += dgT_chunk x icT_chunk ----------------
+        const bool more = (c + a.NB) < n_chunks;
+        if (more) load_chunk(c + a.NB, vdg, vic);  // loads overlap the MFMAs
+        // ---- 32 MFMAs: C[64][128] += dg_chunk^T x ic_chunk ----------------
+        // fragments gathered transposed: 8 scalar LDS reads each (row-major
+        // image; the k rows are the slow dim)
+        const int k0 = (lane >> 4) * 8;
+        ga_bf16x8 bfr[8];
 #pragma unroll
-        for (int jt = 0; jt < 4; ++jt) {
-            ga_bf16x8 afrag = *(const ga_bf16x8*)(DW_DGT(nb) + (size_t)(wid * 64 + jt * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
+        for (int ct = 0; ct < 8; ++ct) {
+            int colc = ct * 16 + (lane & 15);
 #pragma unroll
-            for (int ct = 0; ct < 8; ++ct) {
-                ga_bf16x8 bfrag = *(const ga_bf16x8*)(DW_ICT(nb) + (size_t)(ct * 16 + (lane & 15)) * DW_STR + (lane >> 4) * 8);
-                acc[jt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[jt][ct], 0, 0, 0);
+            for (int e = 0; e < 8; ++e) {
+                unsigned short u = DW_ICT(nb)[(size_t)(k0 + e) * DW_ICS + colc];
+                bfr[ct][e] = *(__bf16*)&u;
             }
         }
+#pragma unroll
+        for (int jt = 0; jt < 4; ++jt) {
+            int j = wid * 64 + jt * 16 + (lane & 15);
+            ga_bf16x8 afrag;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                unsigned short u = DW_DGT(nb)[(size_t)(k0 + e) * DW_DGS + j];
+                afrag[e] = *(__bf16*)&u;
+            }
+#pragma unroll
+            for (int ct = 0; ct < 8; ++ct)
+                acc[jt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfr[ct], acc[jt][ct], 0, 0, 0);
+        }
+        if (more) write_chunk(nb ^ 1, vdg, vic);  // waitcnt lands after MFMAs
         nb ^= 1;
         __syncthreads();
     }
