@@ -1,0 +1,131 @@
+#!/usr/bin/env python3
+"""Fleet benchmark (BASELINE config #5): build N per-machine anomaly models
+through the fleet builder, then measure anomaly-score latency through the
+serving runtime (the fused scale→forward→|err|→L2 HIP kernel per model).
+
+  python bench_serving.py --machines 32 --requests 500 --req-batch 1000
+
+Prints one JSON line: models built/hour + anomaly p50/p99 latency.
+(Single-node; the driver's headline bench stays bench.py.)
+"""
+import argparse
+import json
+import os
+import sys
+import tempfile
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+N_TAGS = 50
+
+
+def project_config(n_machines: int, n_tags: int):
+    machines = []
+    for i in range(n_machines):
+        machines.append({"name": f"machine-{i:03d}"})
+    return {
+        "project_name": "bench-fleet",
+        "machines": machines,
+        "globals": {
+            "dataset": {
+                "type": "RandomDataset",
+                "train_start_date": "2020-01-01T00:00:00Z",
+                "train_end_date": "2020-06-01T00:00:00Z",
+                "tag_list": [f"tag-{j}" for j in range(n_tags)],
+            },
+            "model": {
+                "gordo_amd.model.anomaly.diff.DiffBasedAnomalyDetector": {
+                    "base_estimator": {
+                        "sklearn.pipeline.Pipeline": {
+                            "steps": [
+                                "sklearn.preprocessing.MinMaxScaler",
+                                {
+                                    "gordo_amd.model.models.TorchAutoEncoder": {
+                                        "kind": "feedforward_hourglass",
+                                        "encoding_layers": 2,
+                                        "compression_factor": 0.5,
+                                        "epochs": 10,
+                                        "batch_size": 4096,
+                                    }
+                                },
+                            ]
+                        }
+                    }
+                }
+            },
+            "evaluation": {"cv_mode": "no_cv"},
+        },
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--machines", type=int, default=32)
+    ap.add_argument("--tags", type=int, default=N_TAGS)
+    ap.add_argument("--requests", type=int, default=500)
+    ap.add_argument("--req-batch", type=int, default=1000)
+    ap.add_argument("--out-dir", default=None)
+    args = ap.parse_args()
+
+    from gordo_amd.machine import NormalizedConfig
+    from gordo_amd.server.model_io import ModelRuntime
+    from gordo_amd.workflow.fleet import FleetBuilder
+
+    out_dir = args.out_dir or tempfile.mkdtemp(prefix="gordo-fleet-")
+    cfg = NormalizedConfig(project_config(args.machines, args.tags))
+
+    t0 = time.perf_counter()
+    builder = FleetBuilder(cfg, output_dir=out_dir)
+    results = builder.build_all()
+    build_s = time.perf_counter() - t0
+    ok = sum(1 for r in results if r["status"] == "succeeded")
+    assert ok == args.machines, [r for r in results if r["status"] != "succeeded"][:1]
+
+    # ---- serving latency: fused anomaly path, one runtime per machine -----
+    runtimes = [
+        ModelRuntime(os.path.join(out_dir, f"machine-{i:03d}"))
+        for i in range(args.machines)
+    ]
+    import pandas as pd
+
+    rng = np.random.default_rng(0)
+    tags = [f"tag-{j}" for j in range(args.tags)]
+    X = pd.DataFrame(
+        rng.uniform(0, 1, size=(args.req_batch, args.tags)).astype(np.float32),
+        columns=tags,
+        index=pd.date_range("2021-01-01", periods=args.req_batch, freq="10min"),
+    )
+    # warmup
+    for rt in runtimes[: min(8, len(runtimes))]:
+        rt.anomaly(X, None)
+    lat = []
+    for i in range(args.requests):
+        rt = runtimes[i % len(runtimes)]
+        t = time.perf_counter()
+        df = rt.anomaly(X, None)
+        lat.append(time.perf_counter() - t)
+    assert "total-anomaly-score" in set(c[0] for c in df.columns)
+    lat_ms = np.array(lat) * 1e3
+    fused = sum(1 for rt in runtimes if rt._fused is not None)
+
+    print(json.dumps({
+        "metric": "anomaly_p50_latency_ms",
+        "value": float(np.percentile(lat_ms, 50)),
+        "unit": "ms",
+        "p99_ms": float(np.percentile(lat_ms, 99)),
+        "req_batch_rows": args.req_batch,
+        "n_models": args.machines,
+        "fused_runtimes": fused,
+        "models_built": ok,
+        "build_seconds": build_s,
+        "models_per_hour": ok / build_s * 3600,
+        "higher_is_better": False,
+        "data": "synthetic",
+    }))
+
+
+if __name__ == "__main__":
+    main()
