@@ -19,5 +19,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_dw", &lstm_dw_launch, "split-K dWcat + dbias from one dgates pass");
     m.attr("WMAX") = 64;
     m.attr("MAXL") = 5;
-    m.attr("MT") = 64;
+    m.attr("MT") = 32;   // hg_train_step batch rows per block/slab
 }

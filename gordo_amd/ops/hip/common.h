@@ -61,6 +61,23 @@ __device__ __forceinline__ void ga_tile_write1(char* tile, int row, int col, uns
 }
 
 // ---------------------------------------------------------------------------
+// Narrow tiles for the MT=32 hourglass step: [R][32] bf16, 64-B rows, XOR
+// swizzle over the 4 16-B slots per row.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ unsigned int ga_sw32(int row, int col) {
+    unsigned int byte = (unsigned int)(col * 2) ^ (((unsigned int)row & 3u) << 4);
+    return ((unsigned int)row << 6) + byte;
+}
+
+__device__ __forceinline__ ga_bf16x8 ga_t32_read8(const char* tile, int row, int col0) {
+    return *(const ga_bf16x8*)(tile + ga_sw32(row, col0));
+}
+
+__device__ __forceinline__ void ga_t32_write1(char* tile, int row, int col, unsigned short v) {
+    *(unsigned short*)(tile + ga_sw32(row, col)) = v;
+}
+
+// ---------------------------------------------------------------------------
 // Swizzled LDS image for weight matrices read as MFMA B-fragments: 16-lane
 // groups read 16 consecutive ROWS at one column offset; a plain row-major
 // image puts all 16 on one bank (row stride is a multiple of 256 B). XORing

@@ -42,11 +42,17 @@ struct HgTrainArgs {
     float gscale;   // 2 / (B * F_out)
 };
 
-// LDS plan: acts[L] row-major swizzled tiles + 4 dpre scratch tiles + bias.
-// L=5: 5*8K + 4*8K + 1.3K = 73.6 KiB -> 2 blocks/CU.
+// MT=32 geometry: one block = 32 batch rows; B=8192 -> 256 blocks = every
+// CU busy (the original 64-row tiling left half the chip idle at the
+// headline batch). The 4 waves split the work 2-D: wid&1 = 16-row stripe,
+// wid>>1 = 32-col stripe. LDS: acts[L] + 4 dpre scratch tiles, all 4 KiB
+// ([32][64] via ga_sw, [64][32] transposed via ga_sw32) + bias ~ 38 KiB.
+#define HG_MT 32
+#define HG_TILE_BYTES (HG_MT * GA_WMAX * 2)  // 4 KiB
+
 extern __shared__ __attribute__((aligned(16))) char ga_smem[];
 
-__device__ __forceinline__ char* act_tile(int i) { return ga_smem + i * GA_TILE_BYTES; }
+__device__ __forceinline__ char* act_tile(int i) { return ga_smem + i * HG_TILE_BYTES; }
 
 __launch_bounds__(256)
 __global__ void hg_train_step(HgTrainArgs a) {
@@ -54,13 +60,15 @@ __global__ void hg_train_step(HgTrainArgs a) {
     const int lane = tid & 63;
     const int wid = tid >> 6;
     const int tile = blockIdx.x;
-    const int base = tile * GA_MT;
+    const int base = tile * HG_MT;
+    const int wrow = (wid & 1) * 16;   // this wave's 16-row stripe
+    const int wcol = (wid >> 1) * 32;  // this wave's 32-col stripe
 
-    char* scratch0 = ga_smem + a.L * GA_TILE_BYTES;        // dcur
-    char* scratch1 = scratch0 + GA_TILE_BYTES;             // dcurT
-    char* scratch2 = scratch1 + GA_TILE_BYTES;             // dnext
-    char* scratch3 = scratch2 + GA_TILE_BYTES;             // dnextT
-    float* s_bias = (float*)(scratch3 + GA_TILE_BYTES);    // [L][64]
+    char* scratch0 = ga_smem + a.L * HG_TILE_BYTES;        // dcur  [32][64]
+    char* scratch1 = scratch0 + HG_TILE_BYTES;             // dcurT [64][32]
+    char* scratch2 = scratch1 + HG_TILE_BYTES;             // dnext
+    char* scratch3 = scratch2 + HG_TILE_BYTES;             // dnextT
+    float* s_bias = (float*)(scratch3 + HG_TILE_BYTES);    // [L][64]
     float* s_red = s_bias + GA_MAXL * GA_WMAX;             // [4][64] reduce pad
 
     const int step = a.counters[0];
@@ -69,11 +77,10 @@ __global__ void hg_train_step(HgTrainArgs a) {
     // ---- stage biases -------------------------------------------------
     for (int i = tid; i < a.L * GA_WMAX; i += 256) s_bias[i] = a.bias[i];
 
-    // ---- gather minibatch tile into acts[0] ----------------------------
-    // 64 rows x 8 chunks of 8 bf16; 512 chunks over 256 threads.
-    for (int c = tid; c < GA_MT * 8; c += 256) {
-        int m = c >> 3;
-        int col0 = (c & 7) * 8;
+    // ---- gather minibatch tile into acts[0]: one 16-B granule/thread ---
+    {
+        int m = tid >> 3;
+        int col0 = (tid & 7) * 8;
         int gi = base + m;
         ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (gi < a.B) {
@@ -84,32 +91,27 @@ __global__ void hg_train_step(HgTrainArgs a) {
     }
     __syncthreads();
 
-    const int wrow = wid * 16;  // this wave's row stripe
-
     // =====================================================================
-    // Forward: layers 0..L-2 write their activation tile; layer L-1 goes
-    // straight into the loss epilogue producing dpre_{L-1} in scratch0/1.
+    // Forward: per wave [16 rows][32 cols] = 2 MFMA tiles x 2 K-steps.
     // =====================================================================
     for (int li = 0; li < a.L; ++li) {
         const unsigned short* wl = a.w + (size_t)li * GA_WMAX * GA_WMAX;
-        ga_f32x4 acc[GA_NT];
-        for (int nt = 0; nt < GA_NT; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
-
+        ga_f32x4 acc[2];
+        acc[0] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+        acc[1] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
         for (int kk = 0; kk < GA_WMAX; kk += 32) {
             ga_bf16x8 afrag = ga_tile_read8(act_tile(li), wrow + (lane & 15), kk + (lane >> 4) * 8);
 #pragma unroll
-            for (int nt = 0; nt < GA_NT; ++nt) {
-                ga_bf16x8 bfrag = *(const ga_bf16x8*)(wl + (size_t)(nt * 16 + (lane & 15)) * GA_WMAX + kk + (lane >> 4) * 8);
+            for (int nt = 0; nt < 2; ++nt) {
+                ga_bf16x8 bfrag = *(const ga_bf16x8*)(wl + (size_t)(wcol + nt * 16 + (lane & 15)) * GA_WMAX + kk + (lane >> 4) * 8);
                 acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
             }
         }
 
         if (li + 1 < a.L) {
-            // hidden epilogue: bias + act -> acts[li+1] (a different tile than
-            // the one being read, so no barrier needed before the writes)
 #pragma unroll
-            for (int nt = 0; nt < GA_NT; ++nt) {
-                int col = nt * 16 + (lane & 15);
+            for (int nt = 0; nt < 2; ++nt) {
+                int col = wcol + nt * 16 + (lane & 15);
                 float b = s_bias[li * GA_WMAX + col];
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
@@ -123,8 +125,8 @@ __global__ void hg_train_step(HgTrainArgs a) {
             // ---- loss epilogue: dpre_{L-1} = (yhat - y) * gscale --------
             float lsum = 0.f;
 #pragma unroll
-            for (int nt = 0; nt < GA_NT; ++nt) {
-                int col = nt * 16 + (lane & 15);
+            for (int nt = 0; nt < 2; ++nt) {
+                int col = wcol + nt * 16 + (lane & 15);
                 float b = s_bias[li * GA_WMAX + col];
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
@@ -146,7 +148,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
                     float dpre = diff * a.gscale * ga_act_grad_from_h(yhat, a.act_id[li]);
                     unsigned short dp = ga_f32_to_bf16(dpre);
                     ga_tile_write1(scratch0, m, col, dp);
-                    ga_tile_write1(scratch1, col, m, dp);
+                    ga_t32_write1(scratch1, col, m, dp);
                 }
             }
             lsum = ga_wave_sum(lsum);
@@ -160,12 +162,9 @@ __global__ void hg_train_step(HgTrainArgs a) {
     __syncthreads();  // s_red is reused by the db reduction below
 
     // =====================================================================
-    // Backward: for li = L-1 .. 0:
-    //   dX (li>0):  dacts[li] = dpre_li x W_li  (B-frag from WT global rows)
-    //               dpre_{li-1} = dacts * act'(acts[li]) -> scratch2/3
-    //   dW:         dW_li = acts[li]^T x dpre_li (B-frag from dcurT rows)
-    //   db:         column sum of dpre_li (via dcurT row sums)
-    // scratch0/1 = dpre_li (row/col major); scratch2/3 = dpre_{li-1}.
+    // Backward: per li:  dX -> dpre_{li-1} (waves split [16r][32c]);
+    //   dW[64 in][64 out] (wave = 16-in stripe, K = 32 rows, 1 K-step);
+    //   db via dcurT row sums.
     // =====================================================================
     char* dcur = scratch0;
     char* dcurT = scratch1;
@@ -173,24 +172,24 @@ __global__ void hg_train_step(HgTrainArgs a) {
     char* dnextT = scratch3;
 
     for (int li = a.L - 1; li >= 0; --li) {
-        // ---- dX: per-wave row stripe over m ----------------------------
         if (li > 0 || a.dx_out != nullptr) {
             const unsigned short* wtl = a.wt + (size_t)li * GA_WMAX * GA_WMAX;
-            ga_f32x4 acc[GA_NT];
-            for (int nt = 0; nt < GA_NT; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            ga_f32x4 acc[2];
+            acc[0] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            acc[1] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
             for (int kk = 0; kk < GA_WMAX; kk += 32) {
                 ga_bf16x8 afrag = ga_tile_read8(dcur, wrow + (lane & 15), kk + (lane >> 4) * 8);
 #pragma unroll
-                for (int nt = 0; nt < GA_NT; ++nt) {
+                for (int nt = 0; nt < 2; ++nt) {
                     // B[k=out][j=in] = WT[in][out]: row-read of WT
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(wtl + (size_t)(nt * 16 + (lane & 15)) * GA_WMAX + kk + (lane >> 4) * 8);
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(wtl + (size_t)(wcol + nt * 16 + (lane & 15)) * GA_WMAX + kk + (lane >> 4) * 8);
                     acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
                 }
             }
             if (li > 0) {
 #pragma unroll
-                for (int nt = 0; nt < GA_NT; ++nt) {
-                    int col = nt * 16 + (lane & 15);
+                for (int nt = 0; nt < 2; ++nt) {
+                    int col = wcol + nt * 16 + (lane & 15);
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
                         int m = wrow + (lane >> 4) * 4 + r;
@@ -198,14 +197,14 @@ __global__ void hg_train_step(HgTrainArgs a) {
                         float d = acc[nt][r] * ga_act_grad_from_h(h, a.act_id[li - 1]);
                         unsigned short dp = ga_f32_to_bf16(d);
                         ga_tile_write1(dnext, m, col, dp);
-                        ga_tile_write1(dnextT, col, m, dp);
+                        ga_t32_write1(dnextT, col, m, dp);
                     }
                 }
             } else {
                 // dL/d(input) for the caller (e.g. the LSTM head): batch order
 #pragma unroll
-                for (int nt = 0; nt < GA_NT; ++nt) {
-                    int col = nt * 16 + (lane & 15);
+                for (int nt = 0; nt < 2; ++nt) {
+                    int col = wcol + nt * 16 + (lane & 15);
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
                         int m = wrow + (lane >> 4) * 4 + r;
@@ -217,25 +216,23 @@ __global__ void hg_train_step(HgTrainArgs a) {
             }
         }
 
-        // ---- dW: per-wave stripe over `in`, K = m ----------------------
+        // ---- dW: wave = 16-in stripe x 64 out, K = m (32, one K-step) ----
         {
             ga_f32x4 acc[GA_NT];
             for (int nt = 0; nt < GA_NT; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
-            for (int kk = 0; kk < GA_MT; kk += 32) {
-                // A[i=in][k=m] = acts[li][m][in]: strided scalar reads
-                ga_bf16x8 afrag;
+            // A[i=in][k=m] = acts[li][m][in]: strided scalar reads
+            ga_bf16x8 afrag;
 #pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    int m = kk + (lane >> 4) * 8 + e;
-                    unsigned short u = ga_tile_read1(act_tile(li), m, wrow + (lane & 15));
-                    afrag[e] = *(__bf16*)&u;
-                }
+            for (int e = 0; e < 8; ++e) {
+                int m = (lane >> 4) * 8 + e;
+                unsigned short u = ga_tile_read1(act_tile(li), m, wid * 16 + (lane & 15));
+                afrag[e] = *(__bf16*)&u;
+            }
 #pragma unroll
-                for (int nt = 0; nt < GA_NT; ++nt) {
-                    // B[k=m][j=out] = dcurT[out][m]: row-read of dcurT
-                    ga_bf16x8 bfrag = ga_tile_read8(dcurT, nt * 16 + (lane & 15), kk + (lane >> 4) * 8);
-                    acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
-                }
+            for (int nt = 0; nt < GA_NT; ++nt) {
+                // B[k=m][j=out] = dcurT[out][m]: row-read of the narrow tile
+                ga_bf16x8 bfrag = ga_t32_read8(dcurT, nt * 16 + (lane & 15), (lane >> 4) * 8);
+                acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
             }
             float* dwl = a.dw + ((size_t)tile * a.L + li) * GA_WMAX * GA_WMAX;
 #pragma unroll
@@ -243,7 +240,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
                 int out = nt * 16 + (lane & 15);
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    int in = wrow + (lane >> 4) * 4 + r;
+                    int in = wid * 16 + (lane >> 4) * 4 + r;
                     dwl[(size_t)in * GA_WMAX + out] = acc[nt][r];
                 }
             }
@@ -251,16 +248,13 @@ __global__ void hg_train_step(HgTrainArgs a) {
 
         // ---- db: column sums of dpre via dcurT row sums ----------------
         {
-            int out = tid >> 2;          // 64 outs x 4 groups
-            int g = tid & 3;             // 16 m's per group
-            float s = 0.f;
+            int out = tid >> 2;          // 64 outs x 4 groups of 8 m's
+            int g = tid & 3;
+            ga_bf16x8 v = ga_t32_read8(dcurT, out, g * 8);
+            float sum = 0.f;
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi) {
-                ga_bf16x8 v = ga_tile_read8(dcurT, out, g * 16 + mi * 8);
-#pragma unroll
-                for (int e = 0; e < 8; ++e) s += (float)v[e];
-            }
-            s_red[g * GA_WMAX + out] = s;
+            for (int e = 0; e < 8; ++e) sum += (float)v[e];
+            s_red[g * GA_WMAX + out] = sum;
             __syncthreads();
             if (tid < GA_WMAX) {
                 float tot = s_red[tid] + s_red[GA_WMAX + tid] + s_red[2 * GA_WMAX + tid] +
@@ -635,9 +629,9 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
     ga_fill_acts(a.act_id, acts, L);
     a.gscale = (float)gscale;
 
-    int n_tiles = (int)((B + GA_MT - 1) / GA_MT);
+    int n_tiles = (int)((B + HG_MT - 1) / HG_MT);
     TORCH_CHECK(dw.size(0) >= n_tiles, "dw slab count too small");
-    size_t smem = (size_t)(L + 4) * GA_TILE_BYTES + GA_MAXL * GA_WMAX * 4 + 4 * GA_WMAX * 4;
+    size_t smem = (size_t)(L + 4) * HG_TILE_BYTES + GA_MAXL * GA_WMAX * 4 + 4 * GA_WMAX * 4;
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(hg_train_step, dim3(n_tiles), dim3(256), smem, stream.stream(), a);
 }

@@ -114,7 +114,8 @@ class FusedHourglassTrainer:
 
         N = Xd.shape[0]
         B = max(64, min(cfg.batch_size, N))
-        n_tiles = (B + 63) // 64
+        MT = int(getattr(self.E, "MT", 32))
+        n_tiles = (B + MT - 1) // MT
         n_batches = max(1, math.ceil(N / B))
 
         dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=device)

@@ -203,6 +203,8 @@ class FusedLSTMTrainer:
         device = self.device
         Lc = self.p.Lc
         bf16 = torch.bfloat16
+        MT = int(getattr(self.E, "MT", 32))
+        n_head_tiles = (B + MT - 1) // MT
         st = {
             "Xd": Xd, "Yd": Yd, "B": B, "T": T, "n_batches": n_batches,
             "n_windows": n_windows, "lookahead": lookahead, "F_out": F_out,
@@ -223,8 +225,8 @@ class FusedLSTMTrainer:
             "dx_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
             "perm_id": torch.arange(B, dtype=torch.int32, device=device),
             "counters0": torch.tensor([0, 1], dtype=torch.int32, device=device),
-            "head_dw": torch.zeros((B + 63) // 64, 1, WMAX * WMAX, dtype=torch.float32, device=device),
-            "head_db": torch.zeros((B + 63) // 64, 1, WMAX, dtype=torch.float32, device=device),
+            "head_dw": torch.zeros(n_head_tiles, 1, WMAX * WMAX, dtype=torch.float32, device=device),
+            "head_db": torch.zeros(n_head_tiles, 1, WMAX, dtype=torch.float32, device=device),
             "loss_acc": torch.zeros(1, dtype=torch.float32, device=device),
             "gscale": 2.0 / (B * F_out),
             "act_out": [ACT_IDS[self.module.out_func]],

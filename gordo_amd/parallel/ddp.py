@@ -117,7 +117,8 @@ class HourglassStepTrainer:
         self.Xd = pad_rows_bf16(np.asarray(self.X, np.float32), device)
         N = self.Xd.shape[0]
         self.N = N
-        self.n_tiles = (self.B + 63) // 64
+        MT = int(getattr(self.E, "MT", 32))
+        self.n_tiles = (self.B + MT - 1) // MT
         self.dw = torch.zeros(self.n_tiles, p.L, WMAX * WMAX, dtype=torch.float32, device=device)
         self.db = torch.zeros(self.n_tiles, p.L, WMAX, dtype=torch.float32, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)

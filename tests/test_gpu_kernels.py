@@ -100,7 +100,8 @@ def test_train_step_gradients_match_autograd():
     L, F_out = p.L, p.widths[-1]
     Xd = pad_rows_bf16(X, DEV)
     B = N
-    n_tiles = (B + 63) // 64
+    MT = int(E.MT)
+    n_tiles = (B + MT - 1) // MT
     dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=DEV)
     db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
