@@ -344,32 +344,38 @@ __global__ void hg_bump(int* counters, int n_batches) {
 __launch_bounds__(256)
 __global__ void hg_slab_reduce(const float* __restrict__ dw,
                                const float* __restrict__ db,
-                               float* __restrict__ partial,  // [G][P]
+                               float* __restrict__ partial,  // [G][P] SLAB layout
                                int* __restrict__ counters,
                                int n_slabs, int G, int L, int n_batches) {
+    // stays in slab layout ([l][in][out]) so every access is a contiguous
+    // float4 stream; the [in][out]->[out][in] transpose happens in the tiny
+    // stage-B Adam instead (1.3 MB vs 21 MB here).
     const int PW = L * GA_WMAX * GA_WMAX;
     const int P = PW + L * GA_WMAX;
     int g = blockIdx.y;
-    int idx = blockIdx.x * blockDim.x + threadIdx.x;
-    if (g == 0 && idx == 0 && counters != nullptr) {
+    int idx4 = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g == 0 && idx4 == 0 && counters != nullptr) {
         counters[0] = (counters[0] + 1) % n_batches;
         counters[1] += 1;
     }
+    int idx = idx4 * 4;
     if (idx >= P) return;
     int per = (n_slabs + G - 1) / G;
     int s0 = g * per, s1 = min(n_slabs, s0 + per);
-    float s = 0.f;
+    float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
     if (idx < PW) {
-        int l = idx / (GA_WMAX * GA_WMAX);
-        int rem = idx - l * GA_WMAX * GA_WMAX;
-        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
-        size_t sidx = (size_t)l * GA_WMAX * GA_WMAX + (size_t)in * GA_WMAX + out;
-        for (int sl = s0; sl < s1; ++sl) s += dw[(size_t)sl * PW + sidx];
+        for (int sl = s0; sl < s1; ++sl) {
+            float4 v = *(const float4*)(dw + (size_t)sl * PW + idx);
+            acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+        }
     } else {
         int bidx = idx - PW;
-        for (int sl = s0; sl < s1; ++sl) s += db[(size_t)sl * (L * GA_WMAX) + bidx];
+        for (int sl = s0; sl < s1; ++sl) {
+            float4 v = *(const float4*)(db + (size_t)sl * (L * GA_WMAX) + bidx);
+            acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+        }
     }
-    partial[(size_t)g * P + idx] = s;
+    *(float4*)(partial + (size_t)g * P + idx) = acc;
 }
 
 __launch_bounds__(256)
@@ -381,10 +387,20 @@ __global__ void hg_adam_p(
     float lr, float beta1, float beta2, float eps) {
     const int PW = L * GA_WMAX * GA_WMAX;
     const int P = PW + L * GA_WMAX;
-    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;  // MASTER-layout index
     if (idx >= P) return;
+    // partial is in SLAB layout: transpose on read
+    size_t sidx;
+    if (idx < PW) {
+        int l = idx / (GA_WMAX * GA_WMAX);
+        int rem = idx - l * GA_WMAX * GA_WMAX;
+        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
+        sidx = (size_t)l * GA_WMAX * GA_WMAX + (size_t)in * GA_WMAX + out;
+    } else {
+        sidx = (size_t)idx;
+    }
     float g = 0.f;
-    for (int i = 0; i < G; ++i) g += partial[(size_t)i * P + idx];
+    for (int i = 0; i < G; ++i) g += partial[(size_t)i * P + sidx];
     int t = counters[1];
     float m = beta1 * m_buf[idx] + (1.f - beta1) * g;
     float v = beta2 * v_buf[idx] + (1.f - beta2) * g * g;
@@ -410,12 +426,23 @@ __global__ void hg_adam_p(
 __launch_bounds__(256)
 __global__ void hg_partial_flat(const float* __restrict__ partial,
                                 float* __restrict__ g_out, int G, int L) {
-    const int P = L * GA_WMAX * GA_WMAX + L * GA_WMAX;
+    // partial is SLAB layout; emit MASTER layout for the DP bucket
+    const int PW = L * GA_WMAX * GA_WMAX;
+    const int P = PW + L * GA_WMAX;
     int idx = blockIdx.x * blockDim.x + threadIdx.x;
     if (idx >= P) return;
-    float s = 0.f;
-    for (int i = 0; i < G; ++i) s += partial[(size_t)i * P + idx];
-    g_out[idx] = s;
+    size_t sidx;
+    if (idx < PW) {
+        int l = idx / (GA_WMAX * GA_WMAX);
+        int rem = idx - l * GA_WMAX * GA_WMAX;
+        int out = rem / GA_WMAX, in = rem - out * GA_WMAX;
+        sidx = (size_t)l * GA_WMAX * GA_WMAX + (size_t)in * GA_WMAX + out;
+    } else {
+        sidx = (size_t)idx;
+    }
+    float sum = 0.f;
+    for (int i = 0; i < G; ++i) sum += partial[(size_t)i * P + sidx];
+    g_out[idx] = sum;
 }
 
 // ===========================================================================
@@ -691,7 +718,7 @@ void hg_slab_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor par
                            c10::optional<torch::Tensor> counters, int64_t n_slabs,
                            int64_t G, int64_t L, int64_t n_batches) {
     int P = (int)(L * GA_WMAX * GA_WMAX + L * GA_WMAX);
-    dim3 grid((P + 255) / 256, (int)G);
+    dim3 grid((P / 4 + 255) / 256, (int)G);
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(hg_slab_reduce, grid, dim3(256), 0, stream.stream(),
                        dw.data_ptr<float>(), db.data_ptr<float>(),
