@@ -39,6 +39,34 @@ from ..util import disk_registry
 
 logger = logging.getLogger(__name__)
 
+
+class _phase_range:
+    """roctx range around a build phase (torch.cuda.nvtx -> roctx on ROCm);
+    makes builder phases visible in rocprof timelines. No-op off-GPU."""
+
+    def __init__(self, name: str):
+        self.name = name
+
+    def __enter__(self):
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                torch.cuda.nvtx.range_push(self.name)
+                self._active = True
+                return self
+        except Exception:
+            pass
+        self._active = False
+        return self
+
+    def __exit__(self, *exc):
+        if getattr(self, "_active", False):
+            import torch
+
+            torch.cuda.nvtx.range_pop()
+        return False
+
 METRIC_FUNCS = {
     "explained_variance_score": explained_variance_score,
     "r2_score": r2_score,
@@ -130,7 +158,8 @@ class ModelBuilder:
 
         dataset = dataset_from_config(machine.dataset)
         t_query = time.time()
-        X, y = dataset.get_data()
+        with _phase_range("gordo:data-query"):
+            X, y = dataset.get_data()
         data_query_duration = time.time() - t_query
 
         model = from_definition(machine.model)
@@ -150,13 +179,15 @@ class ModelBuilder:
         }
 
         if cv_mode in ("full_build", "cross_val_only"):
-            cv_meta = self._cross_validate(model, X, y, evaluation)
+            with _phase_range("gordo:cross-validate"):
+                cv_meta = self._cross_validate(model, X, y, evaluation)
             metadata["model"]["cross-validation"] = cv_meta
 
         if cv_mode != "cross_val_only":
             t_fit = time.time()
             yv = y if y is not None else X
-            model.fit(_values(X), _values(yv))
+            with _phase_range("gordo:fit"):
+                model.fit(_values(X), _values(yv))
             metadata["model"]["model-training-duration-sec"] = time.time() - t_fit
         metadata["model"]["model-builder-version"] = __version__
         if isinstance(model, GordoBase) or hasattr(model, "get_metadata"):

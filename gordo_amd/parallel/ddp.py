@@ -49,7 +49,13 @@ class DPContext:
                 backend = "nccl" if device.type == "cuda" else "gloo"
             if device.type == "cuda":
                 torch.cuda.set_device(device)
-            dist.init_process_group(backend=backend)
+            # watchdog: a hung rank aborts the job instead of deadlocking it
+            import datetime
+
+            timeout_s = float(os.environ.get("GORDO_DIST_TIMEOUT", "600"))
+            dist.init_process_group(
+                backend=backend, timeout=datetime.timedelta(seconds=timeout_s)
+            )
         self.backend = backend
 
     def all_reduce_mean(self, t: torch.Tensor):
@@ -134,7 +140,9 @@ class HourglassStepTrainer:
         self._fill_perm()
         self.gscale = 2.0 / (self.B * self.F_out)
         self._graph = None
-        if getattr(self.cfg, "use_graph", True):
+        # GORDO_NO_GRAPH=1 is the operational escape hatch for the captured
+        # step (e.g. if an RCCL build misbehaves inside graph replay)
+        if getattr(self.cfg, "use_graph", True) and not os.environ.get("GORDO_NO_GRAPH"):
             self._try_capture_graph()
 
     def _try_capture_graph(self):
