@@ -169,12 +169,24 @@ class DiffBasedAnomalyDetector(BaseEstimator, AnomalyDetectorBase):
                 "require_thresholds=True but this model was built without "
                 "cross_validate-derived thresholds"
             )
+        Xv = X.values if isinstance(X, pd.DataFrame) else np.asarray(X)
+        out = self.base_estimator.predict(Xv)
+        return self.anomaly_from_prediction(X, y, out, frequency)
+
+    def anomaly_from_prediction(
+        self, X, y, out: np.ndarray, frequency: Optional[pd.Timedelta] = None
+    ) -> pd.DataFrame:
+        """Assemble the anomaly frame from an already-computed prediction
+        (lets the server's fused HIP forward feed the same frame logic)."""
+        if self.require_thresholds and self.aggregate_threshold_ is None:
+            raise AttributeError(
+                "require_thresholds=True but this model was built without "
+                "cross_validate-derived thresholds"
+            )
         is_frame = isinstance(X, pd.DataFrame)
         index = X.index if is_frame else None
         Xv = X.values if is_frame else np.asarray(X)
         yv = y.values if isinstance(y, pd.DataFrame) else np.asarray(y)
-
-        out = self.base_estimator.predict(Xv)
         offset = len(yv) - len(out)
 
         tags = list(X.columns) if is_frame else [str(i) for i in range(Xv.shape[1])]

@@ -93,6 +93,12 @@ class ModelRuntime:
             raise TypeError("Loaded model is not an anomaly detector")
         if y is None:
             y = X
+        if self._fused is not None:
+            # fused HIP forward for the expensive part; the detector's own
+            # frame/threshold logic consumes the prediction unchanged
+            xt = torch.as_tensor(np.ascontiguousarray(X.values, dtype=np.float32))
+            out = self._fused.forward(xt).cpu().numpy()
+            return self.model.anomaly_from_prediction(X, y, out, frequency)
         return self.model.anomaly(X, y, frequency)
 
     @property
