@@ -34,6 +34,7 @@ def runtime_pair(tmp_path_factory):
                             {
                                 "gordo_amd.model.models.TorchAutoEncoder": {
                                     "kind": "feedforward_hourglass",
+                                    "encoding_layers": 2,
                                     "epochs": 3,
                                     "batch_size": 512,
                                 }
