@@ -41,33 +41,39 @@ struct LstmFwdArgs {
 
 extern __shared__ __attribute__((aligned(16))) char ga_smem_lstm[];
 
-// LDS plan (~121.25 KiB -> 1 block/CU): x, h, c + 4 gate tiles + bias + the
-// whole 64-KiB Wcat (staged ONCE, swizzled; ablation showed the per-t L2
-// B-fragment loads were 74% of the kernel). The per-t x tile is prefetched
-// into registers during the previous GEMM so its HBM latency is hidden.
-__launch_bounds__(256)
+// Gate-specialized design: wave w owns gate w. Its 64x128 weight slice lives
+// entirely in REGISTERS (16 bf16x8 fragments/lane, loaded once) and its bias
+// is pre-added into the MFMA accumulators — the per-timestep weight traffic
+// (originally 74% of the kernel as L2 loads, then an LDS image) is zero.
+// Per t: waves compute their gate pre-activation tile [64 m][64 j] -> LDS;
+// a vectorized elementwise phase (thread = row x 16 cols) applies the cell
+// nonlinearity on 16-B vectors and streams h/c/gates straight to HBM.
+// LDS: x + h + 4 pre-act tiles = 48 KiB -> 2 blocks/CU at ~190 VGPRs.
+__launch_bounds__(256, 2)
 __global__ void lstm_fwd(LstmFwdArgs a) {
     const int tid = threadIdx.x;
     const int lane = tid & 63;
-    const int wid = tid >> 6;
+    const int wid = tid >> 6;  // wave = gate
     const int base = blockIdx.x * GA_MT;
-    const int wrow = wid * 16;
 
     char* x_tile = ga_smem_lstm;                     // [64][64] swizzled bf16
     char* h_tile = ga_smem_lstm + GA_TILE_BYTES;     // [64][64]
-    char* c_tile = ga_smem_lstm + 2 * GA_TILE_BYTES; // [64][64]
-    char* g_tiles = ga_smem_lstm + 3 * GA_TILE_BYTES;  // 4 x [64][64] i,f,g,o
-    float* s_bias = (float*)(ga_smem_lstm + 7 * GA_TILE_BYTES);  // [256]
-    char* w_lds = ga_smem_lstm + 7 * GA_TILE_BYTES + 1024;  // [256][128] swizzled
+    char* pre_tiles = ga_smem_lstm + 2 * GA_TILE_BYTES;  // 4 x [64][64] pre+bias
 
-    for (int i = tid; i < LSTM_4H; i += 256) s_bias[i] = a.bias[i];
-
-    // stage the whole Wcat into LDS once (re-read 4*T times by the GEMM)
-    for (int g = tid; g < LSTM_4H * 16; g += 256) {
-        int row = g >> 4, col0 = (g & 15) * 8;
-        ga_u16x8 v = *(const ga_u16x8*)(a.wcat + (size_t)row * LSTM_K + col0);
-        *(ga_u16x8*)&w_lds[ga_swW(row, col0 * 2, 256)] = v;
-    }
+    // ---- per-lane weight fragments (whole gate slice, registers) ---------
+    // wfrag[jt][kk] = Wcat[wid*64 + jt*16 + (lane&15)][kk*32 + (lane>>4)*8 ..+8]
+    ga_bf16x8 wfrag[4][4];
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt)
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk)
+            wfrag[jt][kk] = *(const ga_bf16x8*)(a.wcat +
+                (size_t)(wid * GA_WMAX + jt * 16 + (lane & 15)) * LSTM_K +
+                kk * 32 + (lane >> 4) * 8);
+    float bias_reg[4];
+#pragma unroll
+    for (int jt = 0; jt < 4; ++jt)
+        bias_reg[jt] = a.bias[wid * GA_WMAX + jt * 16 + (lane & 15)];
 
     // zero h_tile (h_{-1} = 0)
     for (int c = tid; c < GA_MT * 8; c += 256) {
@@ -75,21 +81,17 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
         ga_tile_write8(h_tile, c >> 3, (c & 7) * 8, z);
     }
 
-    // c state in registers: lane owns (rows wrow + (lane>>4)*4 + r, col jt*16 + (lane&15))
-    float c_reg[4][4];  // [jt][r]
+    // elementwise-phase mapping: row em = tid>>2, 16 cols from ec0
+    const int em = tid >> 2;
+    const int ec0 = (tid & 3) * 16;
+    const int egi = base + em;
+    float c_reg[16];
 #pragma unroll
-    for (int jt = 0; jt < 4; ++jt)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) c_reg[jt][r] = 0.f;
-
-    // writer-phase mapping: row m = tid>>2, 16 cols starting at (tid&3)*16
-    const int wm = tid >> 2;
-    const int wc0 = (tid & 3) * 16;
-    const int wgi = base + wm;
+    for (int e = 0; e < 16; ++e) c_reg[e] = 0.f;
 
     // x prefetch: thread owns two fixed (row, col) granules of the x tile
-    const int px_m[2] = {tid >> 3, (tid + 256) >> 3};
-    const int px_c[2] = {(tid & 7) * 8, (tid & 7) * 8};
+    const int px_m[2] = {tid >> 3, (tid >> 3) + 32};
+    const int px_c = (tid & 7) * 8;
     ga_u16x8 px[2];
     auto px_load = [&](int t) {
 #pragma unroll
@@ -99,118 +101,100 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
             if (gi < a.B) {
                 if (a.win != nullptr) {
                     int row = a.win[gi] + t;
-                    v = *(const ga_u16x8*)(a.X + (size_t)row * GA_WMAX + px_c[i]);
+                    v = *(const ga_u16x8*)(a.X + (size_t)row * GA_WMAX + px_c);
                 } else {
-                    v = *(const ga_u16x8*)(a.X + ((size_t)gi * a.T + t) * GA_WMAX + px_c[i]);
+                    v = *(const ga_u16x8*)(a.X + ((size_t)gi * a.T + t) * GA_WMAX + px_c);
                 }
             }
             px[i] = v;
         }
     };
     px_load(0);
-
     __syncthreads();
 
     for (int t = 0; t < a.T; ++t) {
-        // ---- stage x_t tile from the prefetch registers -------------------
+        // ---- stage x_t from the prefetch registers -----------------------
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
             int gi = base + px_m[i];
             if (gi < a.B && a.x_seq_out != nullptr)
-                *(ga_u16x8*)(a.x_seq_out + ((size_t)gi * a.T + t) * GA_WMAX + px_c[i]) = px[i];
-            ga_tile_write8(x_tile, px_m[i], px_c[i], px[i]);
-        }
-        // ---- deferred writer for t-1: 16-B stores of h, c, gates ---------
-        // (overlapped with the x stage: both only read tiles the other
-        // doesn't touch; the barrier below fences phase2's c/g overwrite)
-        if (t > 0 && wgi < a.B) {
-            size_t bt = (size_t)wgi * a.T + (t - 1);
-#pragma unroll
-            for (int half = 0; half < 2; ++half) {
-                int c0 = wc0 + half * 8;
-                *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + c0) =
-                    *(ga_u16x8*)&h_tile[ga_sw(wm, c0)];
-                *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + c0) =
-                    *(ga_u16x8*)&c_tile[ga_sw(wm, c0)];
-#pragma unroll
-                for (int g = 0; g < 4; ++g)
-                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + g * GA_WMAX + c0) =
-                        *(ga_u16x8*)&g_tiles[g * GA_TILE_BYTES + ga_sw(wm, c0)];
-            }
+                *(ga_u16x8*)(a.x_seq_out + ((size_t)gi * a.T + t) * GA_WMAX + px_c) = px[i];
+            ga_tile_write8(x_tile, px_m[i], px_c, px[i]);
         }
         __syncthreads();
         if (t + 1 < a.T) px_load(t + 1);  // overlap next x load with the GEMM
 
-        unsigned short h_stage[4][4];  // deferred h_tile update (see below)
-        // ---- gates GEMM + cell update, one 16-col tile at a time ---------
-        // (per-jt: only the 4 gate acc tiles are live -> 16 f32/lane, which
-        // keeps the kernel under the 128-VGPR budget for 2 waves/SIMD; the
-        // x/h afrag re-reads per jt are cheap LDS hits)
+        // ---- this wave's gate pre-activations: [64 m][64 j] --------------
+        ga_f32x4 acc[4][4];  // [mt][jt], bias pre-added
 #pragma unroll
-        for (int jt = 0; jt < 4; ++jt) {
-            int col = jt * 16 + (lane & 15);
-            ga_f32x4 acc4[4];  // i, f, g, o pre-activations for col tile jt
+        for (int mt = 0; mt < 4; ++mt)
 #pragma unroll
-            for (int g = 0; g < 4; ++g) acc4[g] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
-            for (int kk = 0; kk < LSTM_K; kk += 32) {
-                const char* src = (kk < 64) ? x_tile : h_tile;
-                int klocal = kk & 63;
-                ga_bf16x8 afrag = ga_tile_read8(src, wrow + (lane & 15), klocal + (lane >> 4) * 8);
+            for (int jt = 0; jt < 4; ++jt)
+                acc[mt][jt] = ga_f32x4{bias_reg[jt], bias_reg[jt], bias_reg[jt], bias_reg[jt]};
+        for (int kk = 0; kk < 4; ++kk) {
+            const char* src = (kk < 2) ? x_tile : h_tile;
+            int klocal = (kk & 1) * 32;
 #pragma unroll
-                for (int g = 0; g < 4; ++g) {
-                    // B[k][j] = Wcat[j][k] row-read from the swizzled LDS image
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)&w_lds[ga_swW(
-                        g * GA_WMAX + jt * 16 + (lane & 15), (kk + (lane >> 4) * 8) * 2, 256)];
-                    acc4[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc4[g], 0, 0, 0);
+            for (int mt = 0; mt < 4; ++mt) {
+                ga_bf16x8 afrag = ga_tile_read8(src, mt * 16 + (lane & 15), klocal + (lane >> 4) * 8);
+#pragma unroll
+                for (int jt = 0; jt < 4; ++jt)
+                    acc[mt][jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, wfrag[jt][kk], acc[mt][jt], 0, 0, 0);
+            }
+        }
+        char* my_pre = pre_tiles + wid * GA_TILE_BYTES;
+#pragma unroll
+        for (int mt = 0; mt < 4; ++mt)
+#pragma unroll
+            for (int jt = 0; jt < 4; ++jt) {
+                int col = jt * 16 + (lane & 15);
+#pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    ga_tile_write1(my_pre, mt * 16 + (lane >> 4) * 4 + r, col,
+                                   ga_f32_to_bf16(acc[mt][jt][r]));
+            }
+        __syncthreads();
+
+        // ---- vectorized cell update + direct HBM streaming ---------------
+        {
+            const bool live = egi < a.B;
+            const size_t bt = (size_t)egi * a.T + t;
+#pragma unroll
+            for (int hh = 0; hh < 2; ++hh) {
+                int cc = ec0 + hh * 8;
+                ga_u16x8 vi = *(ga_u16x8*)&pre_tiles[ga_sw(em, cc)];
+                ga_u16x8 vf = *(ga_u16x8*)&pre_tiles[GA_TILE_BYTES + ga_sw(em, cc)];
+                ga_u16x8 vg = *(ga_u16x8*)&pre_tiles[2 * GA_TILE_BYTES + ga_sw(em, cc)];
+                ga_u16x8 vo = *(ga_u16x8*)&pre_tiles[3 * GA_TILE_BYTES + ga_sw(em, cc)];
+                ga_u16x8 oh, oc, oi, of, og, oo;
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    float i_g = ga_sigmoid(ga_bf16_to_f32(vi[e]));
+                    float f_g = ga_sigmoid(ga_bf16_to_f32(vf[e]));
+                    float g_g = ga_tanh(ga_bf16_to_f32(vg[e]));
+                    float o_g = ga_sigmoid(ga_bf16_to_f32(vo[e]));
+                    int ei = hh * 8 + e;
+                    float c_new = f_g * c_reg[ei] + i_g * g_g;
+                    c_reg[ei] = c_new;
+                    oh[e] = ga_f32_to_bf16(o_g * ga_tanh(c_new));
+                    oc[e] = ga_f32_to_bf16(c_new);
+                    oi[e] = ga_f32_to_bf16(i_g);
+                    of[e] = ga_f32_to_bf16(f_g);
+                    og[e] = ga_f32_to_bf16(g_g);
+                    oo[e] = ga_f32_to_bf16(o_g);
+                }
+                *(ga_u16x8*)&h_tile[ga_sw(em, cc)] = oh;
+                if (live) {
+                    *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + cc) = oh;
+                    *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc) = oc;
+                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc) = oi;
+                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc) = of;
+                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc) = og;
+                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc) = oo;
                 }
             }
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int m = wrow + (lane >> 4) * 4 + r;
-                float i_g = ga_sigmoid(acc4[0][r] + s_bias[col]);
-                float f_g = ga_sigmoid(acc4[1][r] + s_bias[GA_WMAX + col]);
-                float g_g = ga_tanh(acc4[2][r] + s_bias[2 * GA_WMAX + col]);
-                float o_g = ga_sigmoid(acc4[3][r] + s_bias[3 * GA_WMAX + col]);
-                float c_new = f_g * c_reg[jt][r] + i_g * g_g;
-                c_reg[jt][r] = c_new;
-                float h_new = o_g * ga_tanh(c_new);
-                // NOTE: h_tile writes must not race the GEMM's h reads of
-                // OTHER waves -> write to c/g tiles now, h to a temp column
-                // in c-order; h_tile is updated after the barrier below.
-                ga_tile_write1(c_tile, m, col, ga_f32_to_bf16(c_new));
-                ga_tile_write1(g_tiles, m, col, ga_f32_to_bf16(i_g));
-                ga_tile_write1(g_tiles + GA_TILE_BYTES, m, col, ga_f32_to_bf16(f_g));
-                ga_tile_write1(g_tiles + 2 * GA_TILE_BYTES, m, col, ga_f32_to_bf16(g_g));
-                ga_tile_write1(g_tiles + 3 * GA_TILE_BYTES, m, col, ga_f32_to_bf16(o_g));
-                h_stage[jt][r] = ga_f32_to_bf16(h_new);
-            }
         }
-        __syncthreads();  // all waves done reading x/h for this t
-#pragma unroll
-        for (int jt = 0; jt < 4; ++jt) {
-            int col = jt * 16 + (lane & 15);
-#pragma unroll
-            for (int r = 0; r < 4; ++r)
-                ga_tile_write1(h_tile, wrow + (lane >> 4) * 4 + r, col, h_stage[jt][r]);
-        }
-        __syncthreads();
-    }
-
-    // trailing writer for t = T-1
-    if (wgi < a.B) {
-        size_t bt = (size_t)wgi * a.T + (a.T - 1);
-#pragma unroll
-        for (int half = 0; half < 2; ++half) {
-            int c0 = wc0 + half * 8;
-            *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + c0) =
-                *(ga_u16x8*)&h_tile[ga_sw(wm, c0)];
-            *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + c0) =
-                *(ga_u16x8*)&c_tile[ga_sw(wm, c0)];
-#pragma unroll
-            for (int g = 0; g < 4; ++g)
-                *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + g * GA_WMAX + c0) =
-                    *(ga_u16x8*)&g_tiles[g * GA_TILE_BYTES + ga_sw(wm, c0)];
-        }
+        __syncthreads();  // h_tile/pre reads-writes settle before next t
     }
 }
 
@@ -595,7 +579,7 @@ void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
     a.T = (int)T;
     a.N = (int)X.size(0);
     int n_tiles = (int)((B + GA_MT - 1) / GA_MT);
-    size_t smem = 7 * GA_TILE_BYTES + LSTM_4H * sizeof(float) + LSTM_4H * LSTM_K * 2;
+    size_t smem = 6 * GA_TILE_BYTES;
     static bool smem_cfg = [] {  // >64 KiB dynamic LDS needs an explicit opt-in
         (void)hipFuncSetAttribute((const void*)lstm_fwd,
                                   hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
