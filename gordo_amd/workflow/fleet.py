@@ -68,7 +68,9 @@ class FleetBuilder:
             n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
         self.n_gpus = n_gpus
         if n_workers is None:
-            n_workers = max(1, n_gpus) if n_gpus else min(8, os.cpu_count() or 1)
+            # CPU workers each re-import torch (~GBs RSS); keep the default
+            # small to avoid OOM-killing the pool on modest hosts
+            n_workers = max(1, n_gpus) if n_gpus else min(2, os.cpu_count() or 1)
         self.n_workers = n_workers
 
     def machine_dir(self, name: str) -> str:
@@ -79,27 +81,47 @@ class FleetBuilder:
         ctx = mp.get_context("spawn")
         results: List[Dict[str, Any]] = []
         t0 = time.time()
-        with ProcessPoolExecutor(max_workers=self.n_workers, mp_context=ctx) as pool:
-            futures = {}
+        done: set = set()
+        try:
+            with ProcessPoolExecutor(max_workers=self.n_workers, mp_context=ctx) as pool:
+                futures = {}
+                for i, machine in enumerate(self.config.machines):
+                    device = f"cuda:{i % self.n_gpus}" if self.n_gpus else "cpu"
+                    fut = pool.submit(
+                        _build_one,
+                        machine.to_dict(),
+                        self.machine_dir(machine.name),
+                        self.registry_dir,
+                        device,
+                    )
+                    futures[fut] = machine.name
+                for fut in as_completed(futures):
+                    res = fut.result()
+                    results.append(res)
+                    done.add(res["machine"])
+                    logger.info(
+                        "fleet build %s: %s (%.1fs)",
+                        res["machine"],
+                        res["status"],
+                        res["seconds"],
+                    )
+        except Exception as e:  # e.g. BrokenProcessPool after a worker OOM
+            logger.warning(
+                "fleet worker pool failed (%s); building remaining machines "
+                "in-process", e,
+            )
             for i, machine in enumerate(self.config.machines):
+                if machine.name in done:
+                    continue
                 device = f"cuda:{i % self.n_gpus}" if self.n_gpus else "cpu"
-                fut = pool.submit(
-                    _build_one,
+                res = _build_one(
                     machine.to_dict(),
                     self.machine_dir(machine.name),
                     self.registry_dir,
                     device,
                 )
-                futures[fut] = machine.name
-            for fut in as_completed(futures):
-                res = fut.result()
                 results.append(res)
-                logger.info(
-                    "fleet build %s: %s (%.1fs)",
-                    res["machine"],
-                    res["status"],
-                    res["seconds"],
-                )
+                done.add(res["machine"])
         elapsed = time.time() - t0
         ok = sum(1 for r in results if r["status"] == "succeeded")
         logger.info(
