@@ -68,6 +68,9 @@ def main():
         from gordo_amd.model.factories.lstm_autoencoder import lstm_hourglass
         from gordo_amd.ops.lstm_engine import LSTMStepTrainer
 
+        if not on_gpu:
+            sys.exit("--model lstm benches the fused HIP cell kernels and needs a GPU")
+
         # 32768 windows/GPU -> 512 blocks = 2 blocks/CU in the cell kernels
         batch = args.batch or 32768
         module = lstm_hourglass(
