@@ -61,6 +61,63 @@ def project_config(n_machines: int, n_tags: int):
     }
 
 
+def measure_http(out_dir: str, args, X):
+    """End-to-end HTTP: uvicorn fleet server in a subprocess, parquet POSTs."""
+    import socket
+    import subprocess
+    import httpx
+    from gordo_amd.server.utils import dataframe_into_parquet_bytes
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    repo = os.path.dirname(os.path.abspath(__file__))
+    env = dict(os.environ, PYTHONPATH=repo + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "gordo_amd.cli.cli", "fleet", "serve",
+         "--models-dir", out_dir, "--project-name", "bench-fleet",
+         "--host", "127.0.0.1", "--port", str(port)],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL, env=env,
+    )
+    try:
+        base = f"http://127.0.0.1:{port}"
+        body = dataframe_into_parquet_bytes(X)
+        with httpx.Client(timeout=30.0) as client:
+            for _ in range(240):  # wait for startup (model loads)
+                try:
+                    if client.get(base + "/healthcheck").status_code == 200:
+                        break
+                except httpx.TransportError:
+                    pass
+                time.sleep(0.5)
+            urls = [
+                f"{base}/gordo/v0/bench-fleet/machine-{i:03d}/anomaly/prediction?format=parquet"
+                for i in range(min(8, args.machines))
+            ]
+            for u in urls:  # warm
+                r = client.post(u, content=body,
+                                headers={"content-type": "application/octet-stream"})
+                assert r.status_code == 200, r.text[:200]
+            lat = []
+            for i in range(min(args.requests, 200)):
+                t = time.perf_counter()
+                r = client.post(urls[i % len(urls)], content=body,
+                                headers={"content-type": "application/octet-stream"})
+                assert r.status_code == 200
+                lat.append(time.perf_counter() - t)
+        lat_ms = np.array(lat) * 1e3
+        return {
+            "http_p50_ms": float(np.percentile(lat_ms, 50)),
+            "http_p99_ms": float(np.percentile(lat_ms, 99)),
+        }
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--machines", type=int, default=32)
@@ -68,6 +125,8 @@ def main():
     ap.add_argument("--requests", type=int, default=500)
     ap.add_argument("--req-batch", type=int, default=1000)
     ap.add_argument("--out-dir", default=None)
+    ap.add_argument("--http", action="store_true",
+                    help="also measure end-to-end HTTP latency through the fleet server")
     args = ap.parse_args()
 
     from gordo_amd.machine import NormalizedConfig
@@ -111,7 +170,12 @@ def main():
     lat_ms = np.array(lat) * 1e3
     fused = sum(1 for rt in runtimes if rt._fused is not None)
 
+    http_stats = {}
+    if args.http:
+        http_stats = measure_http(out_dir, args, X)
+
     print(json.dumps({
+        **http_stats,
         "metric": "anomaly_p50_latency_ms",
         "value": float(np.percentile(lat_ms, 50)),
         "unit": "ms",
