@@ -34,8 +34,8 @@ from ..model.utils import make_base_dataframe
 logger = logging.getLogger(__name__)
 
 
-def build_app(model_dir: str):
-    runtime = ModelRuntime(model_dir)
+def build_app(model_dir: str, device=None):
+    runtime = ModelRuntime(model_dir, device=device)
     app = FastAPI(title="gordo-amd model server", version=__version__)
     app.state.runtime = runtime
 

@@ -29,10 +29,15 @@ def build_fleet_app(models_dir: str, project_name: str = "project",
         )
     app = FastAPI(title=f"gordo-amd fleet server [{project_name}]", version=__version__)
     mounted: Dict[str, str] = {}
-    for name in machines:
+    # shard resident models round-robin across the node's GPUs
+    import torch
+
+    n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    for i, name in enumerate(machines):
         model_dir = os.path.join(models_dir, name)
+        device = f"cuda:{i % n_gpus}" if n_gpus else None
         try:
-            sub = build_app(model_dir)
+            sub = build_app(model_dir, device=device)
         except Exception:
             logger.exception("failed to load model %s", name)
             continue

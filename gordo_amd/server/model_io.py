@@ -9,6 +9,7 @@ error → row L2) runs as ONE fused HIP kernel (SURVEY.md §3.2 / K1+K7+K8).
 from __future__ import annotations
 
 import logging
+import os
 from typing import Any, Dict, Optional
 
 import numpy as np
@@ -22,8 +23,13 @@ logger = logging.getLogger(__name__)
 
 
 class ModelRuntime:
-    def __init__(self, model_dir: str):
+    def __init__(self, model_dir: str, device: Optional[str] = None):
         self.model_dir = model_dir
+        self.device = torch.device(
+            device
+            or os.environ.get("GORDO_DEVICE")
+            or ("cuda:0" if torch.cuda.is_available() else "cpu")
+        )
         self.model = load(model_dir)
         try:
             self.metadata: Dict[str, Any] = load_metadata(model_dir)
@@ -46,7 +52,7 @@ class ModelRuntime:
             est, scaler_in, err_scaler = self._unwrap()
             if est is None or est.model is None or supported(est.model) is not None:
                 return
-            fwd = FusedHourglassForward(est.model, torch.device("cuda:0"))
+            fwd = FusedHourglassForward(est.model, self.device)
             if scaler_in is not None:
                 # MinMaxScaler: x' = x*scale_ + min_
                 fwd.set_input_scaler(scaler_in.scale_, scaler_in.min_)
