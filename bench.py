@@ -81,7 +81,7 @@ def main():
         X = synthetic_sensor_data(rows, args.tags, seed=1000 + rank)
         cfg = TrainConfig(batch_size=batch, lr=1e-3, dtype="bf16")
         trainer = LSTMStepTrainer(module, X, batch, cfg, dp=dp)
-        model_name = "lstm_hourglass-50x" + "x".join(str(d) for d in module.dims)
+        model_name = f"lstm_hourglass-{args.tags}x" + "x".join(str(d) for d in module.dims)
         seq_len = args.lookback
         engine = "fused_hip_lstm"
     else:
@@ -92,7 +92,7 @@ def main():
         X = synthetic_sensor_data(DATASET_ROWS, args.tags, seed=1000 + rank)
         cfg = TrainConfig(batch_size=batch, lr=1e-3, dtype="bf16")
         trainer = HourglassStepTrainer(module, X, batch, cfg, dp=dp)
-        model_name = "feedforward_hourglass-50x38x25x38x50"
+        model_name = "feedforward_hourglass-" + "x".join(str(w) for w in module.layer_widths)
         seq_len = 1
         engine = "fused_hip" if trainer.fused else "eager"
 

@@ -9,13 +9,13 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
                           torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
                           torch::Tensor loss, int64_t B, int64_t F_out,
                           std::vector<int64_t> acts, double gscale, bool y_is_x,
-                          c10::optional<torch::Tensor> dx_out);
+                          c10::optional<torch::Tensor> dx_out, int64_t wmax);
 
 void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                     torch::Tensor dw, torch::Tensor db, torch::Tensor w,
                     torch::Tensor wt, torch::Tensor bias, torch::Tensor counters,
                     int64_t n_slabs, int64_t L, double lr, double beta1, double beta2,
-                    double eps);
+                    double eps, int64_t wmax);
 
 void hg_bump_launch(torch::Tensor counters, int64_t n_batches);
 
@@ -23,26 +23,30 @@ void hg_forward_launch(torch::Tensor X, torch::Tensor w, torch::Tensor bias,
                        torch::Tensor xscale, torch::Tensor xshift, torch::Tensor yhat,
                        c10::optional<torch::Tensor> err, c10::optional<torch::Tensor> total,
                        c10::optional<torch::Tensor> inv_err_scale,
-                       int64_t F_in, int64_t F_out, std::vector<int64_t> acts);
+                       int64_t F_in, int64_t F_out, std::vector<int64_t> acts,
+                       int64_t wmax);
 
 void hg_grad_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,
-                           int64_t n_slabs, int64_t L);
+                           int64_t n_slabs, int64_t L, int64_t wmax);
 
 void hg_adam_flat_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                          torch::Tensor g, torch::Tensor w, torch::Tensor wt,
                          torch::Tensor bias, torch::Tensor counters, int64_t L,
-                         double lr, double beta1, double beta2, double eps);
+                         double lr, double beta1, double beta2, double eps,
+                         int64_t wmax);
 
 void hg_slab_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor partial,
                            c10::optional<torch::Tensor> counters, int64_t n_slabs,
-                           int64_t G, int64_t L, int64_t n_batches);
+                           int64_t G, int64_t L, int64_t n_batches, int64_t wmax);
 
 void hg_adam_p_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                       torch::Tensor partial, torch::Tensor w, torch::Tensor wt,
                       torch::Tensor bias, torch::Tensor counters, int64_t G,
-                      int64_t L, double lr, double beta1, double beta2, double eps);
+                      int64_t L, double lr, double beta1, double beta2, double eps,
+                      int64_t wmax);
 
-void hg_partial_flat_launch(torch::Tensor partial, torch::Tensor g, int64_t G, int64_t L);
+void hg_partial_flat_launch(torch::Tensor partial, torch::Tensor g, int64_t G,
+                            int64_t L, int64_t wmax);
 
 void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
                      torch::Tensor wcat, torch::Tensor bias, torch::Tensor h_seq,

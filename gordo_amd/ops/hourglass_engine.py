@@ -20,16 +20,22 @@ from ..runtime.trainer import TrainHistory
 
 logger = logging.getLogger(__name__)
 
-WMAX = 64
+WMAX = 64         # narrow-variant padded width (also the LSTM head width)
+WMAX_WIDE = 128   # wide-variant padded width (> this -> library-GEMM fallback)
 MAXL = 5
 ACT_IDS = {"linear": 0, None: 0, "tanh": 1, "relu": 2, "sigmoid": 3}
+
+
+def wmax_for(widths) -> int:
+    """Padded kernel width for a stack of layer widths."""
+    return WMAX if max(widths) <= WMAX else WMAX_WIDE
 
 
 def supported(module) -> Optional[str]:
     """None if the fused path supports this FeedForwardNet, else the reason."""
     widths = module.layer_widths
-    if max(widths) > WMAX:
-        return f"width {max(widths)} > {WMAX}"
+    if max(widths) > WMAX_WIDE:
+        return f"width {max(widths)} > {WMAX_WIDE}"
     L = len(widths) - 1
     if L < 1 or L > MAXL:
         return f"L={L} outside 1..{MAXL}"
@@ -47,9 +53,11 @@ class _PackedParams:
         self.L = len(self.widths) - 1
         self.act_ids: List[int] = [ACT_IDS[f] for f in module.layer_funcs]
         self.device = device
-        PW = self.L * WMAX * WMAX
+        WM = wmax_for(self.widths)
+        self.wmax = WM
+        PW = self.L * WM * WM
         self.PW = PW
-        self.P = PW + self.L * WMAX
+        self.P = PW + self.L * WM
 
         master = torch.zeros(self.P, dtype=torch.float32, device=device)
         linears = [m for m in module.net if isinstance(m, torch.nn.Linear)]
@@ -57,20 +65,21 @@ class _PackedParams:
         self.linears = linears
         for i, lin in enumerate(linears):
             out_w, in_w = lin.weight.shape
-            wpad = master[i * WMAX * WMAX : (i + 1) * WMAX * WMAX].view(WMAX, WMAX)
+            wpad = master[i * WM * WM : (i + 1) * WM * WM].view(WM, WM)
             wpad[:out_w, :in_w] = lin.weight.detach().to(device, torch.float32)
-            bpad = master[PW + i * WMAX : PW + (i + 1) * WMAX]
+            bpad = master[PW + i * WM : PW + (i + 1) * WM]
             bpad[:out_w] = lin.bias.detach().to(device, torch.float32)
         self.master = master
         self.bias = master[PW:].clone()  # fp32 bias view used by kernels
-        w = master[:PW].view(self.L, WMAX, WMAX)
+        w = master[:PW].view(self.L, WM, WM)
         self.w_bf16 = w.to(torch.bfloat16).contiguous()
         self.wt_bf16 = w.transpose(1, 2).contiguous().to(torch.bfloat16).contiguous()
 
     def sync_to_module(self):
         """Copy trained padded weights back into the torch module (fp32)."""
-        w = self.master[: self.PW].view(self.L, WMAX, WMAX)
-        b = self.master[self.PW :].view(self.L, WMAX)
+        WM = self.wmax
+        w = self.master[: self.PW].view(self.L, WM, WM)
+        b = self.master[self.PW :].view(self.L, WM)
         with torch.no_grad():
             for i, lin in enumerate(self.linears):
                 out_w, in_w = lin.weight.shape
@@ -78,10 +87,10 @@ class _PackedParams:
                 lin.bias.copy_(b[i, :out_w].to(lin.bias.device))
 
 
-def pad_rows_bf16(X: np.ndarray, device) -> torch.Tensor:
-    """[N,F] fp32 → [N,64] bf16 device tensor, zero-padded."""
+def pad_rows_bf16(X: np.ndarray, device, wmax: int = WMAX) -> torch.Tensor:
+    """[N,F] fp32 → [N,wmax] bf16 device tensor, zero-padded."""
     N, F = X.shape
-    t = torch.zeros(N, WMAX, dtype=torch.bfloat16, device=device)
+    t = torch.zeros(N, wmax, dtype=torch.bfloat16, device=device)
     t[:, :F] = torch.as_tensor(X, device=device, dtype=torch.float32).to(torch.bfloat16)
     return t
 
@@ -109,8 +118,9 @@ class FusedHourglassTrainer:
         L, F_in, F_out = p.L, p.widths[0], p.widths[-1]
 
         y_is_x = y is X or (X.shape == y.shape and np.shares_memory(X, y))
-        Xd = pad_rows_bf16(np.asarray(X, np.float32), device)
-        Yd = Xd if y_is_x else pad_rows_bf16(np.asarray(y, np.float32), device)
+        WM = p.wmax
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), device, WM)
+        Yd = Xd if y_is_x else pad_rows_bf16(np.asarray(y, np.float32), device, WM)
 
         N = Xd.shape[0]
         B = max(64, min(cfg.batch_size, N))
@@ -118,8 +128,8 @@ class FusedHourglassTrainer:
         n_tiles = (B + MT - 1) // MT
         n_batches = max(1, math.ceil(N / B))
 
-        dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=device)
-        db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=device)
+        dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=device)
+        db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=device)
         loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
         # counters [batch idx, adam t]: hg_slab_reduce pre-bumps, so t starts 0
         counters = torch.tensor([0, 0], dtype=torch.int32, device=device)
@@ -143,13 +153,14 @@ class FusedHourglassTrainer:
         def one_step():
             self.E.hg_train_step(
                 Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                loss_acc, B, F_out, acts, gscale, bool(y_is_x), None,
+                loss_acc, B, F_out, acts, gscale, bool(y_is_x), None, WM,
             )
             # two-stage slab reduce (counter bump folded into stage A)
-            self.E.hg_slab_reduce(dw, db, partial, counters, n_tiles, G, L, n_batches)
+            self.E.hg_slab_reduce(dw, db, partial, counters, n_tiles, G, L,
+                                  n_batches, WM)
             self.E.hg_adam_p(
                 p.master, m, v, partial, p.w_bf16, p.wt_bf16, p.bias, counters,
-                G, L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+                G, L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
 
         # ---- optional hipGraph capture of the whole step -----------------
@@ -221,14 +232,15 @@ class FusedHourglassForward:
         self.p = _PackedParams(module, device)
         self.F_in = self.p.widths[0]
         self.F_out = self.p.widths[-1]
-        self.identity_scale = torch.zeros(WMAX, dtype=torch.float32, device=device)
+        WM = self.p.wmax
+        self.identity_scale = torch.zeros(WM, dtype=torch.float32, device=device)
         self.identity_scale[: self.F_in] = 1.0
-        self.zero_shift = torch.zeros(WMAX, dtype=torch.float32, device=device)
+        self.zero_shift = torch.zeros(WM, dtype=torch.float32, device=device)
 
     def set_input_scaler(self, scale: np.ndarray, shift: np.ndarray):
         """x_scaled = x*scale + shift (fused MinMaxScaler)."""
-        s = torch.zeros(WMAX, dtype=torch.float32, device=self.device)
-        t = torch.zeros(WMAX, dtype=torch.float32, device=self.device)
+        s = torch.zeros(self.p.wmax, dtype=torch.float32, device=self.device)
+        t = torch.zeros(self.p.wmax, dtype=torch.float32, device=self.device)
         s[: len(scale)] = torch.as_tensor(scale, dtype=torch.float32, device=self.device)
         t[: len(shift)] = torch.as_tensor(shift, dtype=torch.float32, device=self.device)
         self.identity_scale = s
@@ -240,6 +252,7 @@ class FusedHourglassForward:
         self.E.hg_forward(
             X, self.p.w_bf16, self.p.bias, self.identity_scale, self.zero_shift,
             yhat, None, None, None, self.F_in, self.F_out, self.p.act_ids,
+            self.p.wmax,
         )
         return yhat
 
@@ -250,10 +263,11 @@ class FusedHourglassForward:
         yhat = torch.empty(B, self.F_out, dtype=torch.float32, device=self.device)
         err = torch.empty(B, self.F_out, dtype=torch.float32, device=self.device)
         total = torch.empty(B, dtype=torch.float32, device=self.device)
-        inv = torch.zeros(WMAX, dtype=torch.float32, device=self.device)
+        inv = torch.zeros(self.p.wmax, dtype=torch.float32, device=self.device)
         inv[: self.F_out] = torch.as_tensor(inv_err_scale, dtype=torch.float32, device=self.device)
         self.E.hg_forward(
             X, self.p.w_bf16, self.p.bias, self.identity_scale, self.zero_shift,
             yhat, err, total, inv, self.F_in, self.F_out, self.p.act_ids,
+            self.p.wmax,
         )
         return yhat, err, total

@@ -274,7 +274,7 @@ class FusedLSTMTrainer:
             st["h_last"], st["y_head"], st["perm_id"], st["counters0"],
             p.head_w, p.head_wt, p.head_bias, st["head_dw"], st["head_db"],
             st["loss_acc"], B, st["F_out"], st["act_out"], st["gscale"], False,
-            st["dx_head"],
+            st["dx_head"], WMAX,
         )
 
         # ---- BPTT (fused dx) + split-K dW kernel, top layer down ---------

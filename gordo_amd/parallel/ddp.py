@@ -112,21 +112,22 @@ class HourglassStepTrainer:
         self.E = ext()
         device = self.device
         p = _PackedParams(self.module, device)
+        WM = p.wmax
         # ranks must start from identical weights
         self.dp.broadcast(p.master)
         p.bias.copy_(p.master[p.PW:])
-        w = p.master[: p.PW].view(p.L, WMAX, WMAX)
+        w = p.master[: p.PW].view(p.L, WM, WM)
         p.w_bf16.copy_(w.to(torch.bfloat16))
         p.wt_bf16.copy_(w.transpose(1, 2).to(torch.bfloat16))
         self.p = p
         self.F_out = p.widths[-1]
-        self.Xd = pad_rows_bf16(np.asarray(self.X, np.float32), device)
+        self.Xd = pad_rows_bf16(np.asarray(self.X, np.float32), device, WM)
         N = self.Xd.shape[0]
         self.N = N
         MT = int(getattr(self.E, "MT", 32))
         self.n_tiles = (self.B + MT - 1) // MT
-        self.dw = torch.zeros(self.n_tiles, p.L, WMAX * WMAX, dtype=torch.float32, device=device)
-        self.db = torch.zeros(self.n_tiles, p.L, WMAX, dtype=torch.float32, device=device)
+        self.dw = torch.zeros(self.n_tiles, p.L, WM * WM, dtype=torch.float32, device=device)
+        self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.float32, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
         # counters [batch idx, adam t]: hg_slab_reduce pre-bumps, so t starts 0
@@ -167,8 +168,8 @@ class HourglassStepTrainer:
             self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
             self.counters.copy_(state[3])
             self.p.bias.copy_(self.p.master[self.p.PW:])
-            from ..ops.hourglass_engine import WMAX
-            w = self.p.master[: self.p.PW].view(self.p.L, WMAX, WMAX)
+            WM = self.p.wmax
+            w = self.p.master[: self.p.PW].view(self.p.L, WM, WM)
             self.p.w_bf16.copy_(w.to(torch.bfloat16))
             self.p.wt_bf16.copy_(w.transpose(1, 2).contiguous().to(torch.bfloat16))
             self.loss_acc.zero_()
@@ -199,28 +200,29 @@ class HourglassStepTrainer:
     def _fused_step_raw(self):
         p = self.p
         cfg = self.cfg
+        WM = p.wmax
         self.E.hg_train_step(
             self.Xd, self.Xd, self.perm, self.counters, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
-            self.gscale, True, None,
+            self.gscale, True, None, WM,
         )
         # two-stage slab reduce; stage A also bumps the counters
         self.E.hg_slab_reduce(
             self.dw, self.db, self.partial, self.counters, self.n_tiles,
-            self.G, p.L, self.n_batches,
+            self.G, p.L, self.n_batches, WM,
         )
         if self.dp.enabled:
-            self.E.hg_partial_flat(self.partial, self.gflat, self.G, p.L)
+            self.E.hg_partial_flat(self.partial, self.gflat, self.G, p.L, WM)
             self.dp.all_reduce_mean(self.gflat)
             self.E.hg_adam_flat(
                 p.master, self.m, self.v, self.gflat, p.w_bf16, p.wt_bf16, p.bias,
-                self.counters, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+                self.counters, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
         else:
             self.E.hg_adam_p(
                 p.master, self.m, self.v, self.partial, p.w_bf16, p.wt_bf16,
                 p.bias, self.counters, self.G, p.L,
-                cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps,
+                cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
 
     # ------------------------------------------------------------------

@@ -82,3 +82,81 @@ def test_fused_vs_eager_training_quality():
     ev_eager = run("cuda:0", False)
     assert ev_fused > 0.5, ev_fused
     assert ev_fused > ev_eager - 0.1, (ev_fused, ev_eager)
+
+
+# ---------------------------------------------------------------------------
+# wide variant (W=128): tag sets above 64 use the 128-padded kernels
+# ---------------------------------------------------------------------------
+def test_fused_forward_wide():
+    torch.manual_seed(2)
+    net = FeedForwardNet(128, 128, [96, 64, 96], ["tanh"] * 3, "linear")
+    X = np.random.default_rng(3).normal(size=(211, 128)).astype(np.float32)
+    ref = net(torch.as_tensor(X)).detach().numpy()
+    out = FusedHourglassForward(net, DEV).forward(torch.as_tensor(X)).cpu().numpy()
+    np.testing.assert_allclose(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_fused_training_wide_matches_autograd():
+    """Wide (W=128) grads vs fp32 autograd on one step."""
+    from gordo_amd.ops import ext
+    from gordo_amd.ops.hourglass_engine import _PackedParams, pad_rows_bf16
+
+    torch.manual_seed(3)
+    net = FeedForwardNet(100, 100, [72], ["tanh"], "linear")
+    ref_net = FeedForwardNet(100, 100, [72], ["tanh"], "linear")
+    ref_net.load_state_dict(net.state_dict())
+    X = np.random.default_rng(4).normal(size=(256, 100)).astype(np.float32)
+
+    E = ext()
+    p = _PackedParams(net, DEV)
+    assert p.wmax == 128
+    WM, L, F_out = p.wmax, p.L, 100
+    B = 256
+    MT = int(E.MT)
+    n_tiles = (B + MT - 1) // MT
+    Xd = pad_rows_bf16(X, DEV, WM)
+    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
+    db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+    loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+    counters = torch.tensor([0, 1], dtype=torch.int32, device=DEV)
+    perm = torch.arange(B, dtype=torch.int32, device=DEV)
+    gscale = 2.0 / (B * F_out)
+    E.hg_train_step(Xd, Xd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, True, None, WM)
+    torch.cuda.synchronize()
+
+    Xref = Xd[:, :100].float().cpu()
+    out = ref_net(Xref)
+    loss_ref = torch.nn.functional.mse_loss(out, Xref)
+    loss_ref.backward()
+    kernel_loss = float(loss.item()) / (B * F_out)
+    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+
+    linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
+    dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
+    db_sum = db.sum(dim=0).view(L, WM).cpu()
+    for i, lin in enumerate(linears):
+        o, ninp = lin.weight.shape
+        got_w = dw_sum[i, :ninp, :o].T  # kernel stores [in][out]
+        ref_w = lin.weight.grad.detach()
+        scale = ref_w.abs().max().item() or 1.0
+        assert (got_w - ref_w).abs().max().item() < 0.05 * scale + 1e-4, f"layer {i} dW"
+        got_b = db_sum[i, :o]
+        ref_b = lin.bias.grad.detach()
+        bscale = ref_b.abs().max().item() or 1.0
+        assert (got_b - ref_b).abs().max().item() < 0.05 * bscale + 1e-4, f"layer {i} db"
+        assert dw_sum[i, ninp:, :].abs().max().item() == 0.0
+        assert dw_sum[i, :, o:].abs().max().item() == 0.0
+
+
+def test_fused_trainer_wide_learns():
+    rng = np.random.default_rng(5)
+    Z = rng.normal(size=(4096, 12)).astype(np.float32)
+    P = rng.normal(size=(12, 100)).astype(np.float32)
+    X = np.tanh(Z @ P)
+    torch.manual_seed(5)
+    net = FeedForwardNet(100, 100, [72, 48, 72], ["tanh"] * 3, "linear")
+    cfg = TrainConfig(epochs=20, batch_size=1024, lr=3e-3, seed=0)
+    hist = FusedHourglassTrainer(net, DEV, cfg).fit(X, X)
+    assert hist.engine.startswith("fused_hip")
+    assert hist.loss[-1] < hist.loss[0] * 0.7, hist.loss

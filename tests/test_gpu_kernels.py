@@ -109,7 +109,7 @@ def test_train_step_gradients_match_autograd():
     perm = torch.arange(N, dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
     E.hg_train_step(Xd, Xd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                    loss, B, F_out, p.act_ids, gscale, True, None)
+                    loss, B, F_out, p.act_ids, gscale, True, None, WMAX)
     torch.cuda.synchronize()
 
     # --- torch reference on the same bf16-rounded input ------------------
@@ -144,7 +144,7 @@ def test_train_step_gradients_match_autograd():
     m_opt = torch.zeros_like(p.master)
     v_opt = torch.zeros_like(p.master)
     E.hg_adam(p.master, m_opt, v_opt, dw, db, p.w_bf16, p.wt_bf16, p.bias,
-              counters, n_tiles, L, lr, b1, b2, eps)
+              counters, n_tiles, L, lr, b1, b2, eps, WMAX)
     torch.cuda.synchronize()
 
     # manual reference on the kernel's own summed grads (layout-adjusted)
