@@ -160,3 +160,47 @@ def test_fused_trainer_wide_learns():
     hist = FusedHourglassTrainer(net, DEV, cfg).fit(X, X)
     assert hist.engine.startswith("fused_hip")
     assert hist.loss[-1] < hist.loss[0] * 0.7, hist.loss
+
+
+def test_fused_training_y_not_x():
+    """Fused trainer with a distinct regression target (y != X), vs autograd."""
+    from gordo_amd.ops import ext
+    from gordo_amd.ops.hourglass_engine import WMAX, _PackedParams, pad_rows_bf16
+
+    torch.manual_seed(9)
+    net = FeedForwardNet(30, 10, [20], ["tanh"], "linear")
+    ref_net = FeedForwardNet(30, 10, [20], ["tanh"], "linear")
+    ref_net.load_state_dict(net.state_dict())
+    rng = np.random.default_rng(9)
+    X = rng.normal(size=(256, 30)).astype(np.float32)
+    Y = rng.normal(size=(256, 10)).astype(np.float32)
+
+    E = ext()
+    p = _PackedParams(net, DEV)
+    B, L, F_out = 256, p.L, 10
+    MT = int(E.MT)
+    n_tiles = (B + MT - 1) // MT
+    Xd = pad_rows_bf16(X, DEV)
+    Yd = pad_rows_bf16(Y, DEV)
+    dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=DEV)
+    db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=DEV)
+    loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+    counters = torch.tensor([0, 1], dtype=torch.int32, device=DEV)
+    perm = torch.arange(B, dtype=torch.int32, device=DEV)
+    gscale = 2.0 / (B * F_out)
+    E.hg_train_step(Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, False, None, WMAX)
+    torch.cuda.synchronize()
+
+    Xr = Xd[:, :30].float().cpu()
+    Yr = Yd[:, :10].float().cpu()
+    out = ref_net(Xr)
+    loss_ref = torch.nn.functional.mse_loss(out, Yr)
+    loss_ref.backward()
+    kernel_loss = float(loss.item()) / (B * F_out)
+    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+    lin0 = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)][0]
+    got = dw.sum(0).view(L, WMAX, WMAX)[0, :30, :20].T.cpu()
+    ref = lin0.weight.grad.detach()
+    s = ref.abs().max().item() or 1.0
+    assert (got - ref).abs().max().item() < 0.05 * s + 1e-4

@@ -122,3 +122,60 @@ def test_parquet_roundtrip_multiindex():
     out = dataframe_from_parquet_bytes(dataframe_into_parquet_bytes(df))
     assert list(out.columns) == list(df.columns)
     np.testing.assert_allclose(out.values, df.values)
+
+
+def test_lstm_model_serving_offsets(tmp_path):
+    """An LSTM artifact serves /prediction with lookback-aligned rows."""
+    import numpy as np
+    import pandas as pd
+    from fastapi.testclient import TestClient
+
+    from gordo_amd.builder import ModelBuilder
+    from gordo_amd.machine import Machine
+    from gordo_amd.server.app import build_app
+
+    config = {
+        "name": "lstm-serving",
+        "dataset": {
+            "type": "RandomDataset",
+            "train_start_date": "2020-01-01T00:00:00Z",
+            "train_end_date": "2020-01-08T00:00:00Z",
+            "tag_list": [f"tag-{i}" for i in range(5)],
+        },
+        "model": {
+            "gordo_amd.model.models.TorchLSTMAutoEncoder": {
+                "kind": "lstm_symmetric",
+                "dims": [8],
+                "lookback_window": 8,
+                "epochs": 1,
+                "batch_size": 64,
+            }
+        },
+        "evaluation": {"cv_mode": "no_cv"},
+    }
+    out = tmp_path / "lstm-serving"
+    ModelBuilder(Machine.from_config(config)).build(output_dir=str(out))
+    client = TestClient(build_app(str(out)))
+
+    meta = client.get("/metadata").json()
+    assert meta["metadata"]["model"]["model-offset"] == 7
+
+    rng = np.random.default_rng(0)
+    idx = pd.date_range("2021-01-01", periods=50, freq="10min")
+    payload = {
+        "X": {
+            "data": rng.uniform(0, 1, size=(50, 5)).tolist(),
+            "columns": [f"tag-{i}" for i in range(5)],
+            "index": [str(t) for t in idx],
+        }
+    }
+    r = client.post("/prediction", json=payload)
+    assert r.status_code == 200, r.text
+    body = r.json()
+    # 50 input rows, lookback 8 -> 43 aligned output rows
+    assert len(body["index"]) == 43, len(body["index"])
+    out_cols = body["data"]["model-output"]
+    assert len(out_cols) == 5
+    assert all(len(v) == 43 for v in out_cols.values())
+    # index aligned to the window ends (first prediction at row 7)
+    assert body["index"][0].startswith("2021-01-01T01:10")  # row 7 of a 10min index
