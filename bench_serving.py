@@ -125,6 +125,8 @@ def main():
     ap.add_argument("--requests", type=int, default=500)
     ap.add_argument("--req-batch", type=int, default=1000)
     ap.add_argument("--out-dir", default=None)
+    ap.add_argument("--workers", type=int, default=None,
+                    help="builder workers (default: one per GPU)")
     ap.add_argument("--http", action="store_true",
                     help="also measure end-to-end HTTP latency through the fleet server")
     args = ap.parse_args()
@@ -137,7 +139,7 @@ def main():
     cfg = NormalizedConfig(project_config(args.machines, args.tags))
 
     t0 = time.perf_counter()
-    builder = FleetBuilder(cfg, output_dir=out_dir)
+    builder = FleetBuilder(cfg, output_dir=out_dir, n_workers=args.workers)
     results = builder.build_all()
     build_s = time.perf_counter() - t0
     ok = sum(1 for r in results if r["status"] == "succeeded")
