@@ -186,6 +186,7 @@ class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
                 f"Need at least lookback_window={lookback} rows, got {len(X)}"
             )
         device = resolve_device(self.kwargs.get("device", "auto"))
+        use_wide = False
         if device.type == "cuda":
             from ..ops import is_available
             from ..ops.lstm_engine import FusedLSTMForward, lstm_supported
@@ -193,6 +194,7 @@ class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
             if is_available() and lstm_supported(self.model) is None:
                 fwd = FusedLSTMForward(self.model, device)
                 return fwd.predict(X, lookahead=self.lookahead)
+            use_wide = True  # GEMM-loop forward instead of nn.LSTM eager
         model = self.model.to(device)
         model.eval()
         Xt = torch.as_tensor(X, device=device)
@@ -203,7 +205,13 @@ class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
             for s in range(0, n_windows, batch_size):
                 idx = torch.arange(s, min(s + batch_size, n_windows), device=device)
                 w = Xt[idx.unsqueeze(1) + offsets.unsqueeze(0)]
-                outs.append(model(w).float().cpu().numpy())
+                if use_wide:
+                    from ..ops.wide_lstm import wide_forward
+
+                    out = wide_forward(model, w)
+                else:
+                    out = model(w)
+                outs.append(out.float().cpu().numpy())
         return np.concatenate(outs, axis=0)
 
     def predict(self, X) -> np.ndarray:

@@ -36,5 +36,9 @@ def maybe_fused_trainer(module, device: torch.device, cfg) -> Optional[object]:
             return FusedLSTMTrainer(module, device, cfg)
         except Exception as e:
             logger.warning("Fused LSTM engine rejected module: %s", e)
-            return None
+        # widths beyond the HIP cell envelope: semi-fused GEMM-loop trainer
+        # (vastly faster than torch.nn.LSTM's eager path on ROCm)
+        from .wide_lstm import WideLSTMTrainer
+
+        return WideLSTMTrainer(module, device, cfg)
     return None

@@ -119,3 +119,35 @@ def test_lstm_too_few_rows_raises():
     model = TorchLSTMAutoEncoder(kind="lstm_hourglass", lookback_window=8, epochs=1)
     with pytest.raises(ValueError):
         model.fit(X)
+
+
+def test_wide_lstm_gemmloop_matches_module():
+    """wide_forward (the GEMM-loop LSTM) must equal nn.LSTM numerically."""
+    import torch
+
+    from gordo_amd.model.torch_nets import LSTMNet
+    from gordo_amd.ops.wide_lstm import wide_forward
+
+    torch.manual_seed(0)
+    net = LSTMNet(12, 12, [20, 8], [None, None], "linear", lookback_window=9)
+    x = torch.randn(7, 9, 12)
+    ref = net(x)
+    out = wide_forward(net, x)
+    assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
+
+
+def test_wide_lstm_trainer_cpu():
+    import numpy as np
+    import torch
+
+    from gordo_amd.model.torch_nets import LSTMNet
+    from gordo_amd.ops.wide_lstm import WideLSTMTrainer
+    from gordo_amd.runtime.trainer import TrainConfig
+
+    torch.manual_seed(1)
+    net = LSTMNet(6, 6, [10], [None], "linear", lookback_window=5)
+    X = np.random.default_rng(0).normal(size=(200, 6)).astype(np.float32)
+    cfg = TrainConfig(epochs=2, batch_size=64, device="cpu", seed=0)
+    hist = WideLSTMTrainer(net, torch.device("cpu"), cfg).fit(X, X, lookback=5)
+    assert hist.engine == "wide_lstm_gemmloop"
+    assert len(hist.loss) == 2 and np.isfinite(hist.loss).all()
