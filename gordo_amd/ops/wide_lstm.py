@@ -1,16 +1,20 @@
-"""Semi-fused wide-LSTM engine (widths beyond the hand-written kernels'
-128-feature envelope... in practice any width the HIP cell kernels reject).
+"""Semi-fused wide-LSTM engine (widths the hand-written HIP cells don't
+cover; any hidden size).
 
-torch.nn.LSTM's eager path on ROCm measured ~4K windows/s at hidden 128
-(launch-storm-bound); this runs each layer's time loop as ONE rocBLAS GEMM
-per step on the concatenated [x_t | h] input against the packed 4-gate
-weight, with autograd providing BPTT. Library GEMMs are efficient at these
-widths (the hand-written MFMA kernels matter most at tiny widths), so this
-is the right MI355X shape for wide models: ~2 orders of magnitude over
-nn.LSTM while keeping exact torch semantics.
+torch.nn.LSTM's eager path on ROCm measured ~4K windows/s at hidden 128 —
+with a 144-step lookback the launch count per batch (thousands) dominates.
+This engine removes both sources of overhead:
+
+1. the input projection for ALL timesteps runs as ONE rocBLAS GEMM per layer
+   (x [B·T, I] @ W_ih^T), leaving one [B,H]x[H,4H] addmm + the fused-ish
+   pointwise cell per step;
+2. the whole forward+backward is captured in a hipGraph once per fit and
+   replayed per batch (static shapes; indices gathered into a static input
+   buffer), so the 144-step loop costs zero launch overhead after capture.
 
 Weights stay the canonical torch nn.LSTM parameters (state_dict identical;
-serializer artifacts unchanged); the packed views are rebuilt per fit.
+serializer artifacts unchanged). Library GEMMs are efficient at these widths
+— the hand-written MFMA kernels matter most at tiny widths.
 """
 from __future__ import annotations
 
@@ -28,24 +32,26 @@ logger = logging.getLogger(__name__)
 
 
 def _packed_weights(module, device, dtype):
-    """Per layer: WcatT [I+H, 4H] and bias [4H] built FROM the live torch
-    parameters (so autograd grads flow back into the module)."""
+    """Per layer: (W_ih^T [I,4H], W_hh^T [H,4H], bias [4H]) built FROM the
+    live torch parameters (autograd grads flow back into the module)."""
     packed = []
     for cell in module.cells:
-        wcat_t = torch.cat([cell.weight_ih_l0, cell.weight_hh_l0], dim=1).t()
-        b = cell.bias_ih_l0 + cell.bias_hh_l0
-        packed.append((wcat_t.to(device=device, dtype=dtype), b.to(device=device, dtype=dtype)))
+        w_ih_t = cell.weight_ih_l0.t().to(device=device, dtype=dtype)
+        w_hh_t = cell.weight_hh_l0.t().to(device=device, dtype=dtype)
+        b = (cell.bias_ih_l0 + cell.bias_hh_l0).to(device=device, dtype=dtype)
+        packed.append((w_ih_t, w_hh_t, b))
     return packed
 
 
-def _lstm_layer(x, wcat_t, b, H):
-    """x [B,T,I] -> h_seq [B,T,H] via one addmm per timestep."""
-    B, T, _ = x.shape
+def _lstm_layer(x, w_ih_t, w_hh_t, b, H):
+    """x [B,T,I] -> h_seq [B,T,H]; one big x-projection GEMM + per-t addmm."""
+    B, T, I = x.shape
+    gx = torch.addmm(b, x.reshape(-1, I), w_ih_t).view(B, T, 4 * H)
     h = x.new_zeros(B, H)
     c = x.new_zeros(B, H)
     outs = []
     for t in range(T):
-        gates = torch.addmm(b, torch.cat([x[:, t, :], h], dim=1), wcat_t)
+        gates = torch.addmm(gx[:, t], h, w_hh_t)
         i, f, g, o = gates.split(H, dim=1)
         i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
         g = torch.tanh(g)
@@ -57,11 +63,10 @@ def _lstm_layer(x, wcat_t, b, H):
 
 def wide_forward(module, x):
     """Full stacked forward: windows [B,T,F] -> head output [B,F_out]."""
-    dtype = x.dtype
-    packed = _packed_weights(module, x.device, dtype)
+    packed = _packed_weights(module, x.device, x.dtype)
     h = x
-    for (wcat_t, b), cell in zip(packed, module.cells):
-        h = _lstm_layer(h, wcat_t, b, cell.hidden_size)
+    for (w_ih_t, w_hh_t, b), cell in zip(packed, module.cells):
+        h = _lstm_layer(h, w_ih_t, w_hh_t, b, cell.hidden_size)
     last = h[:, -1, :]
     return module.head_act(module.head(last))
 
@@ -90,32 +95,82 @@ class WideLSTMTrainer:
         B = max(1, min(cfg.batch_size, n_windows))
         n_batches = max(1, math.ceil(n_windows / B))
         offsets = torch.arange(T, device=device)
+        F_in = Xt.shape[1]
+        F_out = module.n_features_out
 
         opt = torch.optim.Adam(module.parameters(), lr=cfg.lr, betas=cfg.betas,
                                eps=cfg.eps, foreach=True)
         loss_fn = torch.nn.MSELoss()
         use_bf16 = device.type == "cuda" and cfg.dtype == "bf16"
 
-        history = TrainHistory(engine="wide_lstm_gemmloop")
-        torch.cuda.synchronize() if device.type == "cuda" else None
+        def fwd_loss(w, tgt):
+            if use_bf16:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    out = wide_forward(module, w)
+                return loss_fn(out.float(), tgt)
+            return loss_fn(wide_forward(module, w), tgt)
+
+        # ---- hipGraph capture of the whole fwd+bwd (static shapes) --------
+        graph = None
+        static_w = static_tgt = static_loss = None
+        if device.type == "cuda" and getattr(cfg, "use_graph", True):
+            try:
+                static_w = torch.zeros(B, T, F_in, device=device)
+                static_tgt = torch.zeros(B, F_out, device=device)
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    for _ in range(2):  # warmup: allocator + autocast caches
+                        opt.zero_grad(set_to_none=True)
+                        fwd_loss(static_w, static_tgt).backward()
+                torch.cuda.current_stream().wait_stream(s)
+                torch.cuda.synchronize()
+                opt.zero_grad(set_to_none=True)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    static_loss = fwd_loss(static_w, static_tgt)
+                    static_loss.backward()
+                graph = g
+                # reset weights perturbed by nothing (no opt.step during
+                # warmup), but grads are stale: zero them for the real run
+                for p in module.parameters():
+                    if p.grad is not None:
+                        p.grad.zero_()
+                torch.cuda.synchronize()
+            except Exception as e:
+                logger.warning("wide-LSTM graph capture failed (%s); eager loop", e)
+                graph = None
+
+        history = TrainHistory(
+            engine="wide_lstm_gemmloop" + ("_graph" if graph else "")
+        )
+        if device.type == "cuda":
+            torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(cfg.epochs):
             perm = torch.randperm(n_windows, device=device)
+            if n_batches * B > n_windows:  # pad so every batch is exactly B
+                extra = torch.randint(0, n_windows, (n_batches * B - n_windows,), device=device)
+                perm = torch.cat([perm, extra])
             total = 0.0
-            for s in range(n_batches):
-                idx = perm[s * B : (s + 1) * B]
-                w = Xt[idx.unsqueeze(1) + offsets.unsqueeze(0)]
-                tgt = yt[idx + T - 1 + lookahead]
-                opt.zero_grad(set_to_none=True)
-                if use_bf16:
-                    with torch.autocast("cuda", dtype=torch.bfloat16):
-                        out = wide_forward(module, w)
-                    loss = loss_fn(out.float(), tgt)
+            for s_i in range(n_batches):
+                idx = perm[s_i * B : (s_i + 1) * B]
+                if graph is not None:
+                    torch.index_select(Xt, 0, (idx.unsqueeze(1) + offsets.unsqueeze(0)).reshape(-1),
+                                       out=static_w.view(B * T, F_in))
+                    torch.index_select(yt, 0, idx + T - 1 + lookahead, out=static_tgt)
+                    graph.replay()
+                    opt.step()
+                    opt.zero_grad(set_to_none=False)  # keep graph's grad buffers
+                    total += float(static_loss.detach()) * B
                 else:
-                    loss = loss_fn(wide_forward(module, w), tgt)
-                loss.backward()
-                opt.step()
-                total += float(loss.detach()) * len(idx)
+                    w = Xt[idx.unsqueeze(1) + offsets.unsqueeze(0)]
+                    tgt = yt[idx + T - 1 + lookahead]
+                    opt.zero_grad(set_to_none=True)
+                    loss = fwd_loss(w, tgt)
+                    loss.backward()
+                    opt.step()
+                    total += float(loss.detach()) * B
             history.loss.append(total / (n_batches * B))
         if device.type == "cuda":
             torch.cuda.synchronize()
