@@ -1,20 +1,17 @@
 """Semi-fused wide-LSTM engine (widths the hand-written HIP cells don't
 cover; any hidden size).
 
-torch.nn.LSTM's eager path on ROCm measured ~4K windows/s at hidden 128 —
-with a 144-step lookback the launch count per batch (thousands) dominates.
-This engine removes both sources of overhead:
-
-1. the input projection for ALL timesteps runs as ONE rocBLAS GEMM per layer
-   (x [B·T, I] @ W_ih^T), leaving one [B,H]x[H,4H] addmm + the fused-ish
-   pointwise cell per step;
-2. the whole forward+backward is captured in a hipGraph once per fit and
-   replayed per batch (static shapes; indices gathered into a static input
-   buffer), so the 144-step loop costs zero launch overhead after capture.
+torch.nn.LSTM's eager path on ROCm measured ~4K windows/s at hidden 128
+with a 144-step lookback. This engine batches the input projection for ALL
+timesteps into one rocBLAS GEMM per layer and captures the whole
+forward+backward in a hipGraph replayed per batch (static shapes) —
+measured 2.5x nn.LSTM. The remaining ceiling is per-kernel execution cost
+of the ~7000 small captured kernels in the 144-step autograd chain;
+closing it needs the persistent fused cell kernels generalized past
+64-wide (docs/ROADMAP.md).
 
 Weights stay the canonical torch nn.LSTM parameters (state_dict identical;
-serializer artifacts unchanged). Library GEMMs are efficient at these widths
-— the hand-written MFMA kernels matter most at tiny widths.
+serializer artifacts unchanged).
 """
 from __future__ import annotations
 
