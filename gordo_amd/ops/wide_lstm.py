@@ -41,20 +41,33 @@ def _packed_weights(module, device, dtype):
 
 
 def _lstm_layer(x, w_ih_t, w_hh_t, b, H):
-    """x [B,T,I] -> h_seq [B,T,H]; one big x-projection GEMM + per-t addmm."""
+    """x [B,T,I] -> h_seq [B,T,H]: one x-projection GEMM for all timesteps,
+    then per step one mm + ATen's fused differentiable LSTM cell (the same
+    pointwise kernel torch.nn.LSTM uses on CUDA; dispatching it directly
+    bypasses the slow MIOpen RNN route on ROCm)."""
     B, T, I = x.shape
-    gx = torch.addmm(b, x.reshape(-1, I), w_ih_t).view(B, T, 4 * H)
+    # [T,B,4H] so each timestep slice is contiguous for the fused cell
+    gx = (
+        torch.addmm(b, x.reshape(-1, I), w_ih_t)
+        .view(B, T, 4 * H).transpose(0, 1).contiguous()
+    )
     h = x.new_zeros(B, H)
     c = x.new_zeros(B, H)
     outs = []
-    for t in range(T):
-        gates = torch.addmm(gx[:, t], h, w_hh_t)
-        i, f, g, o = gates.split(H, dim=1)
-        i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
-        g = torch.tanh(g)
-        c = f * c + i * g
-        h = o * torch.tanh(c)
-        outs.append(h)
+    if x.is_cuda:
+        for t in range(T):
+            gh = h.mm(w_hh_t)
+            h, c, _ = torch.ops.aten._thnn_fused_lstm_cell(gx[t], gh, c)
+            outs.append(h)
+    else:
+        for t in range(T):
+            gates = torch.addmm(gx[t], h, w_hh_t)
+            i, f, g, o = gates.split(H, dim=1)
+            i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
+            g = torch.tanh(g)
+            c = f * c + i * g
+            h = o * torch.tanh(c)
+            outs.append(h)
     return torch.stack(outs, dim=1)
 
 
@@ -98,13 +111,10 @@ class WideLSTMTrainer:
         opt = torch.optim.Adam(module.parameters(), lr=cfg.lr, betas=cfg.betas,
                                eps=cfg.eps, foreach=True)
         loss_fn = torch.nn.MSELoss()
-        use_bf16 = device.type == "cuda" and cfg.dtype == "bf16"
 
+        # fp32 throughout: the step is kernel-count-bound, not GEMM-bound,
+        # and the fused ATen cell is fp32-first
         def fwd_loss(w, tgt):
-            if use_bf16:
-                with torch.autocast("cuda", dtype=torch.bfloat16):
-                    out = wide_forward(module, w)
-                return loss_fn(out.float(), tgt)
             return loss_fn(wide_forward(module, w), tgt)
 
         # ---- hipGraph capture of the whole fwd+bwd (static shapes) --------
