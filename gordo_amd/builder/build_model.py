@@ -19,7 +19,6 @@ from typing import Any, Dict, Optional, Tuple, Union
 
 import numpy as np
 import pandas as pd
-from sklearn.base import clone
 from sklearn.metrics import (
     explained_variance_score,
     mean_absolute_error,

@@ -15,7 +15,6 @@ from datetime import datetime
 from typing import List, Optional, Union
 
 import aiohttp
-import numpy as np
 import pandas as pd
 
 from . import io as client_io

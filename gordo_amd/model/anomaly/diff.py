@@ -20,7 +20,6 @@ from sklearn.base import BaseEstimator
 from sklearn.model_selection import TimeSeriesSplit, cross_validate
 
 from .base import AnomalyDetectorBase
-from ..utils import make_base_dataframe
 from ...serializer import from_definition, into_definition
 
 logger = logging.getLogger(__name__)

@@ -26,7 +26,6 @@ from sklearn.metrics import explained_variance_score
 
 from .base import GordoBase
 from .register import lookup_factory
-from .utils import rolling_windows
 from ..runtime.trainer import TrainConfig, Trainer, resolve_device
 
 logger = logging.getLogger(__name__)

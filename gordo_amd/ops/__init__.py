@@ -13,7 +13,6 @@ tests).
 from __future__ import annotations
 
 import logging
-import os
 
 import torch
 

@@ -18,8 +18,6 @@ from __future__ import annotations
 import logging
 import math
 import time
-from typing import List
-
 import numpy as np
 import torch
 

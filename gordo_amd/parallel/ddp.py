@@ -20,7 +20,6 @@ from __future__ import annotations
 import logging
 import math
 import os
-import time
 from typing import Optional
 
 import numpy as np
@@ -107,7 +106,7 @@ class HourglassStepTrainer:
     # ------------------------------------------------------------------
     def _init_fused(self):
         from ..ops import ext
-        from ..ops.hourglass_engine import WMAX, _PackedParams, pad_rows_bf16
+        from ..ops.hourglass_engine import _PackedParams, pad_rows_bf16
 
         self.E = ext()
         device = self.device

@@ -15,7 +15,6 @@ gordo_amd.parallel (flat-bucket all-reduce of the fused gradient buffer).
 from __future__ import annotations
 
 import logging
-import math
 import os
 import time
 from dataclasses import dataclass, field
