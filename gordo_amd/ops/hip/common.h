@@ -134,6 +134,42 @@ __device__ __forceinline__ float ga_act_grad_from_h(float h, int id) {
     }
 }
 
+// ---------------------------------------------------------------------------
+// On-the-fly minibatch shuffle: a keyed bijection on [0, n) via a 4-round
+// Feistel network over the next-pow2 domain with cycle walking. Replaces the
+// host-side randperm + perm buffer (rocprim sorts were ~8% of the hourglass
+// step and forced host work at epoch boundaries; with the epoch key coming
+// from a device counter the whole training run is pure graph replay).
+// kbits = even ceil(log2 n); kbits == 0 means identity.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ unsigned int ga_mix(unsigned int x, unsigned int key) {
+    x ^= key;
+    x *= 0x9E3779B9u;
+    x ^= x >> 16;
+    x *= 0x85EBCA6Bu;
+    x ^= x >> 13;
+    return x;
+}
+
+__device__ __forceinline__ unsigned int ga_permute(unsigned int i, unsigned int n,
+                                                   int kbits, unsigned int key) {
+    if (kbits == 0) return i;
+    const unsigned int half = (unsigned int)kbits >> 1;
+    const unsigned int hmask = (1u << half) - 1u;
+    unsigned int x = i;
+    do {
+        unsigned int L = x >> half, R = x & hmask;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            unsigned int nl = R;
+            R = (L ^ ga_mix(R, key + (unsigned int)r * 0x6A09E667u)) & hmask;
+            L = nl;
+        }
+        x = (L << half) | R;
+    } while (x >= n);
+    return x;
+}
+
 // wave-wide f32 sum (64 lanes)
 __device__ __forceinline__ float ga_wave_sum(float v) {
     for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);

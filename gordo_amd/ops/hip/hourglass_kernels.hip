@@ -26,8 +26,10 @@
 struct HgTrainArgs {
     const unsigned short* __restrict__ X;   // [N][W] bf16, pre-scaled, padded
     const unsigned short* __restrict__ Y;   // may equal X (autoencoder)
-    const int* __restrict__ perm;           // [n_batches*B] row indices
-    const int* __restrict__ counters;       // [0]=batch idx, [1]=adam t
+    const int* __restrict__ counters;       // [0]=batch idx, [1]=adam t,
+                                            // [2]=epoch (permutation key)
+    int kbits;                              // Feistel domain bits (0=identity)
+    unsigned int seed;                      // shuffle seed
     const unsigned short* __restrict__ w;   // [L][W][W] bf16 (out,in)
     const unsigned short* __restrict__ wt;  // [L][W][W] bf16 (in,out)
     const float* __restrict__ bias;         // [L][W] fp32
@@ -101,6 +103,12 @@ __global__ void hg_train_step(HgTrainArgs a) {
 
     const int step = a.counters[0];
     const long long pbase = (long long)step * a.B + base;
+    // per-epoch shuffle key from the device-side epoch counter
+    const unsigned int pkey = a.seed ^ ((unsigned int)a.counters[2] * 0x9E3779B1u);
+    auto sample_row = [&](int m) -> unsigned int {
+        unsigned int g = (unsigned int)((pbase + m) % (long long)a.N);
+        return ga_permute(g, (unsigned int)a.N, a.kbits, pkey);
+    };
 
     // ---- stage biases -------------------------------------------------
     for (int i = tid; i < a.L * W; i += 256) s_bias[i] = a.bias[i];
@@ -112,7 +120,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
         int gi = base + m;
         ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (gi < a.B) {
-            int row = a.perm[pbase + m];
+            unsigned int row = sample_row(m);
             v = *(const ga_u16x8*)(a.X + (long long)row * W + col0);
         }
         hg_write8<W>(act_tile(0), m, col0, v);
@@ -167,7 +175,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
                         if (a.y_is_x) {
                             yval = ga_bf16_to_f32(hg_read1<W>(act_tile(0), m, col));
                         } else {
-                            int row = a.perm[pbase + m];
+                            unsigned int row = sample_row(m);
                             yval = ga_bf16_to_f32(a.Y[(long long)row * W + col]);
                         }
                     }
@@ -388,8 +396,10 @@ __global__ void hg_slab_reduce(const float* __restrict__ dw,
     int g = blockIdx.y;
     int idx4 = blockIdx.x * blockDim.x + threadIdx.x;
     if (g == 0 && idx4 == 0 && counters != nullptr) {
-        counters[0] = (counters[0] + 1) % n_batches;
+        int nb = (counters[0] + 1) % n_batches;
+        counters[0] = nb;
         counters[1] += 1;
+        if (nb == 0) counters[2] += 1;  // new epoch -> new shuffle key
     }
     int idx = idx4 * 4;
     if (idx >= P) return;
@@ -674,19 +684,22 @@ static void ga_fill_acts(int* dst, const std::vector<int64_t>& acts, int L) {
         }                                                     \
     } while (0)
 
-void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
+void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
                           torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
                           torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
                           torch::Tensor loss, int64_t B, int64_t F_out,
                           std::vector<int64_t> acts, double gscale, bool y_is_x,
-                          c10::optional<torch::Tensor> dx_out, int64_t wmax) {
+                          c10::optional<torch::Tensor> dx_out, int64_t wmax,
+                          int64_t kbits, int64_t seed) {
     CHECK_GPU(X);
+    TORCH_CHECK(counters.numel() >= 3, "counters must hold [batch, t, epoch]");
     int L = (int)(w.numel() / (wmax * wmax));
     HgTrainArgs a;
     a.X = (const unsigned short*)X.data_ptr();
     a.Y = (const unsigned short*)Y.data_ptr();
-    a.perm = perm.data_ptr<int>();
     a.counters = counters.data_ptr<int>();
+    a.kbits = (int)kbits;
+    a.seed = (unsigned int)seed;
     a.w = (const unsigned short*)w.data_ptr();
     a.wt = (const unsigned short*)wt.data_ptr();
     a.bias = bias.data_ptr<float>();

@@ -4,12 +4,13 @@
 #include <torch/extension.h>
 #include <vector>
 
-void hg_train_step_launch(torch::Tensor X, torch::Tensor Y, torch::Tensor perm,
+void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
                           torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
                           torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
                           torch::Tensor loss, int64_t B, int64_t F_out,
                           std::vector<int64_t> acts, double gscale, bool y_is_x,
-                          c10::optional<torch::Tensor> dx_out, int64_t wmax);
+                          c10::optional<torch::Tensor> dx_out, int64_t wmax,
+                          int64_t kbits, int64_t seed);
 
 void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                     torch::Tensor dw, torch::Tensor db, torch::Tensor w,

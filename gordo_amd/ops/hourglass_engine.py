@@ -87,6 +87,12 @@ class _PackedParams:
                 lin.bias.copy_(b[i, :out_w].to(lin.bias.device))
 
 
+def feistel_kbits(n: int) -> int:
+    """Even bit-width of the shuffle's Feistel domain for n rows."""
+    k = max(2, int(math.ceil(math.log2(max(2, n)))))
+    return k + (k & 1)
+
+
 def pad_rows_bf16(X: np.ndarray, device, wmax: int = WMAX) -> torch.Tensor:
     """[N,F] fp32 → [N,wmax] bf16 device tensor, zero-padded."""
     N, F = X.shape
@@ -131,29 +137,23 @@ class FusedHourglassTrainer:
         dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=device)
         db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=device)
         loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        # counters [batch idx, adam t]: hg_slab_reduce pre-bumps, so t starts 0
-        counters = torch.tensor([0, 0], dtype=torch.int32, device=device)
+        # counters [batch idx, adam t, epoch]: hg_slab_reduce pre-bumps
+        counters = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
         G = min(16, n_tiles)
         partial = torch.zeros(G, p.P, dtype=torch.float32, device=device)
         m = torch.zeros_like(p.master)
         v = torch.zeros_like(p.master)
-        perm = torch.zeros(n_batches * B, dtype=torch.int32, device=device)
 
         gscale = 2.0 / (B * F_out)
         acts = [int(i) for i in p.act_ids]
-
-        def fill_perm():
-            idx = torch.randperm(N, device=device)
-            need = n_batches * B
-            if need > N:
-                extra = torch.randint(0, N, (need - N,), device=device)
-                idx = torch.cat([idx, extra])
-            perm.copy_(idx[:need].to(torch.int32))
+        kbits = feistel_kbits(N)
+        seed = int(cfg.seed) if cfg.seed is not None else 0x5EED
 
         def one_step():
             self.E.hg_train_step(
-                Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                Xd, Yd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
                 loss_acc, B, F_out, acts, gscale, bool(y_is_x), None, WM,
+                kbits, seed,
             )
             # two-stage slab reduce (counter bump folded into stage A)
             self.E.hg_slab_reduce(dw, db, partial, counters, n_tiles, G, L,
@@ -167,7 +167,6 @@ class FusedHourglassTrainer:
         graph = None
         if getattr(cfg, "use_graph", True) and n_batches >= 1:
             try:
-                fill_perm()
                 # warm-up on a side stream (allocations + kernel modules)
                 s = torch.cuda.Stream()
                 s.wait_stream(torch.cuda.current_stream())
@@ -176,7 +175,7 @@ class FusedHourglassTrainer:
                 torch.cuda.current_stream().wait_stream(s)
                 torch.cuda.synchronize()
                 # reset state disturbed by warm-up
-                counters.copy_(torch.tensor([0, 0], dtype=torch.int32, device=device))
+                counters.copy_(torch.tensor([0, 0, 0], dtype=torch.int32, device=device))
                 loss_acc.zero_()
                 m.zero_(); v.zero_()
                 p2 = _PackedParams(self.module, device)
@@ -186,7 +185,7 @@ class FusedHourglassTrainer:
                 with torch.cuda.graph(graph):
                     one_step()
                 # capture replays with pristine state below
-                counters.copy_(torch.tensor([0, 0], dtype=torch.int32, device=device))
+                counters.copy_(torch.tensor([0, 0, 0], dtype=torch.int32, device=device))
                 loss_acc.zero_(); m.zero_(); v.zero_()
                 p.master.copy_(p2.master); p.bias.copy_(p2.bias)
                 p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)
@@ -199,7 +198,6 @@ class FusedHourglassTrainer:
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(cfg.epochs):
-            fill_perm()
             if graph is not None:
                 for _ in range(n_batches):
                     graph.replay()

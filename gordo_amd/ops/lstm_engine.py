@@ -223,8 +223,7 @@ class FusedLSTMTrainer:
             "h_last": torch.zeros(B, WMAX, dtype=bf16, device=device),
             "y_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
             "dx_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
-            "perm_id": torch.arange(B, dtype=torch.int32, device=device),
-            "counters0": torch.tensor([0, 1], dtype=torch.int32, device=device),
+            "counters0": torch.tensor([0, 1, 0], dtype=torch.int32, device=device),
             "head_dw": torch.zeros(n_head_tiles, 1, WMAX * WMAX, dtype=torch.float32, device=device),
             "head_db": torch.zeros(n_head_tiles, 1, WMAX, dtype=torch.float32, device=device),
             "loss_acc": torch.zeros(1, dtype=torch.float32, device=device),
@@ -271,10 +270,10 @@ class FusedLSTMTrainer:
         tgt = win.long() + (T - 1 + st["lookahead"])
         torch.index_select(st["Yd"], 0, tgt, out=st["y_head"])
         E.hg_train_step(
-            st["h_last"], st["y_head"], st["perm_id"], st["counters0"],
+            st["h_last"], st["y_head"], st["counters0"],
             p.head_w, p.head_wt, p.head_bias, st["head_dw"], st["head_db"],
             st["loss_acc"], B, st["F_out"], st["act_out"], st["gscale"], False,
-            st["dx_head"], WMAX,
+            st["dx_head"], WMAX, 0, 0,
         )
 
         # ---- BPTT (fused dx) + split-K dW kernel, top layer down ---------

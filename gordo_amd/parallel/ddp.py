@@ -129,15 +129,18 @@ class HourglassStepTrainer:
         self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.float32, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        # counters [batch idx, adam t]: hg_slab_reduce pre-bumps, so t starts 0
-        self.counters = torch.tensor([0, 0], dtype=torch.int32, device=device)
+        # counters [batch idx, adam t, epoch]: hg_slab_reduce pre-bumps
+        self.counters = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
         self.G = min(16, self.n_tiles)
         self.partial = torch.zeros(self.G, p.P, dtype=torch.float32, device=device)
         self.m = torch.zeros_like(p.master)
         self.v = torch.zeros_like(p.master)
         self.n_batches = max(1, math.ceil(N / self.B))
-        self.perm = torch.zeros(self.n_batches * self.B, dtype=torch.int32, device=device)
-        self._fill_perm()
+        from ..ops.hourglass_engine import feistel_kbits
+
+        self.kbits = feistel_kbits(N)
+        # every rank shuffles its own shard with its own seed
+        self.seed = 0x5EED ^ (self.dp.rank * 0x9E3779B1)
         self.gscale = 2.0 / (self.B * self.F_out)
         self._graph = None
         # GORDO_NO_GRAPH=1 is the operational escape hatch for the captured
@@ -179,31 +182,20 @@ class HourglassStepTrainer:
             logger.warning("hipGraph capture failed (%s); plain launches", e)
             self._graph = None
 
-    def _fill_perm(self):
-        N, need = self.N, self.perm.numel()
-        idx = torch.randperm(N, device=self.device)
-        if need > N:
-            idx = torch.cat([idx, torch.randint(0, N, (need - N,), device=self.device)])
-        self.perm.copy_(idx[:need].to(torch.int32))
-
     def _fused_step(self):
         if self._graph is not None:
             self._graph.replay()
         else:
             self._fused_step_raw()
-        self._steps_since_perm = getattr(self, "_steps_since_perm", 0) + 1
-        if self._steps_since_perm >= self.n_batches:
-            self._fill_perm()
-            self._steps_since_perm = 0
 
     def _fused_step_raw(self):
         p = self.p
         cfg = self.cfg
         WM = p.wmax
         self.E.hg_train_step(
-            self.Xd, self.Xd, self.perm, self.counters, p.w_bf16, p.wt_bf16, p.bias,
+            self.Xd, self.Xd, self.counters, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
-            self.gscale, True, None, WM,
+            self.gscale, True, None, WM, self.kbits, self.seed,
         )
         # two-stage slab reduce; stage A also bumps the counters
         self.E.hg_slab_reduce(

@@ -118,11 +118,10 @@ def test_fused_training_wide_matches_autograd():
     dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
     db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
-    counters = torch.tensor([0, 1], dtype=torch.int32, device=DEV)
-    perm = torch.arange(B, dtype=torch.int32, device=DEV)
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
-    E.hg_train_step(Xd, Xd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                    loss, B, F_out, p.act_ids, gscale, True, None, WM)
+    E.hg_train_step(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, True, None, WM, 0, 0)
     torch.cuda.synchronize()
 
     Xref = Xd[:, :100].float().cpu()
@@ -185,11 +184,10 @@ def test_fused_training_y_not_x():
     dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=DEV)
     db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
-    counters = torch.tensor([0, 1], dtype=torch.int32, device=DEV)
-    perm = torch.arange(B, dtype=torch.int32, device=DEV)
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
-    E.hg_train_step(Xd, Yd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                    loss, B, F_out, p.act_ids, gscale, False, None, WMAX)
+    E.hg_train_step(Xd, Yd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, False, None, WMAX, 0, 0)
     torch.cuda.synchronize()
 
     Xr = Xd[:, :30].float().cpu()

@@ -105,11 +105,11 @@ def test_train_step_gradients_match_autograd():
     dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=DEV)
     db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
-    counters = torch.tensor([0, 1], dtype=torch.int32, device=DEV)
-    perm = torch.arange(N, dtype=torch.int32, device=DEV)
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
-    E.hg_train_step(Xd, Xd, perm, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
-                    loss, B, F_out, p.act_ids, gscale, True, None, WMAX)
+    # kbits=0 -> identity gather (row m = sample m)
+    E.hg_train_step(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, True, None, WMAX, 0, 0)
     torch.cuda.synchronize()
 
     # --- torch reference on the same bf16-rounded input ------------------
