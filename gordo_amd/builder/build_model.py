@@ -220,10 +220,20 @@ class ModelBuilder:
         from sklearn.metrics import make_scorer
 
         scorers = {}
-        tag_names = [c if isinstance(c, str) else c[0] for c in getattr(X, "columns", range(Xv.shape[1]))]
+        tag_names = [
+            str(c) if isinstance(c, str) else str(c)
+            for c in getattr(X, "columns", range(Xv.shape[1]))
+        ]
+        per_tag = bool(evaluation.get("per_tag_metrics", True))
         for mname, func in metrics.items():
-            wrapped = metric_wrapper(func, scoring_scaler)
-            scorers[mname] = make_scorer(wrapped)
+            scorers[mname] = make_scorer(metric_wrapper(func, scoring_scaler))
+            if per_tag:
+                # per-tag scores (reference parity); sklearn's multimetric
+                # scorer shares one predict across all of these
+                for i, tag in enumerate(tag_names):
+                    scorers[f"{mname}-{tag}"] = make_scorer(
+                        metric_wrapper(func, scoring_scaler, col=i)
+                    )
 
         cv_duration = None
         # threshold-deriving CV path: only when the TOP-level model is the
@@ -257,7 +267,6 @@ class ModelBuilder:
         }
         if anomaly_target is not None and anomaly_target.aggregate_threshold_ is not None:
             out["thresholds"] = anomaly_target.get_metadata()
-        _ = tag_names  # per-tag metric expansion handled via scaler-wrapped metrics
         return out
 
     @staticmethod

@@ -68,10 +68,11 @@ def make_base_dataframe(
     return df
 
 
-def metric_wrapper(metric: Callable, scaler=None) -> Callable:
+def metric_wrapper(metric: Callable, scaler=None, col: int = None) -> Callable:
     """Wrap an sklearn metric so y/y_pred are transformed by ``scaler`` first
     (parity: the reference scales targets before computing CV metrics so
-    per-tag magnitudes don't dominate)."""
+    per-tag magnitudes don't dominate). ``col`` restricts the metric to one
+    tag column (the reference's per-tag CV scores)."""
 
     def _wrapped(y_true, y_pred, *args, **kwargs):
         y_true = np.asarray(y_true)
@@ -84,6 +85,9 @@ def metric_wrapper(metric: Callable, scaler=None) -> Callable:
         if scaler is not None:
             y_true = scaler.transform(y_true)
             y_pred = scaler.transform(y_pred)
+        if col is not None and y_true.ndim == 2 and y_true.shape[1] > col:
+            y_true = y_true[:, col]
+            y_pred = y_pred[:, col]
         return metric(y_true, y_pred, *args, **kwargs)
 
     _wrapped.__name__ = getattr(metric, "__name__", "metric")
