@@ -8,6 +8,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hg_train_step", &hg_train_step_launch, "fused hourglass train step");
     m.def("hg_adam", &hg_adam_launch, "slab-reduce + Adam + bf16 weight emit");
     m.def("hg_bump", &hg_bump_launch, "advance step counters");
+    m.def("hg_permute_fill", &hg_permute_fill_launch, "materialize Feistel shuffle (tests)");
     m.def("hg_forward", &hg_forward_launch, "fused serving forward (+anomaly)");
     m.def("hg_grad_reduce", &hg_grad_reduce_launch, "slab reduce -> flat grad bucket");
     m.def("hg_adam_flat", &hg_adam_flat_launch, "Adam from reduced flat grad");

@@ -373,6 +373,12 @@ __global__ void hg_adam(
     }
 }
 
+// test support: materialize the Feistel permutation (bijectivity checks)
+__global__ void hg_permute_fill(int* out, int n, int kbits, unsigned int key) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) out[i] = (int)ga_permute((unsigned int)i, (unsigned int)n, kbits, key);
+}
+
 __global__ void hg_bump(int* counters, int n_batches) {
     counters[0] = (counters[0] + 1) % n_batches;
     counters[1] += 1;
@@ -749,6 +755,14 @@ void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                            bias.data_ptr<float>(), counters.data_ptr<int>(), (int)n_slabs,
                            (int)L, (float)lr, (float)beta1, (float)beta2, (float)eps);
     });
+}
+
+void hg_permute_fill_launch(torch::Tensor out, int64_t kbits, int64_t key) {
+    int n = (int)out.numel();
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(hg_permute_fill, dim3((n + 255) / 256), dim3(256), 0,
+                       stream.stream(), out.data_ptr<int>(), n, (int)kbits,
+                       (unsigned int)key);
 }
 
 void hg_bump_launch(torch::Tensor counters, int64_t n_batches) {

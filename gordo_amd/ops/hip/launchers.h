@@ -20,6 +20,8 @@ void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
 
 void hg_bump_launch(torch::Tensor counters, int64_t n_batches);
 
+void hg_permute_fill_launch(torch::Tensor out, int64_t kbits, int64_t key);
+
 void hg_forward_launch(torch::Tensor X, torch::Tensor w, torch::Tensor bias,
                        torch::Tensor xscale, torch::Tensor xshift, torch::Tensor yhat,
                        c10::optional<torch::Tensor> err, c10::optional<torch::Tensor> total,

@@ -192,3 +192,26 @@ def test_estimator_uses_fused_engine():
     assert model.history["engine"].startswith("fused_hip"), model.history
     out = model.predict(X)
     assert out.shape == X.shape
+
+
+def test_feistel_shuffle_is_bijective():
+    """The device-side minibatch shuffle must be a true permutation of [0, N)
+    for every epoch key (a broken shuffle silently corrupts training)."""
+    from gordo_amd.ops.hourglass_engine import feistel_kbits
+
+    E = ext()
+    for n in (7, 64, 1000, 131072):
+        kbits = feistel_kbits(n)
+        for key in (0, 1, 0xDEADBEEF):
+            out = torch.zeros(n, dtype=torch.int32, device=DEV)
+            E.hg_permute_fill(out, kbits, key)
+            torch.cuda.synchronize()
+            vals = out.cpu().numpy()
+            assert sorted(vals) == list(range(n)), (n, key)
+    # different keys give different orders (the point of the epoch key)
+    a = torch.zeros(1000, dtype=torch.int32, device=DEV)
+    b = torch.zeros(1000, dtype=torch.int32, device=DEV)
+    E.hg_permute_fill(a, feistel_kbits(1000), 1)
+    E.hg_permute_fill(b, feistel_kbits(1000), 2)
+    torch.cuda.synchronize()
+    assert not torch.equal(a, b)
