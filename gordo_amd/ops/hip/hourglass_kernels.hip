@@ -413,12 +413,14 @@ __global__ void hg_slab_reduce(const float* __restrict__ dw,
     int s0 = g * per, s1 = min(n_slabs, s0 + per);
     float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
     if (idx < PW) {
+#pragma unroll 4
         for (int sl = s0; sl < s1; ++sl) {
             float4 v = *(const float4*)(dw + (size_t)sl * PW + idx);
             acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
         }
     } else {
         int bidx = idx - PW;
+#pragma unroll 4
         for (int sl = s0; sl < s1; ++sl) {
             float4 v = *(const float4*)(db + (size_t)sl * (L * W) + bidx);
             acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
@@ -450,6 +452,7 @@ __global__ void hg_adam_p(
         sidx = (size_t)idx;
     }
     float g = 0.f;
+#pragma unroll 4
     for (int i = 0; i < G; ++i) g += partial[(size_t)i * P + sidx];
     int t = counters[1];
     float m = beta1 * m_buf[idx] + (1.f - beta1) * g;
