@@ -494,6 +494,7 @@ __global__ void hg_partial_flat(const float* __restrict__ partial,
         sidx = (size_t)idx;
     }
     float sum = 0.f;
+#pragma unroll 4
     for (int i = 0; i < G; ++i) sum += partial[(size_t)i * P + sidx];
     g_out[idx] = sum;
 }
