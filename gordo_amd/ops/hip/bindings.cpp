@@ -18,6 +18,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_fwd", &lstm_fwd_launch, "persistent fused 4-gate LSTM forward");
     m.def("lstm_bwd_cell", &lstm_bwd_cell_launch, "reverse-time BPTT cell kernel (+fused dx)");
     m.def("lstm_dw", &lstm_dw_launch, "split-K dWcat + dbias from one dgates pass");
+    m.def("lstm_fwd_wide", &lstm_fwd_wide_launch, "wide (W=128) persistent LSTM forward");
+    m.def("lstm_bwd_wide", &lstm_bwd_wide_launch, "wide reverse-time BPTT cell (+fused dx)");
+    m.def("lstm_dw_wide", &lstm_dw_wide_launch, "wide split-K dWcat quadrants + dbias");
     m.attr("WMAX") = 64;
     m.attr("MAXL") = 5;
     m.attr("MT") = 32;   // hg_train_step batch rows per block/slab

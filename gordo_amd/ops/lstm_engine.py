@@ -41,8 +41,17 @@ from ..runtime.trainer import TrainHistory
 
 logger = logging.getLogger(__name__)
 
-H4 = 4 * WMAX  # 256
+H4 = 4 * WMAX  # 256 (narrow variant; wide uses per-instance 4*W)
 KCAT = 2 * WMAX  # 128
+WMAX_WIDE = 128
+
+
+def lstm_width(module) -> int:
+    """Padded kernel width: 64 (chip-resident weights) or 128 (wide)."""
+    dims = [module.n_features, module.n_features_out]
+    for cell in module.cells:
+        dims += [cell.hidden_size, cell.input_size]
+    return WMAX if max(dims) <= WMAX else WMAX_WIDE
 
 
 def lstm_supported(module) -> Optional[str]:
@@ -51,13 +60,13 @@ def lstm_supported(module) -> Optional[str]:
 
     if not isinstance(module, LSTMNet):
         return f"{type(module).__name__} is not LSTMNet"
-    if module.n_features > WMAX:
-        return f"n_features {module.n_features} > {WMAX}"
-    if module.n_features_out > WMAX:
-        return f"n_features_out {module.n_features_out} > {WMAX}"
+    if module.n_features > WMAX_WIDE:
+        return f"n_features {module.n_features} > {WMAX_WIDE}"
+    if module.n_features_out > WMAX_WIDE:
+        return f"n_features_out {module.n_features_out} > {WMAX_WIDE}"
     for cell in module.cells:
-        if cell.hidden_size > WMAX or cell.input_size > WMAX:
-            return f"LSTM dims {cell.input_size}->{cell.hidden_size} > {WMAX}"
+        if cell.hidden_size > WMAX_WIDE or cell.input_size > WMAX_WIDE:
+            return f"LSTM dims {cell.input_size}->{cell.hidden_size} > {WMAX_WIDE}"
         if cell.num_layers != 1 or cell.bidirectional:
             return "only stacked unidirectional single-layer cells"
     if module.out_func not in ACT_IDS:
@@ -72,6 +81,10 @@ class _LSTMPackedParams:
         self.module = module
         self.device = device
         self.Lc = len(module.cells)
+        W = lstm_width(module)
+        self.W = W
+        self.H4 = 4 * W
+        self.KCAT = 2 * W
         self.wcat_master: List[torch.Tensor] = []
         self.bias_master: List[torch.Tensor] = []
         self.wcat_bf16: List[torch.Tensor] = []
@@ -80,22 +93,22 @@ class _LSTMPackedParams:
         for cell in module.cells:
             H, I = cell.hidden_size, cell.input_size
             self.h_real.append(H)
-            wcat = torch.zeros(H4, KCAT, dtype=torch.float32, device=device)
-            bias = torch.zeros(H4, dtype=torch.float32, device=device)
+            wcat = torch.zeros(self.H4, self.KCAT, dtype=torch.float32, device=device)
+            bias = torch.zeros(self.H4, dtype=torch.float32, device=device)
             w_ih = cell.weight_ih_l0.detach().to(device, torch.float32)  # [4H, I]
             w_hh = cell.weight_hh_l0.detach().to(device, torch.float32)  # [4H, H]
             b = (cell.bias_ih_l0 + cell.bias_hh_l0).detach().to(device, torch.float32)
             for g in range(4):
-                wcat[g * WMAX : g * WMAX + H, :I] = w_ih[g * H : (g + 1) * H]
-                wcat[g * WMAX : g * WMAX + H, WMAX : WMAX + H] = w_hh[g * H : (g + 1) * H]
-                bias[g * WMAX : g * WMAX + H] = b[g * H : (g + 1) * H]
+                wcat[g * W : g * W + H, :I] = w_ih[g * H : (g + 1) * H]
+                wcat[g * W : g * W + H, W : W + H] = w_hh[g * H : (g + 1) * H]
+                bias[g * W : g * W + H] = b[g * H : (g + 1) * H]
             self.wcat_master.append(wcat)
             self.bias_master.append(bias)
             self.wcat_bf16.append(wcat.to(torch.bfloat16).contiguous())
             self.wcat_t_bf16.append(wcat.t().contiguous().to(torch.bfloat16))
-        # dense head: bf16 w [1][64][64] (out,in), wt (in,out), fp32 bias [64]
-        self.head_wm = torch.zeros(WMAX, WMAX, dtype=torch.float32, device=device)
-        self.head_bm = torch.zeros(WMAX, dtype=torch.float32, device=device)
+        # dense head: bf16 w [1][W][W] (out,in), wt (in,out), fp32 bias [W]
+        self.head_wm = torch.zeros(W, W, dtype=torch.float32, device=device)
+        self.head_bm = torch.zeros(W, dtype=torch.float32, device=device)
         out_w, in_w = module.head.weight.shape
         self.head_wm[:out_w, :in_w] = module.head.weight.detach().to(device, torch.float32)
         self.head_bm[:out_w] = module.head.bias.detach().to(device, torch.float32)
@@ -116,19 +129,20 @@ class _LSTMPackedParams:
 
     def sync_to_module(self):
         with torch.no_grad():
+            W = self.W
             for l, cell in enumerate(self.module.cells):
                 H, I = cell.hidden_size, cell.input_size
                 wcat = self.wcat_master[l]
                 bias = self.bias_master[l]
                 for g in range(4):
                     cell.weight_ih_l0[g * H : (g + 1) * H].copy_(
-                        wcat[g * WMAX : g * WMAX + H, :I].to(cell.weight_ih_l0.device)
+                        wcat[g * W : g * W + H, :I].to(cell.weight_ih_l0.device)
                     )
                     cell.weight_hh_l0[g * H : (g + 1) * H].copy_(
-                        wcat[g * WMAX : g * WMAX + H, WMAX : WMAX + H].to(cell.weight_hh_l0.device)
+                        wcat[g * W : g * W + H, W : W + H].to(cell.weight_hh_l0.device)
                     )
                     # split the fused bias evenly (torch keeps two bias vectors)
-                    half = bias[g * WMAX : g * WMAX + H].to(cell.bias_ih_l0.device) * 0.5
+                    half = bias[g * W : g * W + H].to(cell.bias_ih_l0.device) * 0.5
                     cell.bias_ih_l0[g * H : (g + 1) * H].copy_(half)
                     cell.bias_hh_l0[g * H : (g + 1) * H].copy_(half)
             out_w, in_w = self.module.head.weight.shape
@@ -164,8 +178,8 @@ class FusedLSTMTrainer:
         F_out = self.module.n_features_out
 
         y_is_x = y is X or (X.shape == y.shape and np.shares_memory(X, y))
-        Xd = pad_rows_bf16(np.asarray(X, np.float32), device)
-        Yd = Xd if y_is_x else pad_rows_bf16(np.asarray(y, np.float32), device)
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), device, p.W)
+        Yd = Xd if y_is_x else pad_rows_bf16(np.asarray(y, np.float32), device, p.W)
         N = Xd.shape[0]
         n_windows = N - T + 1 - lookahead
         if n_windows <= 0:
@@ -201,7 +215,8 @@ class FusedLSTMTrainer:
     # ------------------------------------------------------------------
     def _alloc_step_state(self, Xd, Yd, B, T, n_batches, n_windows, lookahead, F_out):
         device = self.device
-        Lc = self.p.Lc
+        p = self.p
+        Lc, W, WH4, WKC = p.Lc, p.W, p.H4, p.KCAT
         bf16 = torch.bfloat16
         MT = int(getattr(self.E, "MT", 32))
         n_head_tiles = (B + MT - 1) // MT
@@ -209,27 +224,32 @@ class FusedLSTMTrainer:
             "Xd": Xd, "Yd": Yd, "B": B, "T": T, "n_batches": n_batches,
             "n_windows": n_windows, "lookahead": lookahead, "F_out": F_out,
             "perm": torch.zeros(n_batches * B, dtype=torch.int32, device=device),
-            "h_seq": [torch.zeros(B, T, WMAX, dtype=bf16, device=device) for _ in range(Lc)],
-            "c_seq": [torch.zeros(B, T, WMAX, dtype=bf16, device=device) for _ in range(Lc)],
-            "gates_seq": [torch.zeros(B, T, H4, dtype=bf16, device=device) for _ in range(Lc)],
-            "x_seq0": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
-            "dgates": torch.zeros(B, T, H4, dtype=bf16, device=device),
-            "dh_seq": torch.zeros(B, T, WMAX, dtype=bf16, device=device),
-            # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
-            "dw_part": torch.zeros(512, H4 * KCAT, dtype=torch.float32, device=device),
-            "db_part": torch.zeros(512, H4, dtype=torch.float32, device=device),
-            "gw": [torch.zeros(H4 * KCAT, dtype=torch.float32, device=device) for _ in range(Lc)],
-            "gb": [torch.zeros(H4, dtype=torch.float32, device=device) for _ in range(Lc)],
-            "h_last": torch.zeros(B, WMAX, dtype=bf16, device=device),
-            "y_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
-            "dx_head": torch.zeros(B, WMAX, dtype=bf16, device=device),
+            "h_seq": [torch.zeros(B, T, W, dtype=bf16, device=device) for _ in range(Lc)],
+            "c_seq": [torch.zeros(B, T, W, dtype=bf16, device=device) for _ in range(Lc)],
+            "gates_seq": [torch.zeros(B, T, WH4, dtype=bf16, device=device) for _ in range(Lc)],
+            "x_seq0": torch.zeros(B, T, W, dtype=bf16, device=device),
+            "dgates": torch.zeros(B, T, WH4, dtype=bf16, device=device),
+            "dh_seq": torch.zeros(B, T, W, dtype=bf16, device=device),
+            "gw": [torch.zeros(WH4 * WKC, dtype=torch.float32, device=device) for _ in range(Lc)],
+            "gb": [torch.zeros(WH4, dtype=torch.float32, device=device) for _ in range(Lc)],
+            "h_last": torch.zeros(B, W, dtype=bf16, device=device),
+            "y_head": torch.zeros(B, W, dtype=bf16, device=device),
+            "dx_head": torch.zeros(B, W, dtype=bf16, device=device),
             "counters0": torch.tensor([0, 1, 0], dtype=torch.int32, device=device),
-            "head_dw": torch.zeros(n_head_tiles, 1, WMAX * WMAX, dtype=torch.float32, device=device),
-            "head_db": torch.zeros(n_head_tiles, 1, WMAX, dtype=torch.float32, device=device),
+            "head_dw": torch.zeros(n_head_tiles, 1, W * W, dtype=torch.float32, device=device),
+            "head_db": torch.zeros(n_head_tiles, 1, W, dtype=torch.float32, device=device),
             "loss_acc": torch.zeros(1, dtype=torch.float32, device=device),
             "gscale": 2.0 / (B * F_out),
             "act_out": [ACT_IDS[self.module.out_func]],
         }
+        if W == WMAX:
+            # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
+            st["dw_part"] = torch.zeros(512, WH4 * WKC, dtype=torch.float32, device=device)
+            st["db_part"] = torch.zeros(512, WH4, dtype=torch.float32, device=device)
+        else:
+            # wide: [4 quadrants][NB][256*128] + dbias [2][NB][256]
+            st["dw_part"] = torch.zeros(4, 256, 256 * 128, dtype=torch.float32, device=device)
+            st["db_part"] = torch.zeros(2, 256, 256, dtype=torch.float32, device=device)
         return st
 
     def _fill_perm(self, st):
@@ -244,9 +264,10 @@ class FusedLSTMTrainer:
         """Run the stacked fused forward for window starts ``win`` [B] int32."""
         E, p = self.E, self.p
         B, T = st["B"], st["T"]
+        k_fwd = E.lstm_fwd if p.W == WMAX else E.lstm_fwd_wide
         for l in range(p.Lc):
             src = st["Xd"] if l == 0 else st["h_seq"][l - 1]
-            E.lstm_fwd(
+            k_fwd(
                 src, win if l == 0 else None, p.wcat_bf16[l], p.bias_master[l],
                 st["h_seq"][l], st["gates_seq"][l], st["c_seq"][l],
                 st["x_seq0"] if l == 0 else None, B, T,
@@ -273,33 +294,50 @@ class FusedLSTMTrainer:
             st["h_last"], st["y_head"], st["counters0"],
             p.head_w, p.head_wt, p.head_bias, st["head_dw"], st["head_db"],
             st["loss_acc"], B, st["F_out"], st["act_out"], st["gscale"], False,
-            st["dx_head"], WMAX, 0, 0,
+            st["dx_head"], p.W, 0, 0,
         )
 
         # ---- BPTT (fused dx) + split-K dW kernel, top layer down ---------
         # dh_seq doubles as the dx output: within a block, dh_out[gi][t] is
         # read (elementwise phase) before dx_seq[gi][t] is written (GEMM
         # phase of the same t), and blocks own disjoint gi rows.
+        k_bwd = E.lstm_bwd_cell if p.W == WMAX else E.lstm_bwd_wide
         for l in reversed(range(p.Lc)):
             dh = st["dx_head"] if l == p.Lc - 1 else st["dh_seq"]
-            E.lstm_bwd_cell(
+            k_bwd(
                 dh, st["gates_seq"][l], st["c_seq"][l], p.wcat_t_bf16[l],
                 st["dgates"], st["dh_seq"] if l > 0 else None, B, T,
                 l == p.Lc - 1,
             )
             x_seq = st["x_seq0"] if l == 0 else st["h_seq"][l - 1]
-            E.lstm_dw(
-                st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
-                B * T, T,
-            )
-            torch.sum(st["dw_part"], dim=0, out=st["gw"][l])
-            torch.sum(st["db_part"], dim=0, out=st["gb"][l])
-            p.wcat_master[l].grad = st["gw"][l].view(H4, KCAT)
-            p.bias_master[l].grad = st["gb"][l]
+            if p.W == WMAX:
+                E.lstm_dw(
+                    st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
+                    B * T, T,
+                )
+                torch.sum(st["dw_part"], dim=0, out=st["gw"][l])
+                torch.sum(st["db_part"], dim=0, out=st["gb"][l])
+                p.wcat_master[l].grad = st["gw"][l].view(p.H4, p.KCAT)
+                p.bias_master[l].grad = st["gb"][l]
+            else:
+                E.lstm_dw_wide(
+                    st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
+                    B * T, T,
+                )
+                # quadrants [gh][ih] -> gw [512][256]
+                q = st["dw_part"].sum(1).view(2, 2, 256, 128)
+                gw = st["gw"][l].view(p.H4, p.KCAT)
+                gw[:256, :128].copy_(q[0, 0])
+                gw[:256, 128:].copy_(q[0, 1])
+                gw[256:, :128].copy_(q[1, 0])
+                gw[256:, 128:].copy_(q[1, 1])
+                st["gb"][l].view(2, 256).copy_(st["db_part"].sum(1))
+                p.wcat_master[l].grad = gw
+                p.bias_master[l].grad = st["gb"][l]
 
         # ---- head grads from the hg slabs (slab layout is [in][out]) -----
-        p.head_wm.grad = st["head_dw"].sum(0).view(WMAX, WMAX).t().contiguous()
-        p.head_bm.grad = st["head_db"].sum(0).view(WMAX)
+        p.head_wm.grad = st["head_dw"].sum(0).view(p.W, p.W).t().contiguous()
+        p.head_bm.grad = st["head_db"].sum(0).view(p.W)
 
 
 class LSTMStepTrainer:
@@ -392,7 +430,7 @@ class FusedLSTMForward:
         p = self.p
         T = self.module.lookback_window
         F_out = self.module.n_features_out
-        Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device)
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device, p.W)
         N = Xd.shape[0]
         n_windows = N - T + 1
         if n_windows <= 0:
@@ -400,9 +438,10 @@ class FusedLSTMForward:
         outs = []
         bf16 = torch.bfloat16
         B = min(batch_size, n_windows)
-        h_seq = [torch.zeros(B, T, WMAX, dtype=bf16, device=self.device) for _ in range(p.Lc)]
-        c_seq = torch.zeros(B, T, WMAX, dtype=bf16, device=self.device)
-        gates = torch.zeros(B, T, H4, dtype=bf16, device=self.device)
+        h_seq = [torch.zeros(B, T, p.W, dtype=bf16, device=self.device) for _ in range(p.Lc)]
+        c_seq = torch.zeros(B, T, p.W, dtype=bf16, device=self.device)
+        gates = torch.zeros(B, T, p.H4, dtype=bf16, device=self.device)
+        k_fwd = self.E.lstm_fwd if p.W == WMAX else self.E.lstm_fwd_wide
         for s in range(0, n_windows, B):
             n = min(B, n_windows - s)
             win = torch.arange(s, s + n, dtype=torch.int32, device=self.device)
@@ -410,7 +449,7 @@ class FusedLSTMForward:
                 win = torch.cat([win, torch.zeros(B - n, dtype=torch.int32, device=self.device)])
             for l in range(p.Lc):
                 src = Xd if l == 0 else h_seq[l - 1]
-                self.E.lstm_fwd(
+                k_fwd(
                     src, win if l == 0 else None, p.wcat_bf16[l], p.bias_master[l],
                     h_seq[l], gates, c_seq, None, B, T,
                 )

@@ -151,3 +151,94 @@ def test_lstm_forecast_fused():
     assert model.history["engine"] == "fused_hip_lstm", model.history
     score = model.score(X)
     assert np.isfinite(score)
+
+
+# ---------------------------------------------------------------------------
+# wide variant (W=128): feature/hidden widths 65..128
+# ---------------------------------------------------------------------------
+def test_wide_lstm_fused_forward_matches_torch():
+    T = 16
+    net = make_lstm(n_features=100, dims=(96,), T=T, seed=7)
+    from gordo_amd.ops.lstm_engine import lstm_width
+
+    assert lstm_width(net) == 128
+    X = np.random.default_rng(7).normal(size=(300, 100)).astype(np.float32)
+    ref = ref_forward(net, X, T)
+    out = FusedLSTMForward(net, DEV).predict(X)
+    assert out.shape == ref.shape
+    np.testing.assert_allclose(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_wide_lstm_grads_match_autograd():
+    T, B, F = 12, 256, 100
+    net = make_lstm(n_features=F, dims=(96,), T=T, seed=8)
+    ref_net = make_lstm(n_features=F, dims=(96,), T=T, seed=8)
+    X = np.random.default_rng(8).normal(size=(B + T + 2, F)).astype(np.float32)
+
+    from gordo_amd.ops.hourglass_engine import pad_rows_bf16
+    from gordo_amd.runtime.trainer import TrainConfig
+
+    cfg = TrainConfig(epochs=1, batch_size=B)
+    tr = FusedLSTMTrainer(net, DEV, cfg)
+    pp = _LSTMPackedParams(net, DEV)
+    assert pp.W == 128
+    tr.p = pp
+    Xd = pad_rows_bf16(X, DEV, pp.W)
+    st = tr._alloc_step_state(Xd, Xd, B, T, 1, B, 0, F)
+    win = torch.arange(B, dtype=torch.int32, device=DEV)
+    tr._compute_grads(st, win)
+    torch.cuda.synchronize()
+
+    Xb = torch.as_tensor(X).to(torch.bfloat16).float()
+    wins = torch.stack([Xb[i : i + T] for i in range(B)])
+    tgt = torch.stack([Xb[i + T - 1] for i in range(B)])
+    loss = torch.nn.functional.mse_loss(ref_net(wins), tgt)
+    loss.backward()
+
+    kern_loss = float(st["loss_acc"].item()) / (B * F)
+    assert abs(kern_loss - float(loss.detach())) / max(float(loss), 1e-9) < 6e-2
+
+    cell = ref_net.cells[0]
+    H, I = cell.hidden_size, cell.input_size
+    W = pp.W
+    gw = pp.wcat_master[0].grad.cpu()
+    gb = pp.bias_master[0].grad.cpu()
+    for g in range(4):
+        ref_ih = cell.weight_ih_l0.grad[g * H : (g + 1) * H]
+        ref_hh = cell.weight_hh_l0.grad[g * H : (g + 1) * H]
+        ref_b = cell.bias_ih_l0.grad[g * H : (g + 1) * H]
+        got_ih = gw[g * W : g * W + H, :I]
+        got_hh = gw[g * W : g * W + H, W : W + H]
+        got_b = gb[g * W : g * W + H]
+        s = ref_ih.abs().max().item() or 1.0
+        assert (got_ih - ref_ih).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} dW_ih"
+        s = ref_hh.abs().max().item() or 1.0
+        assert (got_hh - ref_hh).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} dW_hh"
+        s = ref_b.abs().max().item() or 1.0
+        assert (got_b - ref_b).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} db"
+        assert gw[g * W + H : (g + 1) * W].abs().max().item() == 0.0
+    ref_hw = ref_net.head.weight.grad
+    got_hw = pp.head_wm.grad.cpu()[: ref_hw.shape[0], : ref_hw.shape[1]]
+    s = ref_hw.abs().max().item() or 1.0
+    assert (got_hw - ref_hw).abs().max().item() < 0.08 * s + 2e-4, "head dW"
+
+
+def test_wide_lstm_estimator_learns_fused():
+    from gordo_amd.model.models import TorchLSTMAutoEncoder
+
+    rng = np.random.default_rng(9)
+    t = np.arange(4000)
+    X = np.stack(
+        [np.sin(t / (8 + i % 11)) + 0.05 * rng.standard_normal(len(t)) for i in range(100)],
+        axis=1,
+    ).astype(np.float32)
+    model = TorchLSTMAutoEncoder(
+        kind="lstm_symmetric", dims=[96], lookback_window=16, epochs=4,
+        batch_size=1024, lr=3e-3,
+    )
+    model.fit(X)
+    assert model.history["engine"] == "fused_hip_lstm", model.history
+    assert model.history["loss"][-1] < model.history["loss"][0]
+    out = model.predict(X[:200])
+    assert out.shape == (200 - 16 + 1, 100)
+    assert np.isfinite(out).all()
