@@ -264,54 +264,89 @@ __global__ void lstm_bwd_wide(WBwdArgs a) {
         const bool need_dx = (a.dx_seq != nullptr);
         if (!need_rec && !need_dx) break;
 
-        // ---- shared-A GEMMs over dgates tiles (B-frags from L2) -----------
+        // ---- GEMMs over dgates tiles (B-frags from L2) --------------------
         //   dh_rec[m][j<W] = dgates[m][512] x W_hh (WcatT rows W+j)
         //   dx[m][j<W]     = dgates[m][512] x W_ih (WcatT rows j)
-        ga_f32x4 acc[8], accx[8];
+        // When both outputs are needed, wave pairs specialize (0,1 -> rec;
+        // 2,3 -> dx) and each wave covers TWO 16-row stripes per B-fragment
+        // load — halving the L2 loads per MFMA vs every wave doing both.
+        if (need_rec && need_dx) {
+            const int otype = wid >> 1;   // 0 = rec, 1 = dx
+            const int sub = wid & 1;      // row half: rows sub*32 .. +32
+            const int brow0 = (otype == 0) ? WW : 0;
+            ga_f32x4 acc2[2][8];
 #pragma unroll
-        for (int nt = 0; nt < 8; ++nt) {
-            acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
-            accx[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
-        }
-        for (int kk = 0; kk < W4H; kk += 32) {
-            const char* dgt = dg_tiles + (kk >> 7) * WTB;
-            ga_bf16x8 afrag = *(const ga_bf16x8*)&dgt[sw128(wrow + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
-            if (need_rec)
+            for (int st2 = 0; st2 < 2; ++st2)
+#pragma unroll
+                for (int nt = 0; nt < 8; ++nt) acc2[st2][nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            for (int kk = 0; kk < W4H; kk += 32) {
+                const char* dgt = dg_tiles + (kk >> 7) * WTB;
+                ga_bf16x8 bfrag[8];
+#pragma unroll
+                for (int nt = 0; nt < 8; ++nt)
+                    bfrag[nt] = *(const ga_bf16x8*)(a.wcat_t +
+                        (size_t)(brow0 + nt * 16 + (lane & 15)) * W4H + kk + (lane >> 4) * 8);
+#pragma unroll
+                for (int st2 = 0; st2 < 2; ++st2) {
+                    ga_bf16x8 afrag = *(const ga_bf16x8*)&dgt[sw128(
+                        sub * 32 + st2 * 16 + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
+#pragma unroll
+                    for (int nt = 0; nt < 8; ++nt)
+                        acc2[st2][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag[nt], acc2[st2][nt], 0, 0, 0);
+                }
+            }
+#pragma unroll
+            for (int st2 = 0; st2 < 2; ++st2)
+#pragma unroll
+                for (int nt = 0; nt < 8; ++nt) {
+                    int col = nt * 16 + (lane & 15);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int m = sub * 32 + st2 * 16 + (lane >> 4) * 4 + r;
+                        if (otype == 0) {
+                            *(unsigned short*)&dh_tile[sw128(m, col)] =
+                                ga_f32_to_bf16(acc2[st2][nt][r]);
+                        } else {
+                            int gi = base + m;
+                            if (gi < a.B)
+                                a.dx_seq[((size_t)gi * a.T + t) * WW + col] =
+                                    ga_f32_to_bf16(acc2[st2][nt][r]);
+                        }
+                    }
+                }
+        } else {
+            // single output: all 4 waves, one 16-row stripe each
+            const bool rec_mode = need_rec;
+            const int brow0 = rec_mode ? WW : 0;
+            ga_f32x4 acc[8];
+#pragma unroll
+            for (int nt = 0; nt < 8; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            for (int kk = 0; kk < W4H; kk += 32) {
+                const char* dgt = dg_tiles + (kk >> 7) * WTB;
+                ga_bf16x8 afrag = *(const ga_bf16x8*)&dgt[sw128(wrow + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
 #pragma unroll
                 for (int nt = 0; nt < 8; ++nt) {
                     ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t +
-                        (size_t)(WW + nt * 16 + (lane & 15)) * W4H + kk + (lane >> 4) * 8);
+                        (size_t)(brow0 + nt * 16 + (lane & 15)) * W4H + kk + (lane >> 4) * 8);
                     acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
                 }
-            if (need_dx)
-#pragma unroll
-                for (int nt = 0; nt < 8; ++nt) {
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t +
-                        (size_t)(nt * 16 + (lane & 15)) * W4H + kk + (lane >> 4) * 8);
-                    accx[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, accx[nt], 0, 0, 0);
-                }
-        }
-        if (need_rec)
-#pragma unroll
-            for (int nt = 0; nt < 8; ++nt) {
-                int col = nt * 16 + (lane & 15);
-#pragma unroll
-                for (int r = 0; r < 4; ++r)
-                    *(unsigned short*)&dh_tile[sw128(wrow + (lane >> 4) * 4 + r, col)] =
-                        ga_f32_to_bf16(acc[nt][r]);
             }
-        if (need_dx)
 #pragma unroll
             for (int nt = 0; nt < 8; ++nt) {
                 int col = nt * 16 + (lane & 15);
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int m = wrow + (lane >> 4) * 4 + r;
-                    int gi = base + m;
-                    if (gi < a.B)
-                        a.dx_seq[((size_t)gi * a.T + t) * WW + col] = ga_f32_to_bf16(accx[nt][r]);
+                    if (rec_mode) {
+                        *(unsigned short*)&dh_tile[sw128(m, col)] = ga_f32_to_bf16(acc[nt][r]);
+                    } else {
+                        int gi = base + m;
+                        if (gi < a.B)
+                            a.dx_seq[((size_t)gi * a.T + t) * WW + col] = ga_f32_to_bf16(acc[nt][r]);
+                    }
                 }
             }
+        }
         __syncthreads();
         if (!need_rec) break;
     }
