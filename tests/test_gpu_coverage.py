@@ -260,3 +260,25 @@ def test_lstm_fused_vs_eager_quality():
 
     assert ev_fused > 0.45, (ev_fused, ev_eager)
     assert ev_fused > ev_eager - 0.15, (ev_fused, ev_eager)
+
+
+def test_fused_forward_fuzz():
+    """Seeded randomized shape fuzz of the fused forward vs fp32 torch:
+    widths 2..128, 1..5 layers, mixed activations, awkward batch sizes."""
+    rng = np.random.default_rng(123)
+    acts = ["tanh", "relu", "sigmoid", "linear"]
+    for trial in range(25):
+        L = int(rng.integers(1, 6))
+        widths = [int(rng.integers(2, 129)) for _ in range(L + 1)]
+        funcs = [str(rng.choice(acts)) for _ in range(L - 1)]
+        out_func = str(rng.choice(acts))
+        B = int(rng.integers(1, 200))
+        torch.manual_seed(trial)
+        net = FeedForwardNet(widths[0], widths[-1], widths[1:-1], funcs, out_func)
+        X = rng.normal(size=(B, widths[0])).astype(np.float32)
+        ref = net(torch.as_tensor(X)).detach().numpy()
+        out = FusedHourglassForward(net, DEV).forward(torch.as_tensor(X)).cpu().numpy()
+        np.testing.assert_allclose(
+            out, ref, atol=6e-2, rtol=6e-2,
+            err_msg=f"trial {trial}: widths={widths} funcs={funcs}+{out_func} B={B}",
+        )

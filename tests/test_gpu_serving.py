@@ -84,3 +84,53 @@ def test_fused_predict_matches_eager(runtime_pair):
     np.testing.assert_allclose(
         fused.predict(X), eager.predict(X), rtol=5e-2, atol=3e-2
     )
+
+
+def test_wide_model_serving_fused(tmp_path):
+    """100-tag machine: builder -> artifact -> fused wide serving."""
+    from gordo_amd.builder import ModelBuilder
+    from gordo_amd.machine import Machine
+    from gordo_amd.server.model_io import ModelRuntime
+
+    config = {
+        "name": "wide-serving",
+        "dataset": {
+            "type": "RandomDataset",
+            "train_start_date": "2020-01-01T00:00:00Z",
+            "train_end_date": "2020-01-15T00:00:00Z",
+            "tag_list": [f"tag-{i}" for i in range(100)],
+        },
+        "model": {
+            "gordo_amd.model.anomaly.diff.DiffBasedAnomalyDetector": {
+                "base_estimator": {
+                    "sklearn.pipeline.Pipeline": {
+                        "steps": [
+                            "sklearn.preprocessing.MinMaxScaler",
+                            {
+                                "gordo_amd.model.models.TorchAutoEncoder": {
+                                    "kind": "feedforward_hourglass",
+                                    "encoding_layers": 2,
+                                    "epochs": 2,
+                                    "batch_size": 512,
+                                }
+                            },
+                        ]
+                    }
+                }
+            }
+        },
+        "evaluation": {"cv_mode": "no_cv"},
+    }
+    out = tmp_path / "wide-serving"
+    ModelBuilder(Machine.from_config(config)).build(output_dir=str(out))
+    rt = ModelRuntime(str(out))
+    assert rt._fused is not None, "wide fused serving path not active"
+    assert rt._fused.p.wmax == 128
+    rng = np.random.default_rng(0)
+    X = pd.DataFrame(
+        rng.uniform(0, 1, size=(200, 100)).astype(np.float32),
+        columns=[f"tag-{i}" for i in range(100)],
+    )
+    df = rt.anomaly(X, None)
+    tot = df[("total-anomaly-score", "")].values
+    assert np.isfinite(tot).all() and (tot >= 0).all()
