@@ -365,7 +365,7 @@ class LSTMStepTrainer:
                 self.dp.broadcast(t)
             p.refresh_bf16()
         T = module.lookback_window
-        Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device)
+        Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device, p.W)
         n_windows = Xd.shape[0] - T + 1 - lookahead
         B = max(64, min(batch_size, n_windows))
         self.B = B
