@@ -9,6 +9,7 @@ from __future__ import annotations
 
 import logging
 import math
+import os
 import time
 from typing import List, Optional
 
@@ -165,7 +166,8 @@ class FusedHourglassTrainer:
 
         # ---- optional hipGraph capture of the whole step -----------------
         graph = None
-        if getattr(cfg, "use_graph", True) and n_batches >= 1:
+        if (getattr(cfg, "use_graph", True) and n_batches >= 1
+                and not os.environ.get("GORDO_NO_GRAPH")):
             try:
                 # warm-up on a side stream (allocations + kernel modules)
                 s = torch.cuda.Stream()

@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import logging
 import math
+import os
 import time
 import numpy as np
 import torch
@@ -118,7 +119,8 @@ class WideLSTMTrainer:
         # ---- hipGraph capture of the whole fwd+bwd (static shapes) --------
         graph = None
         static_w = static_tgt = static_loss = None
-        if device.type == "cuda" and getattr(cfg, "use_graph", True):
+        if (device.type == "cuda" and getattr(cfg, "use_graph", True)
+                and not os.environ.get("GORDO_NO_GRAPH")):
             try:
                 static_w = torch.zeros(B, T, F_in, device=device)
                 static_tgt = torch.zeros(B, F_out, device=device)
