@@ -64,7 +64,8 @@ class DPContext:
 
     def barrier(self):
         if self.enabled:
-            if self.device.type == "cuda":
+            # device_ids is only meaningful (and only accepted) for nccl
+            if self.device.type == "cuda" and self.backend == "nccl":
                 dist.barrier(device_ids=[self.device.index])
             else:
                 dist.barrier()
