@@ -689,10 +689,17 @@ static void ga_fill_acts(int* dst, const std::vector<int64_t>& acts, int L) {
         } else if ((WMAX) == 128) {                           \
             constexpr int W = 128;                            \
             __VA_ARGS__;                                      \
+        } else if ((WMAX) == 256) {                           \
+            constexpr int W = 256;                            \
+            __VA_ARGS__;                                      \
         } else {                                              \
             TORCH_CHECK(false, "unsupported width ", (WMAX)); \
         }                                                     \
     } while (0)
+
+// dynamic-LDS opt-in above the 64 KiB default (gfx950 allows a full 160 KiB
+// workgroup); tracked per kernel instantiation (width tier)
+static inline int hg_wtier(int w) { return w == 64 ? 0 : (w == 128 ? 1 : 2); }
 
 void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
                           torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
@@ -731,12 +738,15 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
     HG_DISPATCH_W(wmax, {
         constexpr size_t TB = HG_MT * W * 2;
         size_t smem = (size_t)(L + 4) * TB + GA_MAXL * W * 4 + 256 * 4;
-        static bool cfg64 = false, cfg128 = false;
-        bool& cfg = (W == 64) ? cfg64 : cfg128;
-        if (!cfg) {
+        TORCH_CHECK(smem <= 160 * 1024, "hourglass L=", L, " W=", W,
+                    " exceeds the 160 KiB LDS budget");
+        static size_t cfg[3] = {0, 0, 0};
+        size_t& c = cfg[hg_wtier(W)];
+        if (smem > 65536 && c < smem) {
             (void)hipFuncSetAttribute((const void*)hg_train_step<W>,
-                                      hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
-            cfg = true;
+                                      hipFuncAttributeMaxDynamicSharedMemorySize,
+                                      (int)smem);
+            c = smem;
         }
         hipLaunchKernelGGL(hg_train_step<W>, dim3(n_tiles), dim3(256), smem,
                            stream.stream(), a);
@@ -803,12 +813,13 @@ void hg_forward_launch(torch::Tensor X, torch::Tensor w, torch::Tensor bias,
     auto stream = at::hip::getCurrentHIPStream();
     HG_DISPATCH_W(wmax, {
         size_t smem = 3 * (size_t)(64 * W * 2) + 4 * 64 * 4;
-        static bool cfg64 = false, cfg128 = false;
-        bool& cfg = (W == 64) ? cfg64 : cfg128;
-        if (!cfg) {
+        static size_t cfg[3] = {0, 0, 0};
+        size_t& c = cfg[hg_wtier(W)];
+        if (smem > 65536 && c < smem) {
             (void)hipFuncSetAttribute((const void*)hg_forward<W>,
-                                      hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
-            cfg = true;
+                                      hipFuncAttributeMaxDynamicSharedMemorySize,
+                                      (int)smem);
+            c = smem;
         }
         hipLaunchKernelGGL(hg_forward<W>, dim3(n_tiles), dim3(256), smem,
                            stream.stream(), a);

@@ -22,21 +22,25 @@ from ..runtime.trainer import TrainHistory
 logger = logging.getLogger(__name__)
 
 WMAX = 64         # narrow-variant padded width (also the LSTM head width)
-WMAX_WIDE = 128   # wide-variant padded width (> this -> library-GEMM fallback)
+WMAX_WIDE = 128   # wide-variant padded width
+WMAX_MAX = 256    # widest instantiation (> this -> library-GEMM fallback)
 MAXL = 5
 ACT_IDS = {"linear": 0, None: 0, "tanh": 1, "relu": 2, "sigmoid": 3}
 
 
 def wmax_for(widths) -> int:
-    """Padded kernel width for a stack of layer widths."""
-    return WMAX if max(widths) <= WMAX else WMAX_WIDE
+    """Padded kernel width (64/128/256 instantiation tier) for a layer stack."""
+    w = max(widths)
+    if w <= WMAX:
+        return WMAX
+    return WMAX_WIDE if w <= WMAX_WIDE else WMAX_MAX
 
 
 def supported(module) -> Optional[str]:
     """None if the fused path supports this FeedForwardNet, else the reason."""
     widths = module.layer_widths
-    if max(widths) > WMAX_WIDE:
-        return f"width {max(widths)} > {WMAX_WIDE}"
+    if max(widths) > WMAX_MAX:
+        return f"width {max(widths)} > {WMAX_MAX}"
     L = len(widths) - 1
     if L < 1 or L > MAXL:
         return f"L={L} outside 1..{MAXL}"

@@ -262,14 +262,91 @@ def test_lstm_fused_vs_eager_quality():
     assert ev_fused > ev_eager - 0.15, (ev_fused, ev_eager)
 
 
+# ---------------------------------------------------------------------------
+# x-wide variant (W=256): tag sets above 128 use the 256-padded kernels
+# ---------------------------------------------------------------------------
+def test_fused_forward_x256():
+    torch.manual_seed(11)
+    net = FeedForwardNet(200, 200, [150, 100, 150], ["tanh"] * 3, "linear")
+    X = np.random.default_rng(11).normal(size=(211, 200)).astype(np.float32)
+    ref = net(torch.as_tensor(X)).detach().numpy()
+    out = FusedHourglassForward(net, DEV).forward(torch.as_tensor(X)).cpu().numpy()
+    np.testing.assert_allclose(out, ref, atol=6e-2, rtol=6e-2)
+
+
+def test_fused_training_x256_matches_autograd():
+    """W=256 grads vs fp32 autograd on one step (200-tag hourglass layer)."""
+    from gordo_amd.ops import ext
+    from gordo_amd.ops.hourglass_engine import _PackedParams, pad_rows_bf16
+
+    torch.manual_seed(12)
+    net = FeedForwardNet(200, 200, [144], ["tanh"], "linear")
+    ref_net = FeedForwardNet(200, 200, [144], ["tanh"], "linear")
+    ref_net.load_state_dict(net.state_dict())
+    X = np.random.default_rng(12).normal(size=(256, 200)).astype(np.float32)
+
+    E = ext()
+    p = _PackedParams(net, DEV)
+    assert p.wmax == 256
+    WM, L, F_out = p.wmax, p.L, 200
+    B = 256
+    MT = int(E.MT)
+    n_tiles = (B + MT - 1) // MT
+    Xd = pad_rows_bf16(X, DEV, WM)
+    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
+    db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+    loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
+    gscale = 2.0 / (B * F_out)
+    E.hg_train_step(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, True, None, WM, 0, 0)
+    torch.cuda.synchronize()
+
+    Xref = Xd[:, :200].float().cpu()
+    out = ref_net(Xref)
+    loss_ref = torch.nn.functional.mse_loss(out, Xref)
+    loss_ref.backward()
+    kernel_loss = float(loss.item()) / (B * F_out)
+    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+
+    linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
+    dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
+    db_sum = db.sum(dim=0).view(L, WM).cpu()
+    for i, lin in enumerate(linears):
+        o, ninp = lin.weight.shape
+        got_w = dw_sum[i, :ninp, :o].T  # kernel stores [in][out]
+        ref_w = lin.weight.grad.detach()
+        scale = ref_w.abs().max().item() or 1.0
+        assert (got_w - ref_w).abs().max().item() < 0.05 * scale + 1e-4, f"layer {i} dW"
+        got_b = db_sum[i, :o]
+        ref_b = lin.bias.grad.detach()
+        bscale = ref_b.abs().max().item() or 1.0
+        assert (got_b - ref_b).abs().max().item() < 0.05 * bscale + 1e-4, f"layer {i} db"
+        assert dw_sum[i, ninp:, :].abs().max().item() == 0.0
+        assert dw_sum[i, :, o:].abs().max().item() == 0.0
+
+
+def test_fused_trainer_x256_learns():
+    rng = np.random.default_rng(13)
+    Z = rng.normal(size=(4096, 16)).astype(np.float32)
+    P = rng.normal(size=(16, 200)).astype(np.float32)
+    X = np.tanh(Z @ P)
+    torch.manual_seed(13)
+    net = FeedForwardNet(200, 200, [150, 100, 150], ["tanh"] * 3, "linear")
+    cfg = TrainConfig(epochs=20, batch_size=1024, lr=3e-3, seed=0)
+    hist = FusedHourglassTrainer(net, DEV, cfg).fit(X, X)
+    assert hist.engine.startswith("fused_hip")
+    assert hist.loss[-1] < hist.loss[0] * 0.7, hist.loss
+
+
 def test_fused_forward_fuzz():
     """Seeded randomized shape fuzz of the fused forward vs fp32 torch:
-    widths 2..128, 1..5 layers, mixed activations, awkward batch sizes."""
+    widths 2..256, 1..5 layers, mixed activations, awkward batch sizes."""
     rng = np.random.default_rng(123)
     acts = ["tanh", "relu", "sigmoid", "linear"]
     for trial in range(25):
         L = int(rng.integers(1, 6))
-        widths = [int(rng.integers(2, 129)) for _ in range(L + 1)]
+        widths = [int(rng.integers(2, 257)) for _ in range(L + 1)]
         funcs = [str(rng.choice(acts)) for _ in range(L - 1)]
         out_func = str(rng.choice(acts))
         B = int(rng.integers(1, 200))

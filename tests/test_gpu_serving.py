@@ -86,6 +86,61 @@ def test_fused_predict_matches_eager(runtime_pair):
     )
 
 
+def test_x256_model_serving_fused(tmp_path):
+    """200-tag machine (W=256 tier): builder -> artifact -> fused serving,
+    with anomaly parity against the eager detector."""
+    from gordo_amd.builder import ModelBuilder
+    from gordo_amd.machine import Machine
+    from gordo_amd.server.model_io import ModelRuntime
+
+    config = {
+        "name": "x256-serving",
+        "dataset": {
+            "type": "RandomDataset",
+            "train_start_date": "2020-01-01T00:00:00Z",
+            "train_end_date": "2020-01-10T00:00:00Z",
+            "tag_list": [f"tag-{i}" for i in range(200)],
+        },
+        "model": {
+            "gordo_amd.model.anomaly.diff.DiffBasedAnomalyDetector": {
+                "base_estimator": {
+                    "sklearn.pipeline.Pipeline": {
+                        "steps": [
+                            "sklearn.preprocessing.MinMaxScaler",
+                            {
+                                "gordo_amd.model.models.TorchAutoEncoder": {
+                                    "kind": "feedforward_hourglass",
+                                    "encoding_layers": 2,
+                                    "epochs": 2,
+                                    "batch_size": 512,
+                                }
+                            },
+                        ]
+                    }
+                }
+            }
+        },
+        "evaluation": {"cv_mode": "no_cv"},
+    }
+    out = tmp_path / "x256-serving"
+    ModelBuilder(Machine.from_config(config)).build(output_dir=str(out))
+    fused = ModelRuntime(str(out))
+    assert fused._fused is not None, "W=256 fused serving path not active"
+    assert fused._fused.p.wmax == 256
+    eager = ModelRuntime(str(out))
+    eager._fused = None
+    rng = np.random.default_rng(0)
+    X = pd.DataFrame(
+        rng.uniform(0, 1, size=(200, 200)).astype(np.float32),
+        columns=[f"tag-{i}" for i in range(200)],
+    )
+    df_f = fused.anomaly(X, None)
+    df_e = eager.anomaly(X, None)
+    tot_f = df_f[("total-anomaly-score", "")].values
+    tot_e = df_e[("total-anomaly-score", "")].values
+    np.testing.assert_allclose(tot_f, tot_e, rtol=6e-2, atol=6e-2)
+
+
 def test_wide_model_serving_fused(tmp_path):
     """100-tag machine: builder -> artifact -> fused wide serving."""
     from gordo_amd.builder import ModelBuilder
