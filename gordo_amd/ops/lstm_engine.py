@@ -412,6 +412,29 @@ class LSTMStepTrainer:
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 
+    # mid-training checkpoint: masters + Adam state + batch
+    # cursor, so a restarted DP job resumes the epoch where it stopped
+    def state_dict(self) -> dict:
+        return {
+            "masters": [t.detach().cpu() for t in self.p.masters()],
+            "opt": self.opt.state_dict(),
+            "s": self._s,
+        }
+
+    def load_state_dict(self, sd: dict):
+        for dst, src in zip(self.p.masters(), sd["masters"]):
+            with torch.no_grad():
+                dst.copy_(src.to(self.device))
+        self.opt.load_state_dict(sd["opt"])
+        self._s = int(sd["s"])
+        self.p.refresh_bf16()
+
+    def save_checkpoint(self, path: str):
+        torch.save(self.state_dict(), path)
+
+    def load_checkpoint(self, path: str):
+        self.load_state_dict(torch.load(path, weights_only=False))
+
 
 class FusedLSTMForward:
     """Serving/predict forward through the fused cell kernels."""

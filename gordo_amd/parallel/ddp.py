@@ -285,3 +285,51 @@ class HourglassStepTrainer:
         if self.fused:
             return self.p.master.detach().clone()
         return torch.nn.utils.parameters_to_vector(self.dp_params).detach().clone()
+
+    # ------------------------------------------------------------------
+    # Checkpoint/resume: the serializer directory is the
+    # whole-model checkpoint; these cover mid-training state for long DP
+    # runs — full optimizer state, so a restarted job continues bit-exact.
+    def state_dict(self) -> dict:
+        if self.fused:
+            return {
+                "fused": True,
+                "master": self.p.master.detach().cpu(),
+                "m": self.m.detach().cpu(),
+                "v": self.v.detach().cpu(),
+                "counters": self.counters.detach().cpu(),
+            }
+        return {
+            "fused": False,
+            "params": torch.nn.utils.parameters_to_vector(self.dp_params).detach().cpu(),
+            "opt": self.opt.state_dict(),
+            "gen": self._gen.get_state(),
+        }
+
+    def load_state_dict(self, sd: dict):
+        if self.fused != bool(sd.get("fused")):
+            raise ValueError("checkpoint engine mismatch (fused vs eager)")
+        if self.fused:
+            p = self.p
+            p.master.copy_(sd["master"].to(self.device))
+            self.m.copy_(sd["m"].to(self.device))
+            self.v.copy_(sd["v"].to(self.device))
+            self.counters.copy_(sd["counters"].to(self.device))
+            p.bias.copy_(p.master[p.PW:])
+            WM = p.wmax
+            w = p.master[: p.PW].view(p.L, WM, WM)
+            p.w_bf16.copy_(w.to(torch.bfloat16))
+            p.wt_bf16.copy_(w.transpose(1, 2).contiguous().to(torch.bfloat16))
+        else:
+            torch.nn.utils.vector_to_parameters(
+                sd["params"].to(self.device), self.dp_params
+            )
+            self.opt.load_state_dict(sd["opt"])
+            if "gen" in sd:
+                self._gen.set_state(sd["gen"])
+
+    def save_checkpoint(self, path: str):
+        torch.save(self.state_dict(), path)
+
+    def load_checkpoint(self, path: str):
+        self.load_state_dict(torch.load(path, weights_only=False))

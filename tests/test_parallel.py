@@ -65,3 +65,42 @@ def test_dp_context_single_process():
     t = torch.ones(4)
     dp.all_reduce_mean(t)  # no-op
     assert t.sum().item() == 4.0
+
+
+def test_checkpoint_resume_bit_exact(tmp_path):
+    """save_checkpoint -> more steps -> load_checkpoint -> same steps replays
+    the identical trajectory (params + Adam moments + sampler state)."""
+    from gordo_amd.model.factories import feedforward_hourglass
+    from gordo_amd.parallel.ddp import DPContext, HourglassStepTrainer
+    from gordo_amd.runtime.trainer import TrainConfig
+
+    torch.manual_seed(7)
+    module = feedforward_hourglass(10, 10, encoding_layers=2, compression_factor=0.5)
+    X = np.random.default_rng(7).normal(size=(512, 10)).astype(np.float32)
+    dp = DPContext(device=torch.device("cpu"))
+    trainer = HourglassStepTrainer(module, X, 128, TrainConfig(batch_size=128, lr=1e-3, dtype="fp32"), dp=dp)
+    for _ in range(3):
+        trainer.step()
+    ckpt = tmp_path / "step3.pt"
+    trainer.save_checkpoint(str(ckpt))
+    for _ in range(3):
+        trainer.step()
+    vec1 = trainer.params_vector().numpy()
+    trainer.load_checkpoint(str(ckpt))
+    for _ in range(3):
+        trainer.step()
+    vec2 = trainer.params_vector().numpy()
+    np.testing.assert_array_equal(vec1, vec2)
+
+
+def test_checkpoint_engine_mismatch_rejected(tmp_path):
+    from gordo_amd.model.factories import feedforward_hourglass
+    from gordo_amd.parallel.ddp import DPContext, HourglassStepTrainer
+    from gordo_amd.runtime.trainer import TrainConfig
+
+    module = feedforward_hourglass(10, 10, encoding_layers=2, compression_factor=0.5)
+    X = np.zeros((256, 10), dtype=np.float32)
+    dp = DPContext(device=torch.device("cpu"))
+    trainer = HourglassStepTrainer(module, X, 128, TrainConfig(batch_size=128, dtype="fp32"), dp=dp)
+    with pytest.raises(ValueError, match="mismatch"):
+        trainer.load_state_dict({"fused": True})
