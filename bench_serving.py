@@ -62,11 +62,15 @@ def project_config(n_machines: int, n_tags: int):
 
 
 def measure_http(out_dir: str, args, X):
-    """End-to-end HTTP: uvicorn fleet server in a subprocess, parquet POSTs."""
+    """End-to-end HTTP: uvicorn fleet server in a subprocess; measures both
+    wire formats (parquet for reference parity, arrow IPC = the fast path)."""
     import socket
     import subprocess
     import httpx
-    from gordo_amd.server.utils import dataframe_into_parquet_bytes
+    from gordo_amd.server.utils import (
+        dataframe_into_arrow_bytes,
+        dataframe_into_parquet_bytes,
+    )
 
     with socket.socket() as s:
         s.bind(("127.0.0.1", 0))
@@ -90,26 +94,31 @@ def measure_http(out_dir: str, args, X):
                 except httpx.TransportError:
                     pass
                 time.sleep(0.5)
-            urls = [
-                f"{base}/gordo/v0/bench-fleet/machine-{i:03d}/anomaly/prediction?format=parquet"
-                for i in range(min(8, args.machines))
-            ]
-            for u in urls:  # warm
-                r = client.post(u, content=body,
-                                headers={"content-type": "application/octet-stream"})
-                assert r.status_code == 200, r.text[:200]
-            lat = []
-            for i in range(min(args.requests, 200)):
-                t = time.perf_counter()
-                r = client.post(urls[i % len(urls)], content=body,
-                                headers={"content-type": "application/octet-stream"})
-                assert r.status_code == 200
-                lat.append(time.perf_counter() - t)
-        lat_ms = np.array(lat) * 1e3
-        return {
-            "http_p50_ms": float(np.percentile(lat_ms, 50)),
-            "http_p99_ms": float(np.percentile(lat_ms, 99)),
-        }
+            out = {}
+            abody = dataframe_into_arrow_bytes(X)
+            for fmt, payload, ctype in (
+                ("parquet", body, "application/octet-stream"),
+                ("arrow", abody, "application/vnd.apache.arrow.stream"),
+            ):
+                urls = [
+                    f"{base}/gordo/v0/bench-fleet/machine-{i:03d}/anomaly/prediction?format={fmt}"
+                    for i in range(min(8, args.machines))
+                ]
+                for u in urls:  # warm
+                    r = client.post(u, content=payload, headers={"content-type": ctype})
+                    assert r.status_code == 200, r.text[:200]
+                lat = []
+                for i in range(min(args.requests, 200)):
+                    t = time.perf_counter()
+                    r = client.post(urls[i % len(urls)], content=payload,
+                                    headers={"content-type": ctype})
+                    assert r.status_code == 200
+                    lat.append(time.perf_counter() - t)
+                lat_ms = np.array(lat) * 1e3
+                key = "http_" + ("" if fmt == "parquet" else "arrow_")
+                out[key + "p50_ms"] = float(np.percentile(lat_ms, 50))
+                out[key + "p99_ms"] = float(np.percentile(lat_ms, 99))
+        return out
     finally:
         proc.terminate()
         try:

@@ -37,6 +37,7 @@ class Client:
         forward_resampled_sensors: bool = False,
         n_retries: int = 5,
         use_parquet: bool = True,
+        wire_format: str = None,
         data_provider=None,
         prediction_forwarder=None,
         max_request_hours: float = 24 * 7,
@@ -51,6 +52,10 @@ class Client:
         self.parallelism = parallelism
         self.n_retries = n_retries
         self.use_parquet = use_parquet
+        # "arrow" | "parquet" | "json"; default follows use_parquet (parity
+        # with the reference client's parquet preference); "arrow" is the
+        # lowest-latency option against this package's server
+        self.wire_format = wire_format or ("parquet" if use_parquet else "json")
         self.data_provider = data_provider
         self.prediction_forwarder = prediction_forwarder
         self.forward_resampled_sensors = forward_resampled_sensors
@@ -159,8 +164,10 @@ class Client:
         X: Optional[pd.DataFrame],
     ) -> PredictionResult:
         from ..server.utils import (
+            dataframe_from_arrow_bytes,
             dataframe_from_dict,
             dataframe_from_parquet_bytes,
+            dataframe_into_arrow_bytes,
             dataframe_into_parquet_bytes,
             dataframe_to_dict,
         )
@@ -182,7 +189,12 @@ class Client:
             for attempt in range(self.n_retries + 1):
                 try:
                     async with sem:
-                        if self.use_parquet:
+                        if self.wire_format == "arrow":
+                            raw = await client_io.post_arrow(
+                                url, session, dataframe_into_arrow_bytes(chunk_X)
+                            )
+                            return dataframe_from_arrow_bytes(raw)
+                        if self.wire_format == "parquet":
                             raw = await client_io.post_parquet(
                                 url, session, dataframe_into_parquet_bytes(chunk_X)
                             )

@@ -24,8 +24,10 @@ from .. import __version__
 from ..serializer import dumps
 from .model_io import ModelRuntime
 from .utils import (
+    dataframe_from_arrow_bytes,
     dataframe_from_dict,
     dataframe_from_parquet_bytes,
+    dataframe_into_arrow_bytes,
     dataframe_into_parquet_bytes,
     dataframe_to_dict,
 )
@@ -43,7 +45,10 @@ def build_app(model_dir: str, device=None):
     async def _extract_frames(request: Request):
         ctype = request.headers.get("content-type", "")
         body = await request.body()
-        if "octet-stream" in ctype or "parquet" in ctype:
+        if "arrow" in ctype:
+            X = dataframe_from_arrow_bytes(body)
+            y = None
+        elif "octet-stream" in ctype or "parquet" in ctype:
             X = dataframe_from_parquet_bytes(body)
             y = None
         else:
@@ -79,6 +84,11 @@ def build_app(model_dir: str, device=None):
     def _respond(df: pd.DataFrame, request: Request) -> Response:
         fmt = request.query_params.get("format", "json")
         accept = request.headers.get("accept", "")
+        if fmt == "arrow" or "arrow" in accept:
+            return Response(
+                content=dataframe_into_arrow_bytes(df),
+                media_type="application/vnd.apache.arrow.stream",
+            )
         if fmt == "parquet" or "octet-stream" in accept:
             return Response(
                 content=dataframe_into_parquet_bytes(df),

@@ -124,6 +124,64 @@ def test_parquet_roundtrip_multiindex():
     np.testing.assert_allclose(out.values, df.values)
 
 
+def test_parquet_roundtrip_datetime_index():
+    from gordo_amd.server.utils import (
+        dataframe_from_arrow_bytes,
+        dataframe_into_arrow_bytes,
+    )
+
+    df = pd.DataFrame(
+        np.random.default_rng(0).uniform(size=(5, 3)).astype(np.float32),
+        columns=pd.MultiIndex.from_tuples(
+            [("model-output", "t0"), ("model-output", "t1"), ("total-anomaly-score", "")]
+        ),
+        index=pd.date_range("2021-06-01", periods=5, freq="10min", tz="UTC"),
+    )
+    for enc, dec in (
+        (dataframe_into_parquet_bytes, dataframe_from_parquet_bytes),
+        (dataframe_into_arrow_bytes, dataframe_from_arrow_bytes),
+    ):
+        out = dec(enc(df))
+        assert list(out.columns) == list(df.columns)
+        assert isinstance(out.index, pd.DatetimeIndex)
+        assert list(out.index) == list(df.index)
+        np.testing.assert_allclose(out.values, df.values, rtol=1e-6)
+
+
+def test_parquet_reads_foreign_pandas_bytes():
+    """Bytes produced by plain df.to_parquet (a non-gordo client) still decode."""
+    import io
+
+    df = pd.DataFrame(
+        {"tag-0": [1.0, 2.0], "tag-1": [3.0, 4.0]},
+        index=pd.date_range("2021-01-01", periods=2, freq="h"),
+    )
+    buf = io.BytesIO()
+    df.to_parquet(buf, engine="pyarrow")
+    out = dataframe_from_parquet_bytes(buf.getvalue())
+    assert isinstance(out.index, pd.DatetimeIndex)
+    np.testing.assert_allclose(out.values, df.values)
+
+
+def test_anomaly_prediction_arrow(client):
+    from gordo_amd.server.utils import (
+        dataframe_from_arrow_bytes,
+        dataframe_into_arrow_bytes,
+    )
+
+    X = _x_frame()
+    r = client.post(
+        "/anomaly/prediction?format=arrow",
+        content=dataframe_into_arrow_bytes(X),
+        headers={"Content-Type": "application/vnd.apache.arrow.stream"},
+    )
+    assert r.status_code == 200
+    assert r.headers["content-type"].startswith("application/vnd.apache.arrow")
+    df = dataframe_from_arrow_bytes(r.content)
+    assert "total-anomaly-score" in set(df.columns.get_level_values(0))
+    assert len(df) == len(X)
+
+
 def test_lstm_model_serving_offsets(tmp_path):
     """An LSTM artifact serves /prediction with lookback-aligned rows."""
     import numpy as np
