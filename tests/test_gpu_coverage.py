@@ -262,6 +262,29 @@ def test_lstm_fused_vs_eager_quality():
     assert ev_fused > ev_eager - 0.15, (ev_fused, ev_eager)
 
 
+def test_lstm_beyond_fused_envelope_gemmloop_gpu():
+    """>128-wide LSTM (200 tags): the fused cells don't apply; the estimator
+    must run the hipGraph GEMM-loop engine on GPU and still learn."""
+    from gordo_amd.model.models import TorchLSTMAutoEncoder
+
+    rng = np.random.default_rng(31)
+    t = np.arange(3000)
+    X = np.stack(
+        [np.sin(t / (8 + i % 13)) + 0.05 * rng.standard_normal(len(t)) for i in range(200)],
+        axis=1,
+    ).astype(np.float32)
+    model = TorchLSTMAutoEncoder(
+        kind="lstm_symmetric", dims=[160], lookback_window=12, epochs=3,
+        batch_size=512, lr=3e-3,
+    )
+    model.fit(X)
+    assert model.history["engine"].startswith("wide_lstm_gemmloop"), model.history
+    assert model.history["loss"][-1] < model.history["loss"][0]
+    out = model.predict(X[:100])
+    assert out.shape == (100 - 12 + 1, 200)
+    assert np.isfinite(out).all()
+
+
 def test_fused_checkpoint_resume_bit_exact(tmp_path):
     """Fused trainer checkpoint: the device-side step/epoch counters + Feistel
     key are part of the state, so a resumed run replays the identical
