@@ -33,7 +33,8 @@ struct LstmFwdArgs {
     const unsigned short* __restrict__ wcat;  // [256][128] bf16
     const float* __restrict__ bias;           // [256] fp32 (b_ih + b_hh)
     unsigned short* __restrict__ h_seq;       // [B][T][64] out
-    unsigned short* __restrict__ gates_seq;   // [B][T][256] out (activated i,f,g,o)
+    unsigned char* __restrict__ gates_seq;    // [B][T][256] u8 out (activated
+                                              // i,f,o as v*255; g as (g+1)/2*255)
     unsigned short* __restrict__ c_seq;       // [B][T][64] out
     unsigned short* __restrict__ x_seq_out;   // [B][T][64] gathered x copy (layer0) or null
     int B, T, N;
@@ -166,7 +167,8 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
                 ga_u16x8 vf = *(ga_u16x8*)&pre_tiles[GA_TILE_BYTES + ga_sw(em, cc)];
                 ga_u16x8 vg = *(ga_u16x8*)&pre_tiles[2 * GA_TILE_BYTES + ga_sw(em, cc)];
                 ga_u16x8 vo = *(ga_u16x8*)&pre_tiles[3 * GA_TILE_BYTES + ga_sw(em, cc)];
-                ga_u16x8 oh, oc, oi, of, og, oo;
+                ga_u16x8 oh, oc;
+                float qi[8], qf[8], qg[8], qo[8];
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
                     float i_g = ga_sigmoid(ga_bf16_to_f32(vi[e]));
@@ -178,19 +180,19 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
                     c_reg[ei] = c_new;
                     oh[e] = ga_f32_to_bf16(o_g * ga_tanh(c_new));
                     oc[e] = ga_f32_to_bf16(c_new);
-                    oi[e] = ga_f32_to_bf16(i_g);
-                    of[e] = ga_f32_to_bf16(f_g);
-                    og[e] = ga_f32_to_bf16(g_g);
-                    oo[e] = ga_f32_to_bf16(o_g);
+                    qi[e] = i_g;
+                    qf[e] = f_g;
+                    qg[e] = g_g * 0.5f + 0.5f;
+                    qo[e] = o_g;
                 }
                 *(ga_u16x8*)&h_tile[ga_sw(em, cc)] = oh;
                 if (live) {
                     *(ga_u16x8*)(a.h_seq + bt * GA_WMAX + cc) = oh;
                     *(ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc) = oc;
-                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc) = oi;
-                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc) = of;
-                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc) = og;
-                    *(ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc) = oo;
+                    *(uint2*)(a.gates_seq + bt * LSTM_4H + cc) = ga_q8_pack8(qi);
+                    *(uint2*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc) = ga_q8_pack8(qf);
+                    *(uint2*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc) = ga_q8_pack8(qg);
+                    *(uint2*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc) = ga_q8_pack8(qo);
                 }
             }
         }
@@ -205,7 +207,7 @@ __global__ void lstm_fwd(LstmFwdArgs a) {
 // ===========================================================================
 struct LstmBwdArgs {
     const unsigned short* __restrict__ dh_out;     // [B][T][64] or [B][64]
-    const unsigned short* __restrict__ gates_seq;  // [B][T][256]
+    const unsigned char* __restrict__ gates_seq;   // [B][T][256] u8 (fwd's layout)
     const unsigned short* __restrict__ c_seq;      // [B][T][64]
     const unsigned short* __restrict__ wcat_t;     // [128][256] bf16 (WcatT)
     unsigned short* __restrict__ dgates_seq;       // [B][T][256] out
@@ -256,19 +258,22 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
     // the following timestep so the HBM latency is off the critical path)
     const int pgi = base + em;
     const bool plive = pgi < a.B;
-    ga_u16x8 pvi[2], pvf[2], pvg[2], pvo[2], pvc[2], pvcp[2], pvdh[2];
+    uint2 pvi[2], pvf[2], pvg[2], pvo[2];
+    ga_u16x8 pvc[2], pvcp[2], pvdh[2];
     auto preload = [&](int t) {
         const size_t bt = (size_t)pgi * a.T + t;
         const ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        const uint2 zq = {0, 0};
 #pragma unroll
         for (int hh = 0; hh < 2; ++hh) {
             int cc = c0 + hh * 8;
-            pvi[hh] = pvf[hh] = pvg[hh] = pvo[hh] = pvc[hh] = pvcp[hh] = pvdh[hh] = z;
+            pvi[hh] = pvf[hh] = pvg[hh] = pvo[hh] = zq;
+            pvc[hh] = pvcp[hh] = pvdh[hh] = z;
             if (plive) {
-                pvi[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + cc);
-                pvf[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
-                pvg[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
-                pvo[hh] = *(const ga_u16x8*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
+                pvi[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + cc);
+                pvf[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
+                pvg[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
+                pvo[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
                 pvc[hh] = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
                 if (t > 0) pvcp[hh] = *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc);
                 if (a.dh_last_only) {
@@ -292,17 +297,17 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
 #pragma unroll
             for (int hh = 0; hh < 2; ++hh) {
                 int cc = c0 + hh * 8;
-                ga_u16x8 vi = pvi[hh], vf = pvf[hh], vg = pvg[hh], vo = pvo[hh],
-                         vc = pvc[hh], vcp = pvcp[hh], vdh = pvdh[hh];
+                uint2 vi = pvi[hh], vf = pvf[hh], vg = pvg[hh], vo = pvo[hh];
+                ga_u16x8 vc = pvc[hh], vcp = pvcp[hh], vdh = pvdh[hh];
                 ga_u16x8 vrec = *(ga_u16x8*)&dh_tile[ga_sw(em, cc)];
                 ga_u16x8 odi, odf, odg, odo;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
                     float dh = ga_bf16_to_f32(vdh[e]) + ga_bf16_to_f32(vrec[e]);
-                    float i_g = ga_bf16_to_f32(vi[e]);
-                    float f_g = ga_bf16_to_f32(vf[e]);
-                    float g_g = ga_bf16_to_f32(vg[e]);
-                    float o_g = ga_bf16_to_f32(vo[e]);
+                    float i_g = ga_q8_at(vi, e);
+                    float f_g = ga_q8_at(vf, e);
+                    float g_g = ga_q8_at(vg, e) * 2.f - 1.f;
+                    float o_g = ga_q8_at(vo, e);
                     float c_t = ga_bf16_to_f32(vc[e]);
                     float c_prev = ga_bf16_to_f32(vcp[e]);
                     float tanhc = ga_tanh(c_t);
@@ -572,7 +577,8 @@ void lstm_fwd_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
     a.wcat = (const unsigned short*)wcat.data_ptr();
     a.bias = bias.data_ptr<float>();
     a.h_seq = (unsigned short*)h_seq.data_ptr();
-    a.gates_seq = (unsigned short*)gates_seq.data_ptr();
+    TORCH_CHECK(gates_seq.scalar_type() == torch::kByte, "gates_seq must be u8");
+    a.gates_seq = (unsigned char*)gates_seq.data_ptr();
     a.c_seq = (unsigned short*)c_seq.data_ptr();
     a.x_seq_out = x_seq_out.has_value() ? (unsigned short*)x_seq_out->data_ptr() : nullptr;
     a.B = (int)B;
@@ -596,7 +602,8 @@ void lstm_bwd_cell_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
                           int64_t B, int64_t T, bool dh_last_only) {
     LstmBwdArgs a;
     a.dh_out = (const unsigned short*)dh_out.data_ptr();
-    a.gates_seq = (const unsigned short*)gates_seq.data_ptr();
+    TORCH_CHECK(gates_seq.scalar_type() == torch::kByte, "gates_seq must be u8");
+    a.gates_seq = (const unsigned char*)gates_seq.data_ptr();
     a.c_seq = (const unsigned short*)c_seq.data_ptr();
     a.wcat_t = (const unsigned short*)wcat_t.data_ptr();
     a.dgates_seq = (unsigned short*)dgates_seq.data_ptr();
