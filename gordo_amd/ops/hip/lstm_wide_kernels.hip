@@ -35,7 +35,8 @@ struct WFwdArgs {
     const unsigned short* __restrict__ wcat;  // [512][256]
     const float* __restrict__ bias;           // [512]
     unsigned short* __restrict__ h_seq;       // [B][T][128]
-    unsigned short* __restrict__ gates_seq;   // [B][T][512]
+    unsigned char* __restrict__ gates_seq;    // [B][T][512] u8 (i,f,o v*255;
+                                              // g (g+1)/2*255 — see common.h)
     unsigned short* __restrict__ c_seq;       // [B][T][128]
     unsigned short* __restrict__ x_seq_out;   // [B][T][128] or null
     int B, T;
@@ -133,7 +134,8 @@ __global__ void lstm_fwd_wide(WFwdArgs a) {
                 ga_u16x8 vf = *(ga_u16x8*)&pre_tiles[WTB + sw128(em, cc)];
                 ga_u16x8 vg = *(ga_u16x8*)&pre_tiles[2 * WTB + sw128(em, cc)];
                 ga_u16x8 vo = *(ga_u16x8*)&pre_tiles[3 * WTB + sw128(em, cc)];
-                ga_u16x8 oh, oc, oi, of, og, oo;
+                ga_u16x8 oh, oc;
+                float qi[8], qf[8], qg[8], qo[8];
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
                     float i_g = ga_sigmoid(ga_bf16_to_f32(vi[e]));
@@ -145,19 +147,19 @@ __global__ void lstm_fwd_wide(WFwdArgs a) {
                     c_reg[ei] = c_new;
                     oh[e] = ga_f32_to_bf16(o_g * ga_tanh(c_new));
                     oc[e] = ga_f32_to_bf16(c_new);
-                    oi[e] = ga_f32_to_bf16(i_g);
-                    of[e] = ga_f32_to_bf16(f_g);
-                    og[e] = ga_f32_to_bf16(g_g);
-                    oo[e] = ga_f32_to_bf16(o_g);
+                    qi[e] = i_g;
+                    qf[e] = f_g;
+                    qg[e] = g_g * 0.5f + 0.5f;
+                    qo[e] = o_g;
                 }
                 *(ga_u16x8*)&h_tile[sw128(em, cc)] = oh;
                 if (live) {
                     *(ga_u16x8*)(a.h_seq + bt * WW + cc) = oh;
                     *(ga_u16x8*)(a.c_seq + bt * WW + cc) = oc;
-                    *(ga_u16x8*)(a.gates_seq + bt * W4H + cc) = oi;
-                    *(ga_u16x8*)(a.gates_seq + bt * W4H + WW + cc) = of;
-                    *(ga_u16x8*)(a.gates_seq + bt * W4H + 2 * WW + cc) = og;
-                    *(ga_u16x8*)(a.gates_seq + bt * W4H + 3 * WW + cc) = oo;
+                    *(uint2*)(a.gates_seq + bt * W4H + cc) = ga_q8_pack8(qi);
+                    *(uint2*)(a.gates_seq + bt * W4H + WW + cc) = ga_q8_pack8(qf);
+                    *(uint2*)(a.gates_seq + bt * W4H + 2 * WW + cc) = ga_q8_pack8(qg);
+                    *(uint2*)(a.gates_seq + bt * W4H + 3 * WW + cc) = ga_q8_pack8(qo);
                 }
             }
         }
@@ -171,7 +173,7 @@ __global__ void lstm_fwd_wide(WFwdArgs a) {
 // ===========================================================================
 struct WBwdArgs {
     const unsigned short* __restrict__ dh_out;     // [B][T][128] or [B][128]
-    const unsigned short* __restrict__ gates_seq;  // [B][T][512]
+    const unsigned char* __restrict__ gates_seq;   // [B][T][512] u8
     const unsigned short* __restrict__ c_seq;      // [B][T][128]
     const unsigned short* __restrict__ wcat_t;     // [256][512]
     unsigned short* __restrict__ dgates_seq;       // [B][T][512] out
@@ -211,12 +213,13 @@ __global__ void lstm_bwd_wide(WBwdArgs a) {
 #pragma unroll
             for (int hh = 0; hh < 4; ++hh) {
                 int cc = c0 + hh * 8;
-                ga_u16x8 vi = z, vf = z, vg = z, vo = z, vc = z, vcp = z, vdh = z;
+                uint2 vi = {0, 0}, vf = {0, 0}, vg = {0, 0}, vo = {0, 0};
+                ga_u16x8 vc = z, vcp = z, vdh = z;
                 if (live) {
-                    vi = *(const ga_u16x8*)(a.gates_seq + bt * W4H + cc);
-                    vf = *(const ga_u16x8*)(a.gates_seq + bt * W4H + WW + cc);
-                    vg = *(const ga_u16x8*)(a.gates_seq + bt * W4H + 2 * WW + cc);
-                    vo = *(const ga_u16x8*)(a.gates_seq + bt * W4H + 3 * WW + cc);
+                    vi = *(const uint2*)(a.gates_seq + bt * W4H + cc);
+                    vf = *(const uint2*)(a.gates_seq + bt * W4H + WW + cc);
+                    vg = *(const uint2*)(a.gates_seq + bt * W4H + 2 * WW + cc);
+                    vo = *(const uint2*)(a.gates_seq + bt * W4H + 3 * WW + cc);
                     vc = *(const ga_u16x8*)(a.c_seq + bt * WW + cc);
                     if (t > 0) vcp = *(const ga_u16x8*)(a.c_seq + (bt - 1) * WW + cc);
                     if (a.dh_last_only) {
@@ -231,10 +234,10 @@ __global__ void lstm_bwd_wide(WBwdArgs a) {
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
                     float dh = ga_bf16_to_f32(vdh[e]) + ga_bf16_to_f32(vrec[e]);
-                    float i_g = ga_bf16_to_f32(vi[e]);
-                    float f_g = ga_bf16_to_f32(vf[e]);
-                    float g_g = ga_bf16_to_f32(vg[e]);
-                    float o_g = ga_bf16_to_f32(vo[e]);
+                    float i_g = ga_q8_at(vi, e);
+                    float f_g = ga_q8_at(vf, e);
+                    float g_g = ga_q8_at(vg, e) * 2.f - 1.f;
+                    float o_g = ga_q8_at(vo, e);
                     float c_t = ga_bf16_to_f32(vc[e]);
                     float c_prev = ga_bf16_to_f32(vcp[e]);
                     float tanhc = ga_tanh(c_t);
@@ -522,7 +525,8 @@ void lstm_fwd_wide_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
     a.wcat = (const unsigned short*)wcat.data_ptr();
     a.bias = bias.data_ptr<float>();
     a.h_seq = (unsigned short*)h_seq.data_ptr();
-    a.gates_seq = (unsigned short*)gates_seq.data_ptr();
+    TORCH_CHECK(gates_seq.scalar_type() == torch::kByte, "gates_seq must be u8");
+    a.gates_seq = (unsigned char*)gates_seq.data_ptr();
     a.c_seq = (unsigned short*)c_seq.data_ptr();
     a.x_seq_out = x_seq_out.has_value() ? (unsigned short*)x_seq_out->data_ptr() : nullptr;
     a.B = (int)B;
@@ -545,7 +549,8 @@ void lstm_bwd_wide_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
                           int64_t B, int64_t T, bool dh_last_only) {
     WBwdArgs a;
     a.dh_out = (const unsigned short*)dh_out.data_ptr();
-    a.gates_seq = (const unsigned short*)gates_seq.data_ptr();
+    TORCH_CHECK(gates_seq.scalar_type() == torch::kByte, "gates_seq must be u8");
+    a.gates_seq = (const unsigned char*)gates_seq.data_ptr();
     a.c_seq = (const unsigned short*)c_seq.data_ptr();
     a.wcat_t = (const unsigned short*)wcat_t.data_ptr();
     a.dgates_seq = (unsigned short*)dgates_seq.data_ptr();

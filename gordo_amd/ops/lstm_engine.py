@@ -218,7 +218,7 @@ class FusedLSTMTrainer:
         p = self.p
         Lc, W, WH4, WKC = p.Lc, p.W, p.H4, p.KCAT
         bf16 = torch.bfloat16
-        gates_dt = torch.uint8 if W == WMAX else bf16
+        gates_dt = torch.uint8  # u8 fixed-point gate streams (common.h ga_q8_*)
         MT = int(getattr(self.E, "MT", 32))
         n_head_tiles = (B + MT - 1) // MT
         st = {
@@ -227,9 +227,8 @@ class FusedLSTMTrainer:
             "perm": torch.zeros(n_batches * B, dtype=torch.int32, device=device),
             "h_seq": [torch.zeros(B, T, W, dtype=bf16, device=device) for _ in range(Lc)],
             "c_seq": [torch.zeros(B, T, W, dtype=bf16, device=device) for _ in range(Lc)],
-            # narrow cells store activated gates quantized u8 (common.h
-            # ga_q8_*: i,f,o as v*255, g offset) — halves the biggest BPTT
-            # stream; the W=128 kernels keep bf16 gates
+            # activated gates stored quantized u8 (common.h ga_q8_*: i,f,o
+            # as v*255, g offset) — halves the biggest BPTT stream
             "gates_seq": [torch.zeros(B, T, WH4, dtype=gates_dt, device=device) for _ in range(Lc)],
             "x_seq0": torch.zeros(B, T, W, dtype=bf16, device=device),
             "dgates": torch.zeros(B, T, WH4, dtype=bf16, device=device),
@@ -467,8 +466,7 @@ class FusedLSTMForward:
         B = min(batch_size, n_windows)
         h_seq = [torch.zeros(B, T, p.W, dtype=bf16, device=self.device) for _ in range(p.Lc)]
         c_seq = torch.zeros(B, T, p.W, dtype=bf16, device=self.device)
-        gates_dt = torch.uint8 if p.W == WMAX else bf16
-        gates = torch.zeros(B, T, p.H4, dtype=gates_dt, device=self.device)
+        gates = torch.zeros(B, T, p.H4, dtype=torch.uint8, device=self.device)
         k_fwd = self.E.lstm_fwd if p.W == WMAX else self.E.lstm_fwd_wide
         for s in range(0, n_windows, B):
             n = min(B, n_windows - s)
