@@ -260,6 +260,9 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
     const bool plive = pgi < a.B;
     uint2 pvi[2], pvf[2], pvg[2], pvo[2];
     ga_u16x8 pvc[2], pvcp[2], pvdh[2];
+    // pvc (c_t) is only LOADED once at t = T-1; afterwards it's carried from
+    // the previous iteration's pvcp — c_{t-1} would otherwise be read twice
+    // (as c_prev at t, then as c_t at t-1), a full re-read of the c stream.
     auto preload = [&](int t) {
         const size_t bt = (size_t)pgi * a.T + t;
         const ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -268,13 +271,12 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
         for (int hh = 0; hh < 2; ++hh) {
             int cc = c0 + hh * 8;
             pvi[hh] = pvf[hh] = pvg[hh] = pvo[hh] = zq;
-            pvc[hh] = pvcp[hh] = pvdh[hh] = z;
+            pvcp[hh] = pvdh[hh] = z;
             if (plive) {
                 pvi[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + cc);
                 pvf[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + GA_WMAX + cc);
                 pvg[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + 2 * GA_WMAX + cc);
                 pvo[hh] = *(const uint2*)(a.gates_seq + bt * LSTM_4H + 3 * GA_WMAX + cc);
-                pvc[hh] = *(const ga_u16x8*)(a.c_seq + bt * GA_WMAX + cc);
                 if (t > 0) pvcp[hh] = *(const ga_u16x8*)(a.c_seq + (bt - 1) * GA_WMAX + cc);
                 if (a.dh_last_only) {
                     if (t == a.T - 1)
@@ -285,6 +287,16 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
             }
         }
     };
+    {
+        const ga_u16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        const size_t bt0 = (size_t)pgi * a.T + (a.T - 1);
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) {
+            pvc[hh] = z;
+            if (plive)
+                pvc[hh] = *(const ga_u16x8*)(a.c_seq + bt0 * GA_WMAX + c0 + hh * 8);
+        }
+    }
     preload(a.T - 1);
     __syncthreads();
 
@@ -299,6 +311,7 @@ __global__ void lstm_bwd_cell(LstmBwdArgs a) {
                 int cc = c0 + hh * 8;
                 uint2 vi = pvi[hh], vf = pvf[hh], vg = pvg[hh], vo = pvo[hh];
                 ga_u16x8 vc = pvc[hh], vcp = pvcp[hh], vdh = pvdh[hh];
+                pvc[hh] = vcp;  // carry: c_{t-1} becomes next iteration's c_t
                 ga_u16x8 vrec = *(ga_u16x8*)&dh_tile[ga_sw(em, cc)];
                 ga_u16x8 odi, odf, odg, odo;
 #pragma unroll
