@@ -96,6 +96,61 @@ def _run_pair(target, n_out=3):
     return results
 
 
+def _rccl_ws1_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = "0"
+    os.environ["LOCAL_RANK"] = "0"
+    os.environ["WORLD_SIZE"] = "1"
+    os.environ.pop("GORDO_NO_GRAPH", None)
+    import torch
+    import torch.distributed as dist
+
+    from gordo_amd.model.factories import feedforward_hourglass
+    from gordo_amd.parallel.ddp import DPContext, HourglassStepTrainer
+    from gordo_amd.runtime.trainer import TrainConfig
+
+    torch.manual_seed(0)
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    dp = DPContext(device=torch.device("cuda:0"))
+    # world_size=1 disables DP by default; force the comm branch so the step
+    # runs hg_partial_flat -> dist.all_reduce (RCCL) -> hg_adam_flat and the
+    # hipGraph capture includes the collective — the 8-rank code path.
+    dp.enabled = True
+    dp.backend = "nccl"
+    module = feedforward_hourglass(20, 20, encoding_layers=2, compression_factor=0.5)
+    X = np.random.default_rng(0).uniform(0, 1, size=(4096, 20)).astype(np.float32)
+    cfg = TrainConfig(batch_size=1024, lr=1e-3, dtype="bf16")
+    trainer = HourglassStepTrainer(module, X, 1024, cfg, dp=dp)
+    captured = trainer._graph is not None
+    for _ in range(10):
+        trainer.step()
+    trainer.synchronize()
+    vec = trainer.params_vector().cpu().numpy()
+    q.put((trainer.fused, captured, bool(np.isfinite(vec).all())))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_gpu_dp_rccl_ws1_graph_capture():
+    """world_size=1 RCCL smoke: a real nccl (RCCL) process
+    group on the single GPU, the fused DP bucket path forced on, and the whole
+    step — INCLUDING the RCCL all_reduce — captured in a hipGraph. This is the
+    code path the driver's 8-rank scaling run takes."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29801 + os.getpid() % 500
+    p = ctx.Process(target=_rccl_ws1_worker, args=(0, 1, port, q))
+    p.start()
+    fused, captured, finite = q.get()
+    p.join(240)
+    assert p.exitcode == 0
+    assert fused, "fused DP path must be active"
+    assert captured, "hipGraph capture of the RCCL all_reduce step failed"
+    assert finite
+
+
 @pytest.mark.timeout(600)
 def test_gpu_dp_hourglass_fused_ranks_stay_in_sync():
     """Broadcast-init + per-step fused bucket all-reduce keeps both ranks'
