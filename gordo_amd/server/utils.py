@@ -157,8 +157,13 @@ def _df_from_table(tab) -> pd.DataFrame:
         ivals = tab.column(iname).to_numpy()
         if iname.startswith(_IDX + "dt"):
             tz = iname[len(_IDX) + 2:] or None
-            df.index = pd.DatetimeIndex(ivals.view("datetime64[ns]"), tz="UTC").tz_convert(tz) if tz \
-                else pd.DatetimeIndex(ivals.view("datetime64[ns]"))
+            idx = pd.DatetimeIndex(ivals.view("datetime64[ns]"))
+            if tz:
+                try:
+                    idx = idx.tz_localize("UTC").tz_convert(tz)
+                except Exception:  # unrecognized tz string: keep UTC
+                    idx = idx.tz_localize("UTC")
+            df.index = idx
         else:
             df.index = ivals
         return _restore_columns(df)
