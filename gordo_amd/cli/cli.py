@@ -17,10 +17,13 @@ import sys
 import click
 import yaml
 
+from .. import __version__
+
 logger = logging.getLogger(__name__)
 
 
 @click.group("gordo-amd")
+@click.version_option(version=__version__, prog_name="gordo-amd")
 @click.option("--log-level", default="INFO")
 def gordo(log_level):
     """gordo-amd: build, serve and monitor time-series anomaly models on
