@@ -153,6 +153,29 @@ def test_lstm_forecast_fused():
     assert np.isfinite(score)
 
 
+def test_lstm_fused_forward_fuzz():
+    """Seeded randomized shape fuzz of the fused LSTM forward vs fp32 torch:
+    features/hidden 2..128 (both width tiers), stacks 1..3, lookback 2..24,
+    awkward row counts — exercises the padding/packing envelope."""
+    rng = np.random.default_rng(77)
+    for trial in range(10):
+        F = int(rng.integers(2, 129))
+        L = int(rng.integers(1, 4))
+        dims = [int(rng.integers(2, 129)) for _ in range(L)]
+        T = int(rng.integers(2, 25))
+        n = int(rng.integers(T + 1, T + 120))
+        torch.manual_seed(1000 + trial)
+        net = LSTMNet(F, F, dims, [None] * L, out_func="linear", lookback_window=T)
+        X = rng.normal(size=(n, F)).astype(np.float32)
+        ref = ref_forward(net, X, T)
+        out = FusedLSTMForward(net, DEV).predict(X)
+        assert out.shape == ref.shape
+        np.testing.assert_allclose(
+            out, ref, atol=8e-2, rtol=8e-2,
+            err_msg=f"trial {trial}: F={F} dims={dims} T={T} n={n}",
+        )
+
+
 # ---------------------------------------------------------------------------
 # wide variant (W=128): feature/hidden widths 65..128
 # ---------------------------------------------------------------------------
