@@ -5,7 +5,13 @@
 #include "launchers.h"
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("hg_train_step", &hg_train_step_launch, "fused hourglass train step");
+    m.def("hg_train_step", &hg_train_step_launch, "fused hourglass train step",
+          py::arg("X"), py::arg("Y"), py::arg("counters"), py::arg("w"),
+          py::arg("wt"), py::arg("bias"), py::arg("dw"), py::arg("db"),
+          py::arg("loss"), py::arg("B"), py::arg("F_out"), py::arg("acts"),
+          py::arg("gscale"), py::arg("y_is_x"), py::arg("dx_out"),
+          py::arg("wmax"), py::arg("kbits"), py::arg("seed"),
+          py::arg("widths") = std::vector<int64_t>{});
     m.def("hg_adam", &hg_adam_launch, "slab-reduce + Adam + bf16 weight emit");
     m.def("hg_bump", &hg_bump_launch, "advance step counters");
     m.def("hg_permute_fill", &hg_permute_fill_launch, "materialize Feistel shuffle (tests)");

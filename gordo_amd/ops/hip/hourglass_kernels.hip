@@ -43,6 +43,11 @@ struct HgTrainArgs {
     int L;          // number of layers (<= GA_MAXL)
     int y_is_x;     // autoencoder fast path
     int act_id[GA_MAXL];  // activation per layer (last = out_func)
+    int width[GA_MAXL + 1];  // real layer widths: hidden activations beyond
+                             // width[li+1] are masked to 0 (an activation
+                             // with act(0) != 0, e.g. sigmoid, would
+                             // otherwise leak into the next layer's padded
+                             // dW rows and drift the padded master weights)
     float gscale;   // 2 / (B * F_out)
 };
 
@@ -149,10 +154,11 @@ __global__ void hg_train_step(HgTrainArgs a) {
             for (int nt = 0; nt < NTW; ++nt) {
                 int col = wcol + nt * 16 + (lane & 15);
                 float b = s_bias[li * W + col];
+                bool pad = col >= a.width[li + 1];
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     int m = wrow + (lane >> 4) * 4 + r;
-                    float h = ga_act(acc[nt][r] + b, a.act_id[li]);
+                    float h = pad ? 0.f : ga_act(acc[nt][r] + b, a.act_id[li]);
                     hg_write1<W>(act_tile(li + 1), m, col, ga_f32_to_bf16(h));
                 }
             }
@@ -707,7 +713,7 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
                           torch::Tensor loss, int64_t B, int64_t F_out,
                           std::vector<int64_t> acts, double gscale, bool y_is_x,
                           c10::optional<torch::Tensor> dx_out, int64_t wmax,
-                          int64_t kbits, int64_t seed) {
+                          int64_t kbits, int64_t seed, std::vector<int64_t> widths) {
     CHECK_GPU(X);
     TORCH_CHECK(counters.numel() >= 3, "counters must hold [batch, t, epoch]");
     int L = (int)(w.numel() / (wmax * wmax));
@@ -730,6 +736,15 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
     a.L = L;
     a.y_is_x = y_is_x ? 1 : 0;
     ga_fill_acts(a.act_id, acts, L);
+    if (widths.empty()) {
+        // no masking info: treat every layer as full-width (only safe when
+        // all hidden activations satisfy act(0) == 0)
+        for (int i = 0; i <= GA_MAXL; ++i) a.width[i] = (int)wmax;
+    } else {
+        TORCH_CHECK((int)widths.size() == L + 1, "widths must have L+1 entries");
+        for (int i = 0; i <= L; ++i) a.width[i] = (int)widths[i];
+        for (int i = L + 1; i <= GA_MAXL; ++i) a.width[i] = (int)wmax;
+    }
     a.gscale = (float)gscale;
 
     int n_tiles = (int)((B + HG_MT - 1) / HG_MT);

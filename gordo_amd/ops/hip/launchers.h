@@ -10,7 +10,8 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
                           torch::Tensor loss, int64_t B, int64_t F_out,
                           std::vector<int64_t> acts, double gscale, bool y_is_x,
                           c10::optional<torch::Tensor> dx_out, int64_t wmax,
-                          int64_t kbits, int64_t seed);
+                          int64_t kbits, int64_t seed,
+                          std::vector<int64_t> widths = {});
 
 void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                     torch::Tensor dw, torch::Tensor db, torch::Tensor w,

@@ -158,7 +158,7 @@ class FusedHourglassTrainer:
             self.E.hg_train_step(
                 Xd, Yd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
                 loss_acc, B, F_out, acts, gscale, bool(y_is_x), None, WM,
-                kbits, seed,
+                kbits, seed, [int(x) for x in p.widths],
             )
             # two-stage slab reduce (counter bump folded into stage A)
             self.E.hg_slab_reduce(dw, db, partial, counters, n_tiles, G, L,

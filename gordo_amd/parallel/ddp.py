@@ -197,6 +197,7 @@ class HourglassStepTrainer:
             self.Xd, self.Xd, self.counters, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
             self.gscale, True, None, WM, self.kbits, self.seed,
+            [int(x) for x in p.widths],
         )
         # two-stage slab reduce; stage A also bumps the counters
         self.E.hg_slab_reduce(

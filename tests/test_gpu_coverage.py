@@ -418,6 +418,69 @@ def test_fused_trainer_x256_learns():
     assert hist.loss[-1] < hist.loss[0] * 0.7, hist.loss
 
 
+def test_fused_training_grads_fuzz():
+    """Randomized-shape one-step grad parity vs fp32 autograd (all width
+    tiers, random layer stacks/activations, y != X half the time)."""
+    from gordo_amd.ops import ext
+    from gordo_amd.ops.hourglass_engine import _PackedParams, pad_rows_bf16
+
+    E = ext()
+    MT = int(E.MT)
+    rng = np.random.default_rng(321)
+    acts = ["tanh", "relu", "sigmoid", "linear"]
+    for trial in range(6):
+        L = int(rng.integers(1, 5))
+        widths = [int(rng.integers(3, 257)) for _ in range(L + 1)]
+        funcs = [str(rng.choice(acts))] * (L - 1)
+        torch.manual_seed(500 + trial)
+        net = FeedForwardNet(widths[0], widths[-1], widths[1:-1], funcs, "tanh")
+        ref_net = FeedForwardNet(widths[0], widths[-1], widths[1:-1], funcs, "tanh")
+        ref_net.load_state_dict(net.state_dict())
+        B = int(rng.integers(1, 5)) * MT
+        y_is_x = bool(rng.integers(0, 2)) and widths[0] == widths[-1]
+        X = rng.normal(size=(B, widths[0])).astype(np.float32)
+        Y = X if y_is_x else rng.normal(size=(B, widths[-1])).astype(np.float32)
+
+        p = _PackedParams(net, DEV)
+        WM, F_out = p.wmax, widths[-1]
+        n_tiles = (B + MT - 1) // MT
+        Xd = pad_rows_bf16(X, DEV, WM)
+        Yd = Xd if y_is_x else pad_rows_bf16(Y, DEV, WM)
+        dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
+        db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+        loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+        counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
+        gscale = 2.0 / (B * F_out)
+        E.hg_train_step(Xd, Yd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                        loss, B, F_out, p.act_ids, gscale, y_is_x, None, WM, 0, 0,
+                        [int(x) for x in widths])
+        torch.cuda.synchronize()
+
+        Xr = Xd[:, : widths[0]].float().cpu()
+        Yr = Xr if y_is_x else Yd[:, :F_out].float().cpu()
+        loss_ref = torch.nn.functional.mse_loss(ref_net(Xr), Yr)
+        loss_ref.backward()
+        msg = f"trial {trial}: widths={widths} funcs={funcs} B={B} y_is_x={y_is_x}"
+        kernel_loss = float(loss.item()) / (B * F_out)
+        assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 6e-2, msg
+
+        linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
+        dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
+        db_sum = db.sum(dim=0).view(L, WM).cpu()
+        for i, lin in enumerate(linears):
+            o, ninp = lin.weight.shape
+            got_w = dw_sum[i, :ninp, :o].T
+            ref_w = lin.weight.grad.detach()
+            s = ref_w.abs().max().item() or 1.0
+            assert (got_w - ref_w).abs().max().item() < 0.06 * s + 2e-4, f"{msg} layer {i} dW"
+            got_b = db_sum[i, :o]
+            ref_b = lin.bias.grad.detach()
+            s = ref_b.abs().max().item() or 1.0
+            assert (got_b - ref_b).abs().max().item() < 0.06 * s + 2e-4, f"{msg} layer {i} db"
+            assert dw_sum[i, ninp:, :].abs().max().item() == 0.0, msg
+            assert dw_sum[i, :, o:].abs().max().item() == 0.0, msg
+
+
 def test_fused_forward_fuzz():
     """Seeded randomized shape fuzz of the fused forward vs fp32 torch:
     widths 2..256, 1..5 layers, mixed activations, awkward batch sizes."""
