@@ -153,6 +153,72 @@ def test_lstm_forecast_fused():
     assert np.isfinite(score)
 
 
+def test_lstm_grads_fuzz():
+    """Randomized-shape one-step grad parity vs fp32 autograd (both width
+    tiers, random stacks/lookbacks); also asserts padded wcat rows stay
+    zero-gradient."""
+    from gordo_amd.ops.hourglass_engine import pad_rows_bf16
+    from gordo_amd.runtime.trainer import TrainConfig
+
+    rng = np.random.default_rng(88)
+    for trial in range(4):
+        F = int(rng.integers(3, 129))
+        L = int(rng.integers(1, 3))
+        dims = [int(rng.integers(4, 129)) for _ in range(L)]
+        T = int(rng.integers(3, 17))
+        B = 256
+        torch.manual_seed(2000 + trial)
+        net = LSTMNet(F, F, dims, [None] * L, out_func="linear", lookback_window=T)
+        ref_net = LSTMNet(F, F, dims, [None] * L, out_func="linear", lookback_window=T)
+        ref_net.load_state_dict(net.state_dict())
+        X = rng.normal(size=(B + T + 2, F)).astype(np.float32)
+        msg = f"trial {trial}: F={F} dims={dims} T={T}"
+
+        cfg = TrainConfig(epochs=1, batch_size=B)
+        tr = FusedLSTMTrainer(net, DEV, cfg)
+        p = _LSTMPackedParams(net, DEV)
+        tr.p = p
+        Xd = pad_rows_bf16(X, DEV, p.W)
+        st = tr._alloc_step_state(Xd, Xd, B, T, 1, B, 0, F)
+        win = torch.arange(B, dtype=torch.int32, device=DEV)
+        tr._compute_grads(st, win)
+        torch.cuda.synchronize()
+
+        Xb = torch.as_tensor(X).to(torch.bfloat16).float()
+        wins = torch.stack([Xb[i : i + T] for i in range(B)])
+        tgt = torch.stack([Xb[i + T - 1] for i in range(B)])
+        loss = torch.nn.functional.mse_loss(ref_net(wins), tgt)
+        loss.backward()
+
+        kern_loss = float(st["loss_acc"].item()) / (B * F)
+        assert abs(kern_loss - float(loss.detach())) / max(float(loss), 1e-9) < 6e-2, msg
+
+        W = p.W
+        for li, cell in enumerate(ref_net.cells):
+            H, I = cell.hidden_size, cell.input_size
+            gw = p.wcat_master[li].grad.cpu()
+            gb = p.bias_master[li].grad.cpu()
+            for g in range(4):
+                ref_ih = cell.weight_ih_l0.grad[g * H : (g + 1) * H]
+                ref_hh = cell.weight_hh_l0.grad[g * H : (g + 1) * H]
+                ref_b = cell.bias_ih_l0.grad[g * H : (g + 1) * H]
+                got_ih = gw[g * W : g * W + H, :I]
+                got_hh = gw[g * W : g * W + H, W : W + H]
+                got_b = gb[g * W : g * W + H]
+                s = ref_ih.abs().max().item() or 1.0
+                assert (got_ih - ref_ih).abs().max().item() < 0.09 * s + 3e-4, f"{msg} l{li} g{g} dW_ih"
+                s = ref_hh.abs().max().item() or 1.0
+                assert (got_hh - ref_hh).abs().max().item() < 0.09 * s + 3e-4, f"{msg} l{li} g{g} dW_hh"
+                s = ref_b.abs().max().item() or 1.0
+                assert (got_b - ref_b).abs().max().item() < 0.09 * s + 3e-4, f"{msg} l{li} g{g} db"
+                assert gw[g * W + H : (g + 1) * W].abs().max().item() == 0.0, msg
+                assert gw[g * W : g * W + H, I:W].abs().max().item() == 0.0, msg
+        ref_hw = ref_net.head.weight.grad
+        got_hw = p.head_wm.grad.cpu()[: ref_hw.shape[0], : ref_hw.shape[1]]
+        s = ref_hw.abs().max().item() or 1.0
+        assert (got_hw - ref_hw).abs().max().item() < 0.09 * s + 3e-4, f"{msg} head dW"
+
+
 def test_lstm_fused_forward_fuzz():
     """Seeded randomized shape fuzz of the fused LSTM forward vs fp32 torch:
     features/hidden 2..128 (both width tiers), stacks 1..3, lookback 2..24,
