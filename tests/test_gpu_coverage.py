@@ -418,6 +418,41 @@ def test_fused_trainer_x256_learns():
     assert hist.loss[-1] < hist.loss[0] * 0.7, hist.loss
 
 
+def test_fused_anomaly_epilogue_fuzz():
+    """hg_forward's anomaly epilogue (scaled |err| + row-L2 total) vs numpy,
+    random widths incl. the scale/shift fusion."""
+    from gordo_amd.ops.hourglass_engine import FusedHourglassForward
+
+    rng = np.random.default_rng(55)
+    for trial in range(6):
+        F = int(rng.integers(2, 257))
+        L = int(rng.integers(1, 4))
+        dims = [int(rng.integers(2, 257)) for _ in range(L - 1)]
+        B = int(rng.integers(1, 150))
+        torch.manual_seed(3000 + trial)
+        net = FeedForwardNet(F, F, dims, ["tanh"] * (L - 1), "linear")
+        fwd = FusedHourglassForward(net, DEV)
+        scale = rng.uniform(0.5, 2.0, F).astype(np.float32)
+        shift = rng.uniform(-0.5, 0.5, F).astype(np.float32)
+        inv = rng.uniform(0.1, 3.0, F).astype(np.float32)
+        fwd.set_input_scaler(scale, shift)
+        X = rng.normal(size=(B, F)).astype(np.float32)
+        yhat, err, total = fwd.anomaly(torch.as_tensor(X), inv)
+        yhat, err, total = yhat.cpu().numpy(), err.cpu().numpy(), total.cpu().numpy()
+
+        xs = X * scale + shift
+        ref_y = net(torch.as_tensor(xs)).detach().numpy()
+        msg = f"trial {trial}: F={F} dims={dims} B={B}"
+        np.testing.assert_allclose(yhat, ref_y, atol=7e-2, rtol=7e-2, err_msg=msg)
+        # epilogue math vs the kernel's own yhat (isolates the epilogue)
+        xs_b = torch.as_tensor(xs).to(torch.bfloat16).float().numpy()
+        ref_err = np.abs(yhat - xs_b) * inv
+        np.testing.assert_allclose(err, ref_err, atol=2e-2, rtol=2e-2, err_msg=msg)
+        np.testing.assert_allclose(
+            total, np.sqrt((ref_err**2).sum(axis=1)), atol=5e-2, rtol=2e-2, err_msg=msg
+        )
+
+
 def test_fused_training_grads_fuzz():
     """Randomized-shape one-step grad parity vs fp32 autograd (all width
     tiers, random layer stacks/activations, y != X half the time)."""
