@@ -158,3 +158,22 @@ def test_dataset_with_file_provider(tmp_path):
     X, _ = ds.get_data()
     assert list(X.columns) == TAGS
     assert len(X) > 50
+
+
+def test_multiple_aggregation_methods():
+    """aggregation_methods as a list yields (tag, method) MultiIndex columns
+    (reference TimeSeriesDataset behavior)."""
+    from gordo_amd.dataset.datasets import TimeSeriesDataset
+
+    ds = TimeSeriesDataset(
+        train_start_date="2020-01-01T00:00:00Z",
+        train_end_date="2020-01-03T00:00:00Z",
+        tag_list=["tag-0", "tag-1"],
+        data_provider={"type": "RandomDataProvider"},
+        aggregation_methods=["mean", "max"],
+    )
+    X, _ = ds.get_data()
+    assert isinstance(X.columns, pd.MultiIndex)
+    assert ("tag-0", "mean") in X.columns and ("tag-1", "max") in X.columns
+    assert X.shape[1] == 4
+    assert (X["tag-0"]["max"] >= X["tag-0"]["mean"] - 1e-9).all()
