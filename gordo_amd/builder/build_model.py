@@ -33,7 +33,7 @@ from ..machine import Machine
 from ..model.anomaly.base import AnomalyDetectorBase
 from ..model.base import GordoBase
 from ..model.utils import metric_wrapper
-from ..serializer import dump, from_definition, into_definition
+from ..serializer import dump, from_definition
 from ..util import disk_registry
 
 logger = logging.getLogger(__name__)

@@ -19,7 +19,7 @@ from .base import GordoBaseDataset, InsufficientDataError
 from .data_provider.base import GordoBaseDataProvider
 from .data_provider.providers import RandomDataProvider
 from .filter_rows import pandas_filter_rows
-from ..machine.sensor_tag import SensorTag, normalize_sensor_tags, to_list_of_strings
+from ..machine.sensor_tag import normalize_sensor_tags, to_list_of_strings
 
 logger = logging.getLogger(__name__)
 
