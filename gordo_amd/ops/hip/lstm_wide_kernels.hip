@@ -69,7 +69,12 @@ __global__ void lstm_fwd_wide(WFwdArgs a) {
     for (int e = 0; e < 32; ++e) c_reg[e] = 0.f;
     __syncthreads();
 
+    const unsigned short* wcat_p = a.wcat;
     for (int t = 0; t < a.T; ++t) {
+        // the weight addresses are loop-invariant; without this barrier the
+        // compiler hoists ~100 B-fragments into registers across the t-loop
+        // and spills (measured 20-96 spills/lane, occupancy 1 wave/SIMD)
+        asm volatile("" : "+s"(wcat_p));
         // ---- stage x_t (window gather on layer 0) -------------------------
         for (int c = tid; c < 64 * 16; c += 256) {
             int m = c >> 4, col0 = (c & 15) * 8;
@@ -104,7 +109,7 @@ __global__ void lstm_fwd_wide(WFwdArgs a) {
                 ga_bf16x8 afrag = *(const ga_bf16x8*)&src[sw128(mt * 16 + (lane & 15), klocal + (lane >> 4) * 8)];
 #pragma unroll
                 for (int jt = 0; jt < 8; ++jt) {
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat +
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(wcat_p +
                         (size_t)(wid * WW + jt * 16 + (lane & 15)) * WK + kk * 32 + (lane >> 4) * 8);
                     acc[mt][jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[mt][jt], 0, 0, 0);
                 }
@@ -203,7 +208,9 @@ __global__ void lstm_bwd_wide(WBwdArgs a) {
     for (int e = 0; e < 32; ++e) dc_reg[e] = 0.f;
     __syncthreads();
 
+    const unsigned short* wct_p = a.wcat_t;
     for (int t = a.T - 1; t >= 0; --t) {
+        asm volatile("" : "+s"(wct_p));  // see lstm_fwd_wide: stop weight-LICM
         // ---- elementwise BPTT (16-B vector IO) ----------------------------
         {
             const int gi = base + em;
@@ -287,7 +294,7 @@ __global__ void lstm_bwd_wide(WBwdArgs a) {
                 ga_bf16x8 bfrag[8];
 #pragma unroll
                 for (int nt = 0; nt < 8; ++nt)
-                    bfrag[nt] = *(const ga_bf16x8*)(a.wcat_t +
+                    bfrag[nt] = *(const ga_bf16x8*)(wct_p +
                         (size_t)(brow0 + nt * 16 + (lane & 15)) * W4H + kk + (lane >> 4) * 8);
 #pragma unroll
                 for (int st2 = 0; st2 < 2; ++st2) {
@@ -329,7 +336,7 @@ __global__ void lstm_bwd_wide(WBwdArgs a) {
                 ga_bf16x8 afrag = *(const ga_bf16x8*)&dgt[sw128(wrow + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
 #pragma unroll
                 for (int nt = 0; nt < 8; ++nt) {
-                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(a.wcat_t +
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(wct_p +
                         (size_t)(brow0 + nt * 16 + (lane & 15)) * W4H + kk + (lane >> 4) * 8);
                     acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
                 }
