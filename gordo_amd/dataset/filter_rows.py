@@ -12,8 +12,8 @@ from typing import List, Union
 import pandas as pd
 
 _ALLOWED = re.compile(
-    r"""^[\w\s\.\,\'\"\`\<\>\=\!\&\|\~\+\-\*\/\(\)\[\]\@]*$"""
-)
+    r"""^[\w\s\.\,\'\"\`\<\>\=\!\&\|\~\+\-\*\/\(\)\[\]\@\:]*$"""
+)  # ':' for timestamps in period filters, e.g. "index <= '2020-01-02 06:00:00'"
 
 
 def parse_pandas_filter_vars(expr: Union[str, List[str]]) -> List[str]:
