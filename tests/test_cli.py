@@ -118,3 +118,26 @@ def test_fleet_build_command(tmp_path):
     assert result.exit_code == 0, result.output
     rep = json.loads(report.read_text())
     assert rep[0]["status"] == "succeeded"
+
+
+def test_version_and_help_surfaces():
+    """Every reference CLI surface parses: --version and each command group's
+    --help exit 0 (run-server/run-watchman/client start servers or need
+    endpoints, so only their arg parsing is smoked here)."""
+    runner = CliRunner()
+    r = runner.invoke(gordo, ["--version"])
+    assert r.exit_code == 0 and "gordo-amd" in r.output
+    for args in (
+        ["build", "--help"],
+        ["run-server", "--help"],
+        ["run-watchman", "--help"],
+        ["client", "--help"],
+        ["client", "predict", "--help"],
+        ["client", "metadata", "--help"],
+        ["client", "download-model", "--help"],
+        ["workflow", "generate", "--help"],
+        ["fleet", "build", "--help"],
+        ["fleet", "serve", "--help"],
+    ):
+        r = runner.invoke(gordo, args)
+        assert r.exit_code == 0, (args, r.output)
