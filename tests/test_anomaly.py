@@ -74,3 +74,42 @@ def test_require_thresholds_enforced():
     with pytest.raises(AttributeError, match="thresholds"):
         det.anomaly(pd.DataFrame(X, columns=[f"t{i}" for i in range(8)]),
                     pd.DataFrame(X, columns=[f"t{i}" for i in range(8)]))
+
+
+def test_metric_wrapper_scaling_and_per_tag():
+    """metric_wrapper transforms through the scaler before scoring and the
+    col argument restricts scoring to one tag."""
+    from sklearn.metrics import mean_squared_error
+    from sklearn.preprocessing import StandardScaler
+
+    from gordo_amd.model.utils import metric_wrapper
+
+    rng = np.random.default_rng(1)
+    y_true = rng.normal(0, [1.0, 100.0], size=(200, 2))
+    y_pred = y_true + rng.normal(0, [0.1, 10.0], size=(200, 2))
+    scaler = StandardScaler().fit(y_true)
+
+    plain = mean_squared_error(y_true, y_pred)
+    scaled = metric_wrapper(mean_squared_error, scaler)(y_true, y_pred)
+    expected = mean_squared_error(scaler.transform(y_true), scaler.transform(y_pred))
+    assert abs(scaled - expected) < 1e-9
+    assert abs(scaled - plain) > 1.0  # the big-magnitude tag dominated plain MSE
+
+    per_tag0 = metric_wrapper(mean_squared_error, scaler, col=0)(y_true, y_pred)
+    per_tag1 = metric_wrapper(mean_squared_error, scaler, col=1)(y_true, y_pred)
+    s_true, s_pred = scaler.transform(y_true), scaler.transform(y_pred)
+    assert abs(per_tag0 - mean_squared_error(s_true[:, 0], s_pred[:, 0])) < 1e-9
+    assert abs(per_tag1 - mean_squared_error(s_true[:, 1], s_pred[:, 1])) < 1e-9
+
+
+def test_metric_wrapper_lookback_alignment():
+    """Lookback models emit fewer rows than y: the wrapper aligns to window
+    ends before scoring."""
+    from sklearn.metrics import mean_absolute_error
+
+    from gordo_amd.model.utils import metric_wrapper
+
+    y_true = np.arange(20, dtype=float).reshape(10, 2)
+    y_pred = y_true[3:] + 1.0  # 7 rows, as if lookback=4
+    got = metric_wrapper(mean_absolute_error)(y_true, y_pred)
+    assert abs(got - 1.0) < 1e-12
