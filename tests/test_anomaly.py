@@ -113,3 +113,34 @@ def test_metric_wrapper_lookback_alignment():
     y_pred = y_true[3:] + 1.0  # 7 rows, as if lookback=4
     got = metric_wrapper(mean_absolute_error)(y_true, y_pred)
     assert abs(got - 1.0) < 1e-12
+
+
+def test_transformer_funcs_in_yaml_pipeline():
+    """InfImputer / multiply are YAML-referenceable and behave (parity:
+    gordo_components/model/transformer_funcs)."""
+    from gordo_amd.model.transformer_funcs.general import InfImputer, multiply
+    from gordo_amd.serializer import from_definition
+
+    X = np.array([[1.0, np.inf], [-np.inf, 2.0], [3.0, 4.0]])
+    imp = InfImputer().fit(X)
+    out = imp.transform(X)
+    assert np.isfinite(out).all()
+    assert out[0, 1] == 4.0 + 2.0 and out[1, 0] == 1.0 - 2.0
+
+    imp2 = InfImputer(inf_fill_value=99.0, neg_inf_fill_value=-99.0).fit(X)
+    out2 = imp2.transform(X)
+    assert out2[0, 1] == 99.0 and out2[1, 0] == -99.0
+
+    assert (multiply(np.ones((2, 2)), factor=3.0) == 3.0).all()
+
+    # resolvable from a pipeline definition, like the reference YAMLs
+    pipe = from_definition({
+        "sklearn.pipeline.Pipeline": {
+            "steps": [
+                "gordo_amd.model.transformer_funcs.general.InfImputer",
+                "sklearn.preprocessing.MinMaxScaler",
+            ]
+        }
+    })
+    got = pipe.fit_transform(X)
+    assert np.isfinite(got).all() and got.min() >= 0 and got.max() <= 1
