@@ -144,3 +144,19 @@ def test_transformer_funcs_in_yaml_pipeline():
     })
     got = pipe.fit_transform(X)
     assert np.isfinite(got).all() and got.min() >= 0 and got.max() <= 1
+
+
+def test_rolling_windows_lookahead_semantics():
+    """rolling_windows: windows[i] = X[i:i+lookback], target[i] =
+    X[i+lookback-1+lookahead] (forecast = 1-step-ahead)."""
+    from gordo_amd.model.utils import rolling_windows
+
+    X = np.arange(10, dtype=np.float32)[:, None]
+    w, t = rolling_windows(X, lookback=3, lookahead=0)
+    assert w.shape == (8, 3, 1) and t.shape == (8, 1)
+    assert (w[0].ravel() == [0, 1, 2]).all() and t[0, 0] == 2  # AE: last row
+    w1, t1 = rolling_windows(X, lookback=3, lookahead=1)
+    assert w1.shape == (7, 3, 1)
+    assert (w1[0].ravel() == [0, 1, 2]).all() and t1[0, 0] == 3  # forecast
+    with pytest.raises(Exception):
+        rolling_windows(X[:2], lookback=3, lookahead=0)
