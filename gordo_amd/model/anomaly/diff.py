@@ -146,14 +146,24 @@ class DiffBasedAnomalyDetector(BaseEstimator, AnomalyDetectorBase):
             agg = np.linalg.norm(err, axis=1)
             per_fold_aggregate.append(np.percentile(agg, 92))
 
-        feature_names = (
-            list(X.columns) if isinstance(X, pd.DataFrame) else list(range(Xv.shape[1]))
-        )
+        # Per-feature errors are computed over the TARGET (y) columns, so the
+        # thresholds are keyed by target tag names — not X's — otherwise
+        # metadata/confidence columns mislabel when target_tag_list != tag_list.
+        if isinstance(y, pd.DataFrame):
+            feature_names = list(y.columns)
+        elif isinstance(X, pd.DataFrame) and Xv.shape[1] == yv.shape[1]:
+            feature_names = list(X.columns)  # autoencoder case: y == X
+        else:
+            feature_names = list(range(yv.shape[1]))
         # conservative: the max over folds (the reference takes the final
         # fold's percentiles; max is a stricter, order-independent choice)
         ft = np.max(np.stack(per_fold_feature_thresholds), axis=0)
-        n_out = min(len(feature_names), len(ft))
-        self.feature_thresholds_ = pd.Series(ft[:n_out], index=feature_names[:n_out])
+        if len(ft) != len(feature_names):
+            raise ValueError(
+                f"threshold count {len(ft)} != target column count "
+                f"{len(feature_names)} — prediction width must match y"
+            )
+        self.feature_thresholds_ = pd.Series(ft, index=feature_names)
         self.aggregate_threshold_ = float(np.max(per_fold_aggregate))
         return cv_output
 
@@ -217,10 +227,15 @@ class DiffBasedAnomalyDetector(BaseEstimator, AnomalyDetectorBase):
             cols.append(("smooth-total-anomaly-score", ""))
 
         if self.feature_thresholds_ is not None:
-            n = min(len(target_tags), len(self.feature_thresholds_))
-            ft = self.feature_thresholds_.values[:n]
-            parts.append(err[:, :n] / np.where(ft == 0, np.inf, ft))
-            cols.extend(("anomaly-confidence", t) for t in target_tags[:n])
+            ft = self.feature_thresholds_.values
+            if len(ft) != err.shape[1]:
+                raise ValueError(
+                    f"feature_thresholds_ has {len(ft)} entries but the "
+                    f"prediction has {err.shape[1]} target columns — model "
+                    "and thresholds disagree"
+                )
+            parts.append(err / np.where(ft == 0, np.inf, ft))
+            cols.extend(("anomaly-confidence", t) for t in target_tags)
         if self.aggregate_threshold_ is not None:
             parts.append((total / self.aggregate_threshold_)[:, None])
             cols.append(("total-anomaly-confidence", ""))

@@ -8,7 +8,7 @@ tanh hidden activations, linear output, Adam + MSE (consumed by the trainer).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Sequence
 
 from ..register import register_model_builder
 from ..torch_nets import FeedForwardNet
@@ -74,14 +74,17 @@ def feedforward_hourglass(
     """Hourglass: dims interpolate n_features → n_features*compression_factor
     over ``encoding_layers`` layers, then mirror back up.
 
-    e.g. n_features=50, encoding_layers=2, compression=0.5 → 50→38→25→38→50.
+    Matches the reference composition hourglass == symmetric(hourglass_calc_dims)
+    — the bottleneck layer is duplicated, e.g. n_features=50, encoding_layers=2,
+    compression=0.5 → hidden stack 38→25→25→38 (i.e. 50→38→25→25→38→50 with the
+    input/output widths).
     """
     dims = hourglass_calc_dims(compression_factor, encoding_layers, n_features)
-    decoding_dims = dims[::-1][1:] + [n_features] if len(dims) > 1 else [n_features]
-    # final decoder width replaced by output layer (n_features_out); the last
-    # entry of decoding_dims is the widest hidden layer before the output.
-    all_dims = dims + dims[::-1][1:]
-    funcs: List[Optional[str]] = [func] * len(all_dims)
-    return FeedForwardNet(
-        n_features, n_features_out or n_features, all_dims, funcs, out_func
+    return feedforward_symmetric(
+        n_features,
+        n_features_out,
+        dims=dims,
+        funcs=[func] * len(dims),
+        out_func=out_func,
+        **kwargs,
     )

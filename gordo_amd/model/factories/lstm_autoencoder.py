@@ -7,7 +7,7 @@ dim logic shared with the feedforward factories via factories/utils.py.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Sequence
 
 from ..register import register_model_builder
 from ..torch_nets import LSTMNet
@@ -84,16 +84,16 @@ def lstm_hourglass(
     out_func: str = "linear",
     **kwargs,
 ) -> LSTMNet:
+    # Reference composition: hourglass == symmetric(hourglass_calc_dims) —
+    # the bottleneck layer is duplicated (e.g. 50 tags, 2 layers, 0.5 →
+    # LSTM stack 38→25→25→38), consistent with feedforward_hourglass.
     dims = hourglass_calc_dims(compression_factor, encoding_layers, n_features)
-    all_dims = dims + dims[::-1][1:] + ([n_features] if len(dims) > 1 else [])
-    if len(dims) == 1:
-        all_dims = dims
-    funcs: List[Optional[str]] = [func] * len(all_dims)
-    return LSTMNet(
+    return lstm_symmetric(
         n_features,
-        n_features_out or n_features,
-        all_dims,
-        funcs,
-        out_func=out_func,
+        n_features_out,
         lookback_window=lookback_window,
+        dims=dims,
+        funcs=[func] * len(dims),
+        out_func=out_func,
+        **kwargs,
     )

@@ -197,7 +197,12 @@ class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
         model = self.model.to(device)
         model.eval()
         Xt = torch.as_tensor(X, device=device)
-        n_windows = len(X) - lookback + 1
+        # Drop trailing windows whose target row (window end + lookahead) is
+        # beyond X: output row i always pairs with y[i + lookback - 1 +
+        # lookahead] under the trailing-offset alignment used by diff.py /
+        # metric_wrapper / make_base_dataframe. Matches the reference, whose
+        # forecast predict emits len(X) - lookback rows.
+        n_windows = len(X) - lookback + 1 - self.lookahead
         offsets = torch.arange(lookback, device=device)
         outs = []
         with torch.no_grad():
@@ -228,20 +233,19 @@ class TorchLSTMAutoEncoder(TorchLSTMBaseEstimator):
 
 class TorchLSTMForecast(TorchLSTMBaseEstimator):
     """Forecasts X[t+1] from the window ending at t
-    (parity: KerasLSTMForecast, lookahead=1)."""
+    (parity: KerasLSTMForecast, lookahead=1). predict() emits
+    len(X) - lookback_window rows — prediction i targets
+    X[i + lookback - 1 + lookahead], so trailing-offset alignment in
+    diff.py / metric_wrapper pairs every prediction with its true target."""
 
     lookahead = 1
 
     def score(self, X, y=None, sample_weight=None) -> float:
         X = self._as_array(X)
         y = X if y is None else self._as_array(y)
-        out = self.predict(X)  # predictions for rows lookback-1+1 .. end
-        # prediction i targets row i + lookback - 1 + lookahead
+        out = self.predict(X)
+        # prediction i targets row i + lookback - 1 + lookahead; predict()
+        # already drops the windows whose target is beyond X, so lengths match
         targets = y[self.lookback_window - 1 + self.lookahead :]
-        out = out[: len(targets)]
+        assert len(targets) == len(out), (len(targets), len(out))
         return explained_variance_score(targets, out)
-
-    def _window_predict(self, X: np.ndarray, batch_size: int = 8192) -> np.ndarray:
-        # For forecasting, the final window (ending at the last row) predicts
-        # a row beyond X; keep it — callers align by offset.
-        return super()._window_predict(X, batch_size)

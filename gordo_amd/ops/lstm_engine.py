@@ -452,15 +452,22 @@ class FusedLSTMForward:
         self.p = _LSTMPackedParams(module, device)
 
     def predict(self, X: np.ndarray, lookahead: int = 0, batch_size: int = 16384) -> np.ndarray:
-        """X [N, F] -> predictions for every window, [N - T + 1, F_out]."""
+        """X [N, F] -> predictions [N - T + 1 - lookahead, F_out].
+
+        Windows whose target row (window end + lookahead) falls beyond X are
+        dropped so downstream trailing-offset alignment (diff.py,
+        metric_wrapper, make_base_dataframe) pairs prediction i with
+        y[i + T - 1 + lookahead] — matching rolling_windows' training targets
+        and the reference's forecast predict row count.
+        """
         p = self.p
         T = self.module.lookback_window
         F_out = self.module.n_features_out
         Xd = pad_rows_bf16(np.asarray(X, np.float32), self.device, p.W)
         N = Xd.shape[0]
-        n_windows = N - T + 1
+        n_windows = N - T + 1 - lookahead
         if n_windows <= 0:
-            raise ValueError(f"need >= {T} rows for lookback={T}")
+            raise ValueError(f"need >= {T + lookahead} rows for lookback={T}")
         outs = []
         bf16 = torch.bfloat16
         B = min(batch_size, n_windows)
