@@ -51,7 +51,18 @@ def main():
         "(LSTM-AE, lookback 144, fused 4-gate cell)",
     )
     ap.add_argument("--lookback", type=int, default=144)
+    ap.add_argument(
+        "--dry-run", action="store_true",
+        help="validate the N-rank launch wiring (env, rendezvous, barrier, "
+        "max-over-ranks reduction) with a tiny eager run — works on CPU/gloo, "
+        "so the 8-GPU launch command can be rehearsed without hardware",
+    )
     args = ap.parse_args()
+    if args.dry_run:
+        args.steps, args.warmup = 2, 1
+        args.batch = min(args.batch or 256, 256)
+        if args.model == "lstm":
+            args.model = "hourglass"  # lstm needs the GPU; wiring is identical
 
     from gordo_amd.model.factories import feedforward_hourglass
     from gordo_amd.parallel.ddp import DPContext, HourglassStepTrainer
@@ -89,7 +100,8 @@ def main():
         module = feedforward_hourglass(
             args.tags, args.tags, encoding_layers=2, compression_factor=0.5
         )
-        X = synthetic_sensor_data(DATASET_ROWS, args.tags, seed=1000 + rank)
+        rows = 4096 if args.dry_run else DATASET_ROWS
+        X = synthetic_sensor_data(rows, args.tags, seed=1000 + rank)
         cfg = TrainConfig(batch_size=batch, lr=1e-3, dtype="bf16")
         trainer = HourglassStepTrainer(module, X, batch, cfg, dp=dp)
         model_name = "feedforward_hourglass-" + "x".join(str(w) for w in module.layer_widths)
@@ -144,6 +156,8 @@ def main():
                 "engine": engine,
             },
         }
+        if args.dry_run:
+            out["dry_run"] = True
         print(json.dumps(out))
 
 

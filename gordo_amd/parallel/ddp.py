@@ -235,8 +235,13 @@ class HourglassStepTrainer:
         self._gen = torch.Generator(device="cpu")
         self._gen.manual_seed(1234 + self.dp.rank)
 
+    def _next_indices(self) -> torch.Tensor:
+        """Minibatch row indices for the eager path; overridable so DP
+        semantics tests can pin deterministic batches."""
+        return torch.randint(0, len(self.Xt), (self.B,), generator=self._gen)
+
     def _eager_step(self):
-        idx = torch.randint(0, len(self.Xt), (self.B,), generator=self._gen).to(self.device)
+        idx = self._next_indices().to(self.device)
         xb = self.Xt[idx]
         self.opt.zero_grad(set_to_none=True)
         use_bf16 = self.device.type == "cuda" and self.cfg.dtype == "bf16"
