@@ -66,17 +66,46 @@ def build_app(model_dir: str, device=None):
                 yp = payload["y"]
                 y = dataframe_from_dict(yp if isinstance(yp, dict) else {"data": yp})
         # column sanity: when metadata lists tags and X has plain columns,
-        # reorder to the training tag order
+        # reorder to the training tag order. A width match alone is NOT
+        # enough for a frame with real column names (a misordered parquet
+        # frame would silently score garbage — the reference 400s on tag
+        # mismatch): named-but-unknown columns require the explicit
+        # X-Gordo-Assume-Tag-Order opt-in. Unnamed (RangeIndex) payloads —
+        # plain JSON value arrays — are positional by construction and are
+        # assigned the training tag order as the reference does.
         tags = runtime.tag_names
         if tags and not isinstance(X.columns, pd.MultiIndex):
             missing = [t for t in tags if t not in X.columns]
-            if not missing and list(X.columns) != tags:
-                X = X[tags]
-            elif missing and X.shape[1] == len(tags):
-                X.columns = tags
-            elif missing:
+            positional = isinstance(X.columns, pd.RangeIndex) or all(
+                isinstance(c, (int, np.integer)) for c in X.columns
+            )
+            assume_order = request.headers.get(
+                "x-gordo-assume-tag-order", ""
+            ).lower() in ("1", "true", "yes")
+            if not missing:
+                if list(X.columns) != tags:
+                    X = X[tags]
+            elif X.shape[1] != len(tags):
                 return None, None, JSONResponse(
-                    {"error": f"Missing expected tag columns: {missing[:5]}"},
+                    {
+                        "error": f"Expected {len(tags)} tag columns, got "
+                        f"{X.shape[1]}; missing: {missing[:5]}"
+                    },
+                    status_code=400,
+                )
+            elif positional or assume_order:
+                X = X.copy(deep=False)
+                X.columns = tags
+            else:
+                return None, None, JSONResponse(
+                    {
+                        "error": "Column names do not match the training tag "
+                        f"list (missing: {missing[:5]}). Send columns named "
+                        "by tag, or positional arrays, or set the "
+                        "X-Gordo-Assume-Tag-Order: true header to assert the "
+                        "columns are already in training order.",
+                        "expected-tags": tags,
+                    },
                     status_code=400,
                 )
         return X, y, None
