@@ -68,9 +68,24 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
         return self.kwargs
 
     # -- persistence -------------------------------------------------------
+    # When _externalize_weights is set (serializer.dump's externalize()
+    # context), pickles carry only a marker + dims; the tensors live in the
+    # artifact dir's weights.safetensors and serializer.load() mmaps them
+    # back in — the zero-pickle fast load path. Plain pickle/dumps stays
+    # self-contained (weights embedded as bytes, the reference's Keras-HDF5-
+    # inside-pickle idiom).
+    _externalize_weights: bool = False
+
     def __getstate__(self):
         state = self.__dict__.copy()
         if self.model is not None:
+            if TorchBaseEstimator._externalize_weights:
+                state["model"] = {
+                    "__external_weights__": True,
+                    "n_features": getattr(self, "_n_features", None),
+                    "n_features_out": getattr(self, "_n_features_out", None),
+                }
+                return state
             buf = io.BytesIO()
             torch.save(
                 {
@@ -88,13 +103,26 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
     def __setstate__(self, state):
         model_bytes = state.get("model")
         self.__dict__ = state
-        if isinstance(model_bytes, (bytes, bytearray)):
+        if isinstance(model_bytes, dict) and model_bytes.get("__external_weights__"):
+            self._n_features = model_bytes["n_features"]
+            self._n_features_out = model_bytes["n_features_out"]
+            self.model = None
+            self._needs_external_weights = True
+        elif isinstance(model_bytes, (bytes, bytearray)):
             payload = torch.load(io.BytesIO(model_bytes), map_location="cpu", weights_only=True)
             self._n_features = payload["n_features"]
             self._n_features_out = payload["n_features_out"]
             self.model = self._build(self._n_features, self._n_features_out)
             self.model.load_state_dict(payload["state_dict"])
             self.model.eval()
+
+    def restore_weights(self, state_dict: Dict[str, torch.Tensor]) -> None:
+        """Rebuild the module and load externally-stored tensors (the
+        weights.safetensors half of an externalized artifact)."""
+        self.model = self._build(self._n_features, self._n_features_out)
+        self.model.load_state_dict(state_dict)
+        self.model.eval()
+        self.__dict__.pop("_needs_external_weights", None)
 
     # -- build / fit / predict ----------------------------------------------
     def _factory_kwargs(self) -> Dict[str, Any]:
@@ -140,6 +168,12 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
 
     def _predict_array(self, X: np.ndarray, batch_size: int = 65536) -> np.ndarray:
         if self.model is None:
+            if getattr(self, "_needs_external_weights", False):
+                raise ValueError(
+                    "Estimator was unpickled from an externalized artifact; "
+                    "load it via serializer.load(dir) so weights.safetensors "
+                    "is restored"
+                )
             raise ValueError("Model has not been fit yet")
         device = resolve_device(self.kwargs.get("device", "auto"))
         model = self.model.to(device)

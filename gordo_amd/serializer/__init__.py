@@ -24,6 +24,7 @@ import json
 import logging
 import os
 import pickle
+from contextlib import contextmanager
 from typing import Any, Dict, Optional
 
 import yaml
@@ -46,36 +47,69 @@ __all__ = [
 ]
 
 
-def _collect_torch_tensors(obj: Any, prefix: str = "") -> Dict[str, "object"]:
-    """Walk the object graph for TorchBaseEstimator members; return
-    name→tensor for safetensors export."""
+def _walk_estimators(obj: Any, prefix: str = ""):
+    """Yield (prefix, TorchBaseEstimator) for every torch estimator in the
+    object graph — the one traversal shared by dump (tensor export) and load
+    (tensor restore), so safetensors key names always agree."""
+    from sklearn.pipeline import FeatureUnion, Pipeline
+
     from ..model.models import TorchBaseEstimator
 
+    if isinstance(obj, TorchBaseEstimator):
+        yield prefix, obj
+    elif isinstance(obj, Pipeline):
+        for name, step in obj.steps:
+            yield from _walk_estimators(step, f"{prefix}{name}.")
+    elif isinstance(obj, FeatureUnion):
+        for name, t in obj.transformer_list:
+            yield from _walk_estimators(t, f"{prefix}{name}.")
+    elif hasattr(obj, "base_estimator"):
+        yield from _walk_estimators(obj.base_estimator, f"{prefix}base_estimator.")
+
+
+def _collect_torch_tensors(obj: Any, prefix: str = "") -> Dict[str, "object"]:
+    """name→tensor for safetensors export."""
     tensors: Dict[str, Any] = {}
-
-    def walk(o: Any, pfx: str):
-        from sklearn.pipeline import FeatureUnion, Pipeline
-
-        if isinstance(o, TorchBaseEstimator) and o.model is not None:
-            for k, v in o.model.state_dict().items():
+    for pfx, est in _walk_estimators(obj, prefix):
+        if est.model is not None:
+            for k, v in est.model.state_dict().items():
                 tensors[f"{pfx}model.{k}"] = v.detach().cpu().contiguous()
-        elif isinstance(o, Pipeline):
-            for name, step in o.steps:
-                walk(step, f"{pfx}{name}.")
-        elif isinstance(o, FeatureUnion):
-            for name, t in o.transformer_list:
-                walk(t, f"{pfx}{name}.")
-        elif hasattr(o, "base_estimator"):
-            walk(o.base_estimator, f"{pfx}base_estimator.")
-
-    walk(obj, prefix)
     return tensors
+
+
+@contextmanager
+def _externalized_weights():
+    from ..model.models import TorchBaseEstimator
+
+    TorchBaseEstimator._externalize_weights = True
+    try:
+        yield
+    finally:
+        TorchBaseEstimator._externalize_weights = False
 
 
 def dump(obj: Any, dest_dir: str, metadata: Optional[Dict] = None):
     os.makedirs(dest_dir, exist_ok=True)
+    # Tensors go to weights.safetensors (mmap-able flat blobs — THE weight
+    # load path); the pickle then carries only the object graph + a marker
+    # per estimator. If safetensors export fails, fall back to a
+    # self-contained pickle with embedded weight bytes.
+    externalized = False
+    try:
+        tensors = _collect_torch_tensors(obj)
+        if tensors:
+            from safetensors.torch import save_file
+
+            save_file(tensors, os.path.join(dest_dir, "weights.safetensors"))
+            externalized = True
+    except Exception as e:
+        logger.warning("weights.safetensors not written: %s", e)
     with open(os.path.join(dest_dir, "model.pkl"), "wb") as f:
-        pickle.dump(obj, f)
+        if externalized:
+            with _externalized_weights():
+                pickle.dump(obj, f)
+        else:
+            pickle.dump(obj, f)
     if metadata is not None:
         with open(os.path.join(dest_dir, "metadata.json"), "w") as f:
             json.dump(metadata, f, default=str, indent=1)
@@ -85,14 +119,6 @@ def dump(obj: Any, dest_dir: str, metadata: Optional[Dict] = None):
             yaml.safe_dump(definition, f, default_flow_style=False)
     except Exception as e:
         logger.debug("definition.yaml not derivable for %s: %s", type(obj).__name__, e)
-    try:
-        tensors = _collect_torch_tensors(obj)
-        if tensors:
-            from safetensors.torch import save_file
-
-            save_file(tensors, os.path.join(dest_dir, "weights.safetensors"))
-    except Exception as e:
-        logger.warning("weights.safetensors not written: %s", e)
 
 
 def dumps(obj: Any) -> bytes:
@@ -106,11 +132,40 @@ def loads(data: bytes) -> Any:
 
 
 def load(source_dir: str) -> Any:
+    """Load an artifact dir: unpickle the object graph, then restore any
+    externalized torch weights from weights.safetensors (zero torch-tensor
+    bytes pass through the pickle on this path)."""
     path = os.path.join(source_dir, "model.pkl")
     if not os.path.exists(path):
         raise FileNotFoundError(f"No model.pkl under {source_dir}")
     with open(path, "rb") as f:
-        return pickle.load(f)
+        obj = pickle.load(f)
+
+    pending = [
+        (pfx, est)
+        for pfx, est in _walk_estimators(obj)
+        if getattr(est, "_needs_external_weights", False)
+    ]
+    if pending:
+        st_path = os.path.join(source_dir, "weights.safetensors")
+        if not os.path.exists(st_path):
+            raise FileNotFoundError(
+                f"{source_dir}: model.pkl references external weights but "
+                "weights.safetensors is missing"
+            )
+        from safetensors import safe_open
+
+        with safe_open(st_path, framework="pt", device="cpu") as sf:
+            keys = list(sf.keys())
+            for pfx, est in pending:
+                want = f"{pfx}model."
+                sd = {k[len(want):]: sf.get_tensor(k) for k in keys if k.startswith(want)}
+                if not sd:
+                    raise KeyError(
+                        f"weights.safetensors has no tensors under '{want}'"
+                    )
+                est.restore_weights(sd)
+    return obj
 
 
 def load_metadata(source_dir: str) -> Dict:
