@@ -1,4 +1,6 @@
 from .base import GordoBaseDataProvider
+from .iroc_reader import IrocFileProvider
+from .ncs_reader import NcsFileProvider
 from .providers import (
     CsvFileProvider,
     InfluxDataProvider,
@@ -12,4 +14,6 @@ __all__ = [
     "InfluxDataProvider",
     "ParquetFileProvider",
     "CsvFileProvider",
+    "NcsFileProvider",
+    "IrocFileProvider",
 ]

@@ -49,7 +49,8 @@ class GordoBaseDataProvider:
             module = importlib.import_module(module_name)
             provider_cls = getattr(module, cls_name)
         else:
-            from . import providers
-
-            provider_cls = getattr(providers, kind)
+            # short names resolve against the package namespace (providers,
+            # ncs_reader, iroc_reader re-exports)
+            pkg = importlib.import_module(__package__)
+            provider_cls = getattr(pkg, kind)
         return provider_cls(**config)
