@@ -25,14 +25,14 @@ from sklearn.metrics import (
     mean_squared_error,
     r2_score,
 )
-from sklearn.model_selection import TimeSeriesSplit, cross_validate
+from sklearn.model_selection import TimeSeriesSplit
 
 from .. import __version__
+from .cv import run_time_series_cv
 from ..dataset import dataset_from_config
 from ..machine import Machine
 from ..model.anomaly.base import AnomalyDetectorBase
 from ..model.base import GordoBase
-from ..model.utils import metric_wrapper
 from ..serializer import dump, from_definition
 from ..util import disk_registry
 
@@ -208,8 +208,11 @@ class ModelBuilder:
         Xv = _values(X)
         n_splits = int(evaluation.get("cv_splits", 3))
         cv = TimeSeriesSplit(n_splits=n_splits)
-        metric_names = evaluation.get("metrics") or ["explained_variance_score"]
-        metrics = {m: METRIC_FUNCS[m] for m in metric_names if m in METRIC_FUNCS}
+        metric_names = [
+            m
+            for m in (evaluation.get("metrics") or ["explained_variance_score"])
+            if m in METRIC_FUNCS
+        ]
 
         scoring_scaler = None
         scaler_def = evaluation.get("scoring_scaler")
@@ -217,39 +220,33 @@ class ModelBuilder:
             scoring_scaler = from_definition(scaler_def)
             scoring_scaler.fit(yv)
 
-        from sklearn.metrics import make_scorer
-
-        scorers = {}
-        tag_names = [
-            str(c) if isinstance(c, str) else str(c)
-            for c in getattr(X, "columns", range(Xv.shape[1]))
-        ]
+        tag_names = [str(c) for c in getattr(X, "columns", range(Xv.shape[1]))]
         per_tag = bool(evaluation.get("per_tag_metrics", True))
-        for mname, func in metrics.items():
-            scorers[mname] = make_scorer(metric_wrapper(func, scoring_scaler))
-            if per_tag:
-                # per-tag scores (reference parity); sklearn's multimetric
-                # scorer shares one predict across all of these
-                for i, tag in enumerate(tag_names):
-                    scorers[f"{mname}-{tag}"] = make_scorer(
-                        metric_wrapper(func, scoring_scaler, col=i)
-                    )
 
-        cv_duration = None
-        # threshold-deriving CV path: only when the TOP-level model is the
+        # torch-vectorized fold scoring (K9): one pass per fold covers every
+        # (metric, tag) pair — on GPU when present — instead of ~200 sklearn
+        # scorer invocations each re-transforming the validation block.
+        # Threshold-deriving CV path: only when the TOP-level model is the
         # anomaly detector (wrapping a full inner pipeline) — running the
         # detector's own CV under a partial pipeline would skip outer steps.
         anomaly_target = model if isinstance(model, AnomalyDetectorBase) else None
+        cv_kwargs = dict(
+            metric_names=metric_names,
+            scoring_scaler=scoring_scaler,
+            per_tag=per_tag,
+            tag_names=tag_names,
+        )
         if anomaly_target is not None:
-            cv_out = anomaly_target.cross_validate(X=Xv, y=yv, cv=cv, scoring=scorers)
+            cv_out = anomaly_target.cross_validate(X=Xv, y=yv, cv=cv, **cv_kwargs)
         else:
-            cv_out = cross_validate(
-                model, Xv, yv, cv=cv, scoring=scorers, return_estimator=False, n_jobs=1
-            )
+            cv_out = run_time_series_cv(model, Xv, yv, cv=cv, **cv_kwargs)
         cv_duration = time.time() - t0
 
         scores: Dict[str, Any] = {}
-        for mname in scorers:
+        score_keys = [m for m in metric_names]
+        if per_tag:
+            score_keys += [f"{m}-{t}" for m in metric_names for t in tag_names]
+        for mname in score_keys:
             key = f"test_{mname}"
             if key in cv_out:
                 vals = [float(v) for v in cv_out[key]]

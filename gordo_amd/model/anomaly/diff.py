@@ -17,7 +17,7 @@ from typing import Any, Dict, Optional
 import numpy as np
 import pandas as pd
 from sklearn.base import BaseEstimator
-from sklearn.model_selection import TimeSeriesSplit, cross_validate
+from sklearn.model_selection import TimeSeriesSplit
 
 from .base import AnomalyDetectorBase
 from ...serializer import from_definition, into_definition
@@ -123,21 +123,31 @@ class DiffBasedAnomalyDetector(BaseEstimator, AnomalyDetectorBase):
     ):
         """Run CV and derive per-feature + aggregate anomaly-score thresholds
         from the fold validation errors (92nd percentile, parity with the
-        reference's threshold derivation)."""
+        reference's threshold derivation).
+
+        Scoring runs through builder.cv.run_time_series_cv — every (metric,
+        tag) score computed in one vectorized torch pass per fold (GPU when
+        present), and the fold predictions are REUSED for the threshold
+        errors instead of re-predicting (the reference predicts twice).
+        kwargs: metric_names, scoring_scaler, per_tag, tag_names.
+        """
+        from ...builder.cv import run_time_series_cv
+
         cv = cv or TimeSeriesSplit(n_splits=3)
         Xv = X.values if isinstance(X, pd.DataFrame) else np.asarray(X)
         yv = y.values if isinstance(y, pd.DataFrame) else np.asarray(y)
 
-        cv_output = cross_validate(
-            self, X=Xv, y=yv, cv=cv, n_jobs=n_jobs if n_jobs is not None else 1,
-            return_estimator=True, **kwargs
+        cv_output = run_time_series_cv(
+            self, Xv, yv, cv=cv, return_estimator=True, return_predictions=True,
+            **kwargs,
         )
 
         per_fold_feature_thresholds = []
         per_fold_aggregate = []
-        for est, (_, test_idx) in zip(cv_output["estimator"], cv.split(Xv, yv)):
-            X_test, y_test = Xv[test_idx], yv[test_idx]
-            out = est.predict(X_test)
+        for est, (test_idx, out) in zip(
+            cv_output["estimator"], cv_output["predictions"]
+        ):
+            y_test = yv[test_idx]
             offset = len(y_test) - len(out)
             scaled_y = est.scaler.transform(y_test[offset:])
             scaled_out = est.scaler.transform(out)

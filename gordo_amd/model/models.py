@@ -155,6 +155,12 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
         self._n_features = X.shape[1]
         self._n_features_out = y.shape[1]
         if self.model is None:
+            # seed BEFORE the build so weight init is deterministic too (the
+            # trainer re-seeds before shuffling; without this the first fit
+            # in a process inherits whatever global RNG state preceded it)
+            seed = {**self.kwargs, **fit_args}.get("seed")
+            if seed is not None:
+                torch.manual_seed(int(seed))
             self.model = self._build(self._n_features, self._n_features_out)
 
         cfg = TrainConfig.from_kwargs({**self.kwargs, **fit_args})

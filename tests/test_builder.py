@@ -112,3 +112,97 @@ def test_cv_per_tag_scores(tmp_path):
     assert "explained_variance_score" in scores
     for i in range(3):
         assert f"explained_variance_score-tag-{i}" in scores
+
+
+# ---------------------------------------------------------------------------
+# Torch-vectorized CV scoring (builder/cv.py, K9) vs sklearn ground truth
+# ---------------------------------------------------------------------------
+
+def test_fold_scores_match_sklearn():
+    from sklearn.metrics import (
+        explained_variance_score,
+        mean_absolute_error,
+        mean_squared_error,
+        r2_score,
+    )
+    from sklearn.preprocessing import RobustScaler
+
+    from gordo_amd.builder.cv import fold_scores
+
+    rng = np.random.default_rng(0)
+    y = rng.normal(size=(300, 7))
+    pred = y + rng.normal(0, 0.3, size=(300, 7))
+    scaler = RobustScaler().fit(y)
+    names = [f"t{i}" for i in range(7)]
+    metrics = [
+        "explained_variance_score",
+        "r2_score",
+        "mean_squared_error",
+        "mean_absolute_error",
+    ]
+    got = fold_scores(y, pred, metrics, scaler=scaler, tag_names=names)
+    ys, ps = scaler.transform(y), scaler.transform(pred)
+    for fname, func in [
+        ("explained_variance_score", explained_variance_score),
+        ("r2_score", r2_score),
+        ("mean_squared_error", mean_squared_error),
+        ("mean_absolute_error", mean_absolute_error),
+    ]:
+        assert got[fname] == pytest.approx(func(ys, ps), rel=1e-10)
+        for i, t in enumerate(names):
+            assert got[f"{fname}-{t}"] == pytest.approx(
+                func(ys[:, i], ps[:, i]), rel=1e-10
+            ), (fname, t)
+
+
+def test_fold_scores_zero_variance_conventions():
+    from sklearn.metrics import explained_variance_score, r2_score
+
+    from gordo_amd.builder.cv import fold_scores
+
+    # constant y_true column: sklearn EV/R2 -> 0.0 when prediction differs,
+    # 1.0 when prediction is exact
+    y = np.ones((50, 1))
+    exact = np.ones((50, 1))
+    off = np.full((50, 1), 2.0)
+    m = ["explained_variance_score", "r2_score"]
+    assert fold_scores(y, exact, m, per_tag=False)["r2_score"] == 1.0
+    assert fold_scores(y, off, m, per_tag=False)["r2_score"] == r2_score(y, off)
+    assert (
+        fold_scores(y, off, m, per_tag=False)["explained_variance_score"]
+        == explained_variance_score(y, off)
+    )
+
+
+def test_fold_scores_trailing_offset_alignment():
+    from gordo_amd.builder.cv import fold_scores
+
+    y = np.arange(20, dtype=float)[:, None]
+    pred = y[5:]  # lookback model: fewer prediction rows, aligned to the end
+    s = fold_scores(y, pred, ["mean_absolute_error"], per_tag=False)
+    assert s["mean_absolute_error"] == 0.0
+
+
+def test_run_time_series_cv_shapes_and_estimators():
+    from sklearn.model_selection import TimeSeriesSplit
+
+    from gordo_amd.builder.cv import run_time_series_cv
+    from gordo_amd.model.models import TorchAutoEncoder
+
+    rng = np.random.default_rng(1)
+    X = rng.normal(size=(200, 5)).astype(np.float32)
+    model = TorchAutoEncoder(kind="feedforward_hourglass", epochs=1, batch_size=64, seed=0)
+    out = run_time_series_cv(
+        model, X, X, cv=TimeSeriesSplit(n_splits=3),
+        metric_names=["explained_variance_score", "mean_squared_error"],
+        tag_names=[f"t{i}" for i in range(5)],
+        return_estimator=True, return_predictions=True,
+    )
+    assert len(out["test_explained_variance_score"]) == 3
+    assert "test_mean_squared_error-t4" in out
+    assert len(out["estimator"]) == 3
+    assert len(out["predictions"]) == 3
+    idx, pred = out["predictions"][0]
+    assert pred.shape == (len(idx), 5)
+    # the original model must be untouched (clones are fit)
+    assert model.model is None
