@@ -44,14 +44,19 @@ logger = logging.getLogger(__name__)
 H4 = 4 * WMAX  # 256 (narrow variant; wide uses per-instance 4*W)
 KCAT = 2 * WMAX  # 128
 WMAX_WIDE = 128
+WMAX_WIDE256 = 256
 
 
 def lstm_width(module) -> int:
-    """Padded kernel width: 64 (chip-resident weights) or 128 (wide)."""
+    """Padded kernel width tier: 64 (chip-resident weights), 128 (wide,
+    L2-streamed weights) or 256 (column-half split cells)."""
     dims = [module.n_features, module.n_features_out]
     for cell in module.cells:
         dims += [cell.hidden_size, cell.input_size]
-    return WMAX if max(dims) <= WMAX else WMAX_WIDE
+    m = max(dims)
+    if m <= WMAX:
+        return WMAX
+    return WMAX_WIDE if m <= WMAX_WIDE else WMAX_WIDE256
 
 
 def lstm_supported(module) -> Optional[str]:
@@ -60,18 +65,27 @@ def lstm_supported(module) -> Optional[str]:
 
     if not isinstance(module, LSTMNet):
         return f"{type(module).__name__} is not LSTMNet"
-    if module.n_features > WMAX_WIDE:
-        return f"n_features {module.n_features} > {WMAX_WIDE}"
-    if module.n_features_out > WMAX_WIDE:
-        return f"n_features_out {module.n_features_out} > {WMAX_WIDE}"
+    if module.n_features > WMAX_WIDE256:
+        return f"n_features {module.n_features} > {WMAX_WIDE256}"
+    if module.n_features_out > WMAX_WIDE256:
+        return f"n_features_out {module.n_features_out} > {WMAX_WIDE256}"
     for cell in module.cells:
-        if cell.hidden_size > WMAX_WIDE or cell.input_size > WMAX_WIDE:
-            return f"LSTM dims {cell.input_size}->{cell.hidden_size} > {WMAX_WIDE}"
+        if cell.hidden_size > WMAX_WIDE256 or cell.input_size > WMAX_WIDE256:
+            return f"LSTM dims {cell.input_size}->{cell.hidden_size} > {WMAX_WIDE256}"
         if cell.num_layers != 1 or cell.bidirectional:
             return "only stacked unidirectional single-layer cells"
     if module.out_func not in ACT_IDS:
         return f"out_func {module.out_func!r} not fused"
     return None
+
+
+def _kernels_for(E, W: int):
+    """(fwd, bwd, dw) kernel triple for a width tier."""
+    if W == WMAX:
+        return E.lstm_fwd, E.lstm_bwd_cell, E.lstm_dw
+    if W == WMAX_WIDE:
+        return E.lstm_fwd_wide, E.lstm_bwd_wide, E.lstm_dw_wide
+    return E.lstm_fwd_wide256, E.lstm_bwd_wide256, E.lstm_dw_wide256
 
 
 class _LSTMPackedParams:
@@ -249,10 +263,15 @@ class FusedLSTMTrainer:
             # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
             st["dw_part"] = torch.zeros(512, WH4 * WKC, dtype=torch.float32, device=device)
             st["db_part"] = torch.zeros(512, WH4, dtype=torch.float32, device=device)
-        else:
+        elif W == WMAX_WIDE:
             # wide: [4 quadrants][NB][256*128] + dbias [2][NB][256]
             st["dw_part"] = torch.zeros(4, 256, 256 * 128, dtype=torch.float32, device=device)
             st["db_part"] = torch.zeros(2, 256, 256, dtype=torch.float32, device=device)
+        else:
+            # W=256: [16 quadrants][NB][256*128] + dbias [4][NB][256];
+            # NB=64 -> 16*64 = 1024 workgroups = 4/CU
+            st["dw_part"] = torch.zeros(16, 64, 256 * 128, dtype=torch.float32, device=device)
+            st["db_part"] = torch.zeros(4, 64, 256, dtype=torch.float32, device=device)
         return st
 
     def _fill_perm(self, st):
@@ -267,7 +286,7 @@ class FusedLSTMTrainer:
         """Run the stacked fused forward for window starts ``win`` [B] int32."""
         E, p = self.E, self.p
         B, T = st["B"], st["T"]
-        k_fwd = E.lstm_fwd if p.W == WMAX else E.lstm_fwd_wide
+        k_fwd, _, _ = _kernels_for(E, p.W)
         for l in range(p.Lc):
             src = st["Xd"] if l == 0 else st["h_seq"][l - 1]
             k_fwd(
@@ -304,7 +323,7 @@ class FusedLSTMTrainer:
         # dh_seq doubles as the dx output: within a block, dh_out[gi][t] is
         # read (elementwise phase) before dx_seq[gi][t] is written (GEMM
         # phase of the same t), and blocks own disjoint gi rows.
-        k_bwd = E.lstm_bwd_cell if p.W == WMAX else E.lstm_bwd_wide
+        _, k_bwd, k_dw = _kernels_for(E, p.W)
         for l in reversed(range(p.Lc)):
             dh = st["dx_head"] if l == p.Lc - 1 else st["dh_seq"]
             k_bwd(
@@ -313,28 +332,25 @@ class FusedLSTMTrainer:
                 l == p.Lc - 1,
             )
             x_seq = st["x_seq0"] if l == 0 else st["h_seq"][l - 1]
+            k_dw(
+                st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
+                B * T, T,
+            )
             if p.W == WMAX:
-                E.lstm_dw(
-                    st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
-                    B * T, T,
-                )
                 torch.sum(st["dw_part"], dim=0, out=st["gw"][l])
                 torch.sum(st["db_part"], dim=0, out=st["gb"][l])
                 p.wcat_master[l].grad = st["gw"][l].view(p.H4, p.KCAT)
                 p.bias_master[l].grad = st["gb"][l]
             else:
-                E.lstm_dw_wide(
-                    st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
-                    B * T, T,
-                )
-                # quadrants [gh][ih] -> gw [512][256]
-                q = st["dw_part"].sum(1).view(2, 2, 256, 128)
+                # quadrant tiles [gh][ih] of [256][128] -> gw [4W][2W]
+                ng = p.H4 // 256
+                ni = p.KCAT // 128
+                q = st["dw_part"].sum(1).view(ng, ni, 256, 128)
                 gw = st["gw"][l].view(p.H4, p.KCAT)
-                gw[:256, :128].copy_(q[0, 0])
-                gw[:256, 128:].copy_(q[0, 1])
-                gw[256:, :128].copy_(q[1, 0])
-                gw[256:, 128:].copy_(q[1, 1])
-                st["gb"][l].view(2, 256).copy_(st["db_part"].sum(1))
+                for gh in range(ng):
+                    for ih in range(ni):
+                        gw[gh * 256 : (gh + 1) * 256, ih * 128 : (ih + 1) * 128].copy_(q[gh, ih])
+                st["gb"][l].view(ng, 256).copy_(st["db_part"].sum(1))
                 p.wcat_master[l].grad = gw
                 p.bias_master[l].grad = st["gb"][l]
 
@@ -474,7 +490,7 @@ class FusedLSTMForward:
         h_seq = [torch.zeros(B, T, p.W, dtype=bf16, device=self.device) for _ in range(p.Lc)]
         c_seq = torch.zeros(B, T, p.W, dtype=bf16, device=self.device)
         gates = torch.zeros(B, T, p.H4, dtype=torch.uint8, device=self.device)
-        k_fwd = self.E.lstm_fwd if p.W == WMAX else self.E.lstm_fwd_wide
+        k_fwd, _, _ = _kernels_for(self.E, p.W)
         for s in range(0, n_windows, B):
             n = min(B, n_windows - s)
             win = torch.arange(s, s + n, dtype=torch.int32, device=self.device)

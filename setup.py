@@ -20,6 +20,7 @@ try:
         "gordo_amd/ops/hip/hourglass_kernels.hip",
         "gordo_amd/ops/hip/lstm_kernels.hip",
         "gordo_amd/ops/hip/lstm_wide_kernels.hip",
+        "gordo_amd/ops/hip/lstm_wide256_kernels.hip",
     ]
     hip_sources = [s for s in hip_sources if os.path.exists(s)]
     if hip_sources and torch.version.hip is not None:
