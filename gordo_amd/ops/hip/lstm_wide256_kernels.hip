@@ -87,10 +87,14 @@ __global__ void lstm_fwd_wide256(QFwdArgs a) {
     const int em = tid >> 3;
     const int ec0 = (tid & 7) * 16;
     const int egi = base + em;
-    float c_reg[32];                   // [half*16 + hh*8 + e]
+    // c/h state as TWO 16-element banks (one per column half). A single
+    // c_reg[32] indexed by the runtime `half` lowered to ~500 v_cndmask
+    // register-select instructions per half-pass (comparable to the 256
+    // MFMAs!); selecting the bank once per access group costs ~100.
+    float c_lo[16], c_hi[16];
 #pragma unroll
-    for (int e = 0; e < 32; ++e) c_reg[e] = 0.f;
-    unsigned short oh_reg[32];
+    for (int e = 0; e < 16; ++e) c_lo[e] = c_hi[e] = 0.f;
+    unsigned short oh_lo[16], oh_hi[16];
     __syncthreads();
 
     const unsigned short* wcat_p = a.wcat;
@@ -160,6 +164,10 @@ __global__ void lstm_fwd_wide256(QFwdArgs a) {
             {
                 const bool live = egi < a.B;
                 const size_t bt = (size_t)egi * a.T + t;
+                float cur[16];
+#pragma unroll
+                for (int e2 = 0; e2 < 16; ++e2) cur[e2] = half ? c_hi[e2] : c_lo[e2];
+                unsigned short ohv[16];
 #pragma unroll
                 for (int hh = 0; hh < 2; ++hh) {
                     int lc = ec0 + hh * 8;            // local col in half
@@ -176,12 +184,11 @@ __global__ void lstm_fwd_wide256(QFwdArgs a) {
                         float f_g = ga_sigmoid(ga_bf16_to_f32(vf[e]));
                         float g_g = ga_tanh(ga_bf16_to_f32(vg[e]));
                         float o_g = ga_sigmoid(ga_bf16_to_f32(vo[e]));
-                        int ei = half * 16 + hh * 8 + e;
-                        float c_new = f_g * c_reg[ei] + i_g * g_g;
-                        c_reg[ei] = c_new;
+                        float c_new = f_g * cur[hh * 8 + e] + i_g * g_g;
+                        cur[hh * 8 + e] = c_new;
                         unsigned short hv = ga_f32_to_bf16(o_g * ga_tanh(c_new));
                         oh[e] = hv;
-                        oh_reg[ei] = hv;
+                        ohv[hh * 8 + e] = hv;
                         oc[e] = ga_f32_to_bf16(c_new);
                         qi[e] = i_g;
                         qf[e] = f_g;
@@ -197,18 +204,26 @@ __global__ void lstm_fwd_wide256(QFwdArgs a) {
                         *(uint2*)(a.gates_seq + bt * Q4H + 3 * QW + cc) = ga_q8_pack8(qo);
                     }
                 }
+#pragma unroll
+                for (int e2 = 0; e2 < 16; ++e2) {
+                    if (half) { c_hi[e2] = cur[e2]; oh_hi[e2] = ohv[e2]; }
+                    else      { c_lo[e2] = cur[e2]; oh_lo[e2] = ohv[e2]; }
+                }
             }
             __syncthreads();  // pre_tiles reused by the next half
         }
 
         // ---- commit h_t to LDS (h_{t-1} fully consumed by both halves) ----
 #pragma unroll
-        for (int ei = 0; ei < 32; ei += 8) {
-            int cc = (ei < 16) ? (ec0 + (ei & 8)) : (128 + ec0 + (ei & 8));
-            ga_u16x8 v;
+        for (int hh = 0; hh < 2; ++hh) {
+            ga_u16x8 vlo, vhi;
 #pragma unroll
-            for (int e = 0; e < 8; ++e) v[e] = oh_reg[ei + e];
-            *(ga_u16x8*)&h_tile[sw256(em, cc)] = v;
+            for (int e = 0; e < 8; ++e) {
+                vlo[e] = oh_lo[hh * 8 + e];
+                vhi[e] = oh_hi[hh * 8 + e];
+            }
+            *(ga_u16x8*)&h_tile[sw256(em, ec0 + hh * 8)] = vlo;
+            *(ga_u16x8*)&h_tile[sw256(em, 128 + ec0 + hh * 8)] = vhi;
         }
         __syncthreads();
     }
@@ -253,14 +268,21 @@ __global__ void lstm_bwd_wide256(QBwdArgs a) {
         asm volatile("" : "+s"(wct_p));
         const bool need_rec = (t > 0);
         const bool need_dx = (a.dx_seq != nullptr);
-        // wave role: 0..3 -> dh_rec stripes, 4..7 -> dx stripes
-        const bool rec_wave = wid < 4;
+        // wave role: (otype, output-col-half, stripe-pair). Each wave covers
+        // TWO 16-row stripes per B-fragment load — a wave-per-stripe split
+        // would have all 4 stripes of an output loading IDENTICAL fragments
+        // (4x redundant L2 traffic); splitting the output columns instead
+        // halves the B loads at the same register cost (acc[2][8] == acc[16]).
+        const bool rec_wave = (wid >> 2) == 0;
         const bool active = rec_wave ? need_rec : need_dx;
-        const int stripe = (wid & 3) * 16;
-        const int brow0 = rec_wave ? QW : 0;  // wcat_t rows: h part | x part
-        ga_f32x4 acc[16];
+        const int np = (wid >> 1) & 1;   // output col half: nt 0..7 | 8..15
+        const int sub = wid & 1;         // stripe pair: rows sub*32..+32
+        const int brow0 = (rec_wave ? QW : 0) + np * 128;
+        ga_f32x4 acc[2][8];
 #pragma unroll
-        for (int nt = 0; nt < 16; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int st2 = 0; st2 < 2; ++st2)
+#pragma unroll
+            for (int nt = 0; nt < 8; ++nt) acc[st2][nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
 
         for (int half = 0; half < 2; ++half) {
             // ---- elementwise BPTT for this half's cols --------------------
@@ -329,14 +351,17 @@ __global__ void lstm_bwd_wide256(QBwdArgs a) {
             if (active) {
                 for (int kk = 0; kk < 512; kk += 32) {
                     const char* dgt = dg_tiles + (kk >> 7) * HTB;
-                    ga_bf16x8 afrag = *(const ga_bf16x8*)&dgt[qsw128(
-                        stripe + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
+                    ga_bf16x8 afrag0 = *(const ga_bf16x8*)&dgt[qsw128(
+                        sub * 32 + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
+                    ga_bf16x8 afrag1 = *(const ga_bf16x8*)&dgt[qsw128(
+                        sub * 32 + 16 + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
                     const size_t gc = (size_t)(kk >> 7) * QW + half * 128 + (kk & 127) + (lane >> 4) * 8;
 #pragma unroll
-                    for (int nt = 0; nt < 16; ++nt) {
+                    for (int nt = 0; nt < 8; ++nt) {
                         ga_bf16x8 bfrag = *(const ga_bf16x8*)(wct_p +
                             (size_t)(brow0 + nt * 16 + (lane & 15)) * Q4H + gc);
-                        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
+                        acc[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag0, bfrag, acc[0][nt], 0, 0, 0);
+                        acc[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag1, bfrag, acc[1][nt], 0, 0, 0);
                     }
                 }
             }
@@ -346,20 +371,22 @@ __global__ void lstm_bwd_wide256(QBwdArgs a) {
         // ---- write GEMM outputs ------------------------------------------
         if (active) {
 #pragma unroll
-            for (int nt = 0; nt < 16; ++nt) {
-                int col = nt * 16 + (lane & 15);
+            for (int st2 = 0; st2 < 2; ++st2)
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    int m = stripe + (lane >> 4) * 4 + r;
-                    if (rec_wave) {
-                        *(unsigned short*)&dh_tile[sw256(m, col)] = ga_f32_to_bf16(acc[nt][r]);
-                    } else {
-                        int gi = base + m;
-                        if (gi < a.B)
-                            a.dx_seq[((size_t)gi * a.T + t) * QW + col] = ga_f32_to_bf16(acc[nt][r]);
+                for (int nt = 0; nt < 8; ++nt) {
+                    int col = np * 128 + nt * 16 + (lane & 15);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int m = sub * 32 + st2 * 16 + (lane >> 4) * 4 + r;
+                        if (rec_wave) {
+                            *(unsigned short*)&dh_tile[sw256(m, col)] = ga_f32_to_bf16(acc[st2][nt][r]);
+                        } else {
+                            int gi = base + m;
+                            if (gi < a.B)
+                                a.dx_seq[((size_t)gi * a.T + t) * QW + col] = ga_f32_to_bf16(acc[st2][nt][r]);
+                        }
                     }
                 }
-            }
         }
         __syncthreads();
         if (!need_rec) break;
