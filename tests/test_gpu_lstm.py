@@ -161,10 +161,12 @@ def test_lstm_grads_fuzz():
     from gordo_amd.runtime.trainer import TrainConfig
 
     rng = np.random.default_rng(88)
-    for trial in range(4):
-        F = int(rng.integers(3, 129))
+    for trial in range(6):
+        # trials 0..3 sample the 64/128 tiers; 4..5 force the W=256 band
+        lo, hi = (3, 129) if trial < 4 else (129, 257)
+        F = int(rng.integers(lo, hi))
         L = int(rng.integers(1, 3))
-        dims = [int(rng.integers(4, 129)) for _ in range(L)]
+        dims = [int(rng.integers(max(4, lo // 2), hi)) for _ in range(L)]
         T = int(rng.integers(3, 17))
         B = 256
         torch.manual_seed(2000 + trial)
@@ -221,13 +223,15 @@ def test_lstm_grads_fuzz():
 
 def test_lstm_fused_forward_fuzz():
     """Seeded randomized shape fuzz of the fused LSTM forward vs fp32 torch:
-    features/hidden 2..128 (both width tiers), stacks 1..3, lookback 2..24,
-    awkward row counts — exercises the padding/packing envelope."""
+    features/hidden 2..256 (all three width tiers), stacks 1..3, lookback
+    2..24, awkward row counts — exercises the padding/packing envelope."""
     rng = np.random.default_rng(77)
-    for trial in range(10):
-        F = int(rng.integers(2, 129))
+    for trial in range(12):
+        # trials 0..9 sample the 64/128 tiers; 10..11 force the W=256 band
+        lo, hi = (2, 129) if trial < 10 else (129, 257)
+        F = int(rng.integers(lo, hi))
         L = int(rng.integers(1, 4))
-        dims = [int(rng.integers(2, 129)) for _ in range(L)]
+        dims = [int(rng.integers(max(2, lo), hi)) for _ in range(L)]
         T = int(rng.integers(2, 25))
         n = int(rng.integers(T + 1, T + 120))
         torch.manual_seed(1000 + trial)
