@@ -263,25 +263,27 @@ def test_lstm_fused_vs_eager_quality():
 
 
 def test_lstm_beyond_fused_envelope_gemmloop_gpu():
-    """>128-wide LSTM (200 tags): the fused cells don't apply; the estimator
-    must run the hipGraph GEMM-loop engine on GPU and still learn."""
+    """>256-wide LSTM (300 tags — past ALL three fused width tiers): the
+    estimator must run the hipGraph GEMM-loop engine on GPU and still learn.
+    (200-tag models moved onto the fused W=256 tier in round 2 and are
+    covered by test_gpu_lstm.py's w256 tests.)"""
     from gordo_amd.model.models import TorchLSTMAutoEncoder
 
     rng = np.random.default_rng(31)
     t = np.arange(3000)
     X = np.stack(
-        [np.sin(t / (8 + i % 13)) + 0.05 * rng.standard_normal(len(t)) for i in range(200)],
+        [np.sin(t / (8 + i % 13)) + 0.05 * rng.standard_normal(len(t)) for i in range(300)],
         axis=1,
     ).astype(np.float32)
     model = TorchLSTMAutoEncoder(
-        kind="lstm_symmetric", dims=[160], lookback_window=12, epochs=3,
+        kind="lstm_symmetric", dims=[280], lookback_window=12, epochs=3,
         batch_size=512, lr=3e-3,
     )
     model.fit(X)
     assert model.history["engine"].startswith("wide_lstm_gemmloop"), model.history
     assert model.history["loss"][-1] < model.history["loss"][0]
     out = model.predict(X[:100])
-    assert out.shape == (100 - 12 + 1, 200)
+    assert out.shape == (100 - 12 + 1, 300)
     assert np.isfinite(out).all()
 
 
