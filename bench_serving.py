@@ -20,6 +20,7 @@ import numpy as np
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 N_TAGS = 50
+EVALUATION = {"cv_mode": "no_cv"}
 
 
 def project_config(n_machines: int, n_tags: int):
@@ -56,7 +57,7 @@ def project_config(n_machines: int, n_tags: int):
                     }
                 }
             },
-            "evaluation": {"cv_mode": "no_cv"},
+            "evaluation": EVALUATION,
         },
     }
 
@@ -138,7 +139,22 @@ def main():
                     help="builder workers (default: one per GPU)")
     ap.add_argument("--http", action="store_true",
                     help="also measure end-to-end HTTP latency through the fleet server")
+    ap.add_argument("--cv", action="store_true",
+                    help="full_build evaluation: 3-fold TimeSeriesSplit CV with "
+                    "per-tag metrics + anomaly thresholds per machine (the "
+                    "torch-vectorized K9 path) instead of no_cv")
     args = ap.parse_args()
+    if args.cv:
+        EVALUATION.clear()
+        EVALUATION.update({
+            "cv_mode": "full_build",
+            "cv_splits": 3,
+            "metrics": [
+                "explained_variance_score", "r2_score",
+                "mean_squared_error", "mean_absolute_error",
+            ],
+            "per_tag_metrics": True,
+        })
 
     from gordo_amd.machine import NormalizedConfig
     from gordo_amd.server.model_io import ModelRuntime
@@ -201,6 +217,7 @@ def main():
         "models_built": ok,
         "build_seconds": build_s,
         "models_per_hour": ok / build_s * 3600,
+        "cv": bool(args.cv),
         "cold_load_seconds": cold_load_s,
         "cold_load_ms_per_model": cold_load_s / args.machines * 1e3,
         "higher_is_better": False,
