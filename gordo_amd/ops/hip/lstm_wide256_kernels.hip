@@ -268,21 +268,21 @@ __global__ void lstm_bwd_wide256(QBwdArgs a) {
         asm volatile("" : "+s"(wct_p));
         const bool need_rec = (t > 0);
         const bool need_dx = (a.dx_seq != nullptr);
-        // wave role: (otype, output-col-half, stripe-pair). Each wave covers
-        // TWO 16-row stripes per B-fragment load — a wave-per-stripe split
-        // would have all 4 stripes of an output loading IDENTICAL fragments
-        // (4x redundant L2 traffic); splitting the output columns instead
-        // halves the B loads at the same register cost (acc[2][8] == acc[16]).
+        // wave role: (otype, output-col-QUARTER). Each wave covers ALL FOUR
+        // 16-row stripes per B-fragment load: 8 waves = 2 outputs x 4 col
+        // quarters with zero cross-wave B-fragment redundancy (the
+        // wave-per-stripe layout was 4x redundant, the half/stripe-pair
+        // layout 2x — each halving measured double-digit gains) at the same
+        // register cost (acc[4][4] == acc[2][8]).
         const bool rec_wave = (wid >> 2) == 0;
         const bool active = rec_wave ? need_rec : need_dx;
-        const int np = (wid >> 1) & 1;   // output col half: nt 0..7 | 8..15
-        const int sub = wid & 1;         // stripe pair: rows sub*32..+32
-        const int brow0 = (rec_wave ? QW : 0) + np * 128;
-        ga_f32x4 acc[2][8];
+        const int nq = wid & 3;          // output col quarter: nt 0..3
+        const int brow0 = (rec_wave ? QW : 0) + nq * 64;
+        ga_f32x4 acc[4][4];
 #pragma unroll
-        for (int st2 = 0; st2 < 2; ++st2)
+        for (int st2 = 0; st2 < 4; ++st2)
 #pragma unroll
-            for (int nt = 0; nt < 8; ++nt) acc[st2][nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            for (int nt = 0; nt < 4; ++nt) acc[st2][nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
 
         for (int half = 0; half < 2; ++half) {
             // ---- elementwise BPTT for this half's cols --------------------
@@ -351,17 +351,19 @@ __global__ void lstm_bwd_wide256(QBwdArgs a) {
             if (active) {
                 for (int kk = 0; kk < 512; kk += 32) {
                     const char* dgt = dg_tiles + (kk >> 7) * HTB;
-                    ga_bf16x8 afrag0 = *(const ga_bf16x8*)&dgt[qsw128(
-                        sub * 32 + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
-                    ga_bf16x8 afrag1 = *(const ga_bf16x8*)&dgt[qsw128(
-                        sub * 32 + 16 + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
+                    ga_bf16x8 afrag[4];
+#pragma unroll
+                    for (int st2 = 0; st2 < 4; ++st2)
+                        afrag[st2] = *(const ga_bf16x8*)&dgt[qsw128(
+                            st2 * 16 + (lane & 15), (kk & 127) + (lane >> 4) * 8)];
                     const size_t gc = (size_t)(kk >> 7) * QW + half * 128 + (kk & 127) + (lane >> 4) * 8;
 #pragma unroll
-                    for (int nt = 0; nt < 8; ++nt) {
+                    for (int nt = 0; nt < 4; ++nt) {
                         ga_bf16x8 bfrag = *(const ga_bf16x8*)(wct_p +
                             (size_t)(brow0 + nt * 16 + (lane & 15)) * Q4H + gc);
-                        acc[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag0, bfrag, acc[0][nt], 0, 0, 0);
-                        acc[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag1, bfrag, acc[1][nt], 0, 0, 0);
+#pragma unroll
+                        for (int st2 = 0; st2 < 4; ++st2)
+                            acc[st2][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[st2], bfrag, acc[st2][nt], 0, 0, 0);
                     }
                 }
             }
@@ -371,13 +373,13 @@ __global__ void lstm_bwd_wide256(QBwdArgs a) {
         // ---- write GEMM outputs ------------------------------------------
         if (active) {
 #pragma unroll
-            for (int st2 = 0; st2 < 2; ++st2)
+            for (int st2 = 0; st2 < 4; ++st2)
 #pragma unroll
-                for (int nt = 0; nt < 8; ++nt) {
-                    int col = np * 128 + nt * 16 + (lane & 15);
+                for (int nt = 0; nt < 4; ++nt) {
+                    int col = nq * 64 + nt * 16 + (lane & 15);
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
-                        int m = sub * 32 + st2 * 16 + (lane >> 4) * 4 + r;
+                        int m = st2 * 16 + (lane >> 4) * 4 + r;
                         if (rec_wave) {
                             *(unsigned short*)&dh_tile[sw256(m, col)] = ga_f32_to_bf16(acc[st2][nt][r]);
                         } else {
