@@ -73,10 +73,19 @@ class Client:
         elif self.watchman_url:
             status = await client_io.fetch_json(self.watchman_url, session)
             for e in status.get("endpoints", []):
+                ep_url = e.get("endpoint", "").rstrip("/")
+                # the fleet server's root reports RELATIVE mount prefixes
+                # (/gordo/v0/<project>/<machine>); join them onto the
+                # discovery base like the reference client does with
+                # watchman-reported paths
+                if ep_url.startswith("/"):
+                    from urllib.parse import urljoin
+
+                    ep_url = urljoin(self.watchman_url, ep_url).rstrip("/")
                 eps.append(
                     EndpointMetadata(
                         target_name=e.get("target-name", e.get("name", "?")),
-                        endpoint=e.get("endpoint", "").rstrip("/"),
+                        endpoint=ep_url,
                         healthy=bool(e.get("healthy")),
                         metadata=e.get("metadata", {}),
                     )
