@@ -16,6 +16,7 @@ from __future__ import annotations
 
 import io
 import logging
+import threading
 from typing import Any, Dict, Optional
 
 import numpy as np
@@ -73,13 +74,14 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
     # artifact dir's weights.safetensors and serializer.load() mmaps them
     # back in — the zero-pickle fast load path. Plain pickle/dumps stays
     # self-contained (weights embedded as bytes, the reference's Keras-HDF5-
-    # inside-pickle idiom).
-    _externalize_weights: bool = False
+    # inside-pickle idiom). Thread-local so a concurrent dumps() (e.g. the
+    # server's /download-model) can never observe another thread's dump().
+    _externalize_tls = threading.local()
 
     def __getstate__(self):
         state = self.__dict__.copy()
         if self.model is not None:
-            if TorchBaseEstimator._externalize_weights:
+            if getattr(TorchBaseEstimator._externalize_tls, "on", False):
                 state["model"] = {
                     "__external_weights__": True,
                     "n_features": getattr(self, "_n_features", None),

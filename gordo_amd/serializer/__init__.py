@@ -81,11 +81,11 @@ def _collect_torch_tensors(obj: Any, prefix: str = "") -> Dict[str, "object"]:
 def _externalized_weights():
     from ..model.models import TorchBaseEstimator
 
-    TorchBaseEstimator._externalize_weights = True
+    TorchBaseEstimator._externalize_tls.on = True
     try:
         yield
     finally:
-        TorchBaseEstimator._externalize_weights = False
+        TorchBaseEstimator._externalize_tls.on = False
 
 
 def dump(obj: Any, dest_dir: str, metadata: Optional[Dict] = None):
