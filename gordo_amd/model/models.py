@@ -214,6 +214,11 @@ class TorchAutoEncoder(TorchBaseEstimator, TransformerMixin):
         X = self._as_array(X)
         return self._predict_array(X)
 
+    def transform(self, X) -> np.ndarray:
+        """TransformerMixin contract (FeatureUnion/pipeline middle steps):
+        the autoencoder's transform IS its reconstruction."""
+        return self.predict(X)
+
 
 class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
     supports_lookback = True
@@ -263,6 +268,10 @@ class TorchLSTMBaseEstimator(TorchBaseEstimator, TransformerMixin):
     def predict(self, X) -> np.ndarray:
         X = self._as_array(X)
         return self._window_predict(X)
+
+    def transform(self, X) -> np.ndarray:
+        """TransformerMixin contract: windowed reconstruction/forecast."""
+        return self.predict(X)
 
 
 class TorchLSTMAutoEncoder(TorchLSTMBaseEstimator):

@@ -169,3 +169,27 @@ def test_dumps_stays_self_contained_after_externalized_dump(tmp_path):
     assert b"PK\x03\x04" in blob  # embedded torch.save archive
     model2 = serializer.loads(blob)
     np.testing.assert_allclose(model.predict(X), model2.predict(X), rtol=1e-5, atol=1e-6)
+
+
+def test_externalized_feature_union_distinct_prefixes(tmp_path):
+    """Two torch estimators inside one FeatureUnion: the safetensors keys
+    must be disambiguated by the walk prefixes and both restored."""
+    fu = FeatureUnion(
+        [
+            ("a", TorchAutoEncoder(kind="feedforward_hourglass", epochs=1, seed=1)),
+            ("b", TorchAutoEncoder(kind="feedforward_hourglass", epochs=1, seed=2)),
+        ]
+    )
+    X = np.random.default_rng(6).normal(size=(80, 5)).astype(np.float32)
+    fu.fit(X)
+    expected = fu.transform(X)
+    serializer.dump(fu, str(tmp_path))
+    loaded = serializer.load(str(tmp_path))
+    np.testing.assert_allclose(expected, loaded.transform(X), rtol=1e-5, atol=1e-6)
+    # both estimators restored from distinct key prefixes
+    import pickle
+
+    with open(tmp_path / "model.pkl", "rb") as f:
+        raw = pickle.load(f)
+    ests = dict(serializer._walk_estimators(raw))
+    assert set(ests) == {"a.", "b."}
