@@ -141,3 +141,25 @@ def test_version_and_help_surfaces():
     ):
         r = runner.invoke(gordo, args)
         assert r.exit_code == 0, (args, r.output)
+
+
+def test_build_command_bad_config_fails_with_message(tmp_path):
+    """Builder-pod error reporting (reference: the build entrypoint surfaces
+    config errors with a non-zero exit instead of a silent crash)."""
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "build",
+            "--name", "broken",
+            "--data-config", '{"type": "RandomDataset", "tag_list": ["t0"], '
+            '"train_start_date": "2020-01-01T00:00:00Z", '
+            '"train_end_date": "2020-02-01T00:00:00Z"}',
+            "--model-config", '{"no.such.module.Model": {}}',
+            "--output-dir", str(tmp_path / "out"),
+        ],
+    )
+    assert result.exit_code != 0
+    # the failure reason reaches the operator (import error of the dotted path)
+    blob = (result.output or "") + str(result.exception)
+    assert "no.such" in blob or "No module" in blob
