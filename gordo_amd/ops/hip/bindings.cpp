@@ -31,6 +31,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_bwd_wide256", &lstm_bwd_wide256_launch, "W=256 reverse-time BPTT cell (+fused dx)");
     m.def("lstm_dw_wide256", &lstm_dw_wide256_launch, "W=256 split-K dWcat 4x4 quadrants + dbias");
     m.attr("WMAX") = 64;
-    m.attr("MAXL") = 5;
+    m.attr("MAXL") = 8;
     m.attr("MT") = 32;   // hg_train_step batch rows per block/slab
 }

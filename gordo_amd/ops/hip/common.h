@@ -7,7 +7,9 @@
 #define GA_WAVE 64
 #define GA_WMAX 64   // padded layer width (max supported by fused path)
 #define GA_MT 64     // batch rows per block tile
-#define GA_MAXL 5    // max layers in fused path
+#define GA_MAXL 8    // arg-array cap; the REAL per-width layer limit
+                     // is LDS ((L+4) act/scratch tiles must fit 160 KiB
+                     // -> L<=8 at W=64/128, L<=5 at W=256)
 
 typedef __bf16 ga_bf16x8 __attribute__((ext_vector_type(8)));
 typedef float ga_f32x4 __attribute__((ext_vector_type(4)));

@@ -683,7 +683,8 @@ __global__ void hg_adam_flat(
 
 static void ga_fill_acts(int* dst, const std::vector<int64_t>& acts, int L) {
     TORCH_CHECK((int)acts.size() == L, "act id list must have L entries");
-    TORCH_CHECK(L >= 1 && L <= GA_MAXL, "fused path supports 1..5 layers");
+    TORCH_CHECK(L >= 1 && L <= GA_MAXL, "fused path supports 1..", GA_MAXL,
+                " layers (LDS-capped per width; see engine supported())");
     for (int i = 0; i < L; ++i) dst[i] = (int)acts[i];
 }
 

@@ -24,7 +24,18 @@ logger = logging.getLogger(__name__)
 WMAX = 64         # narrow-variant padded width (also the LSTM head width)
 WMAX_WIDE = 128   # wide-variant padded width
 WMAX_MAX = 256    # widest instantiation (> this -> library-GEMM fallback)
-MAXL = 5
+MAXL = 8          # kernel arg-array cap (common.h GA_MAXL)
+HG_MT = 32        # batch rows per block tile
+
+
+def max_layers_for(wmax: int) -> int:
+    """Per-width fused layer limit: (L+4) activation/scratch tiles + the
+    bias image must fit the 160 KiB LDS. 8 at W=64/128, 5 at W=256 —
+    covers the factory DEFAULT hourglass (encoding_layers=3 -> L=7 with
+    the duplicated-bottleneck composition) on <=128-wide models."""
+    tb = HG_MT * wmax * 2
+    budget = 160 * 1024 - MAXL * wmax * 4 - 1024
+    return max(1, min(MAXL, budget // tb - 4))
 ACT_IDS = {"linear": 0, None: 0, "tanh": 1, "relu": 2, "sigmoid": 3}
 
 
@@ -42,8 +53,9 @@ def supported(module) -> Optional[str]:
     if max(widths) > WMAX_MAX:
         return f"width {max(widths)} > {WMAX_MAX}"
     L = len(widths) - 1
-    if L < 1 or L > MAXL:
-        return f"L={L} outside 1..{MAXL}"
+    lmax = max_layers_for(wmax_for(widths))
+    if L < 1 or L > lmax:
+        return f"L={L} outside 1..{lmax} (width tier {wmax_for(widths)})"
     for f in module.layer_funcs:
         if f not in ACT_IDS:
             return f"activation {f!r} not fused"

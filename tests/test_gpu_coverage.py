@@ -538,3 +538,101 @@ def test_fused_forward_fuzz():
             out, ref, atol=6e-2, rtol=6e-2,
             err_msg=f"trial {trial}: widths={widths} funcs={funcs}+{out_func} B={B}",
         )
+
+
+def test_default_hourglass_l7_fused_grads_match_autograd():
+    """The FACTORY-DEFAULT hourglass (encoding_layers=3 -> 7 layers with the
+    duplicated-bottleneck composition) must run the fused path at W=64
+    (GA_MAXL raised to 8 in round 2 — it silently fell back to eager
+    before). One-step grads vs fp32 autograd."""
+    from gordo_amd.model.factories import feedforward_hourglass
+    from gordo_amd.ops import ext
+    from gordo_amd.ops.hourglass_engine import (
+        _PackedParams,
+        pad_rows_bf16,
+        supported,
+    )
+
+    torch.manual_seed(9)
+    net = feedforward_hourglass(50, 50)  # defaults: 3 encoding layers
+    assert len(net.layer_widths) == 8  # 50-42-33-25-25-33-42-50
+    assert supported(net) is None
+    ref_net = feedforward_hourglass(50, 50)
+    ref_net.load_state_dict(net.state_dict())
+    X = np.random.default_rng(9).normal(size=(256, 50)).astype(np.float32)
+
+    E = ext()
+    p = _PackedParams(net, DEV)
+    assert p.wmax == 64 and p.L == 7
+    WM, L, F_out = p.wmax, p.L, 50
+    B = 256
+    MT = int(E.MT)
+    n_tiles = (B + MT - 1) // MT
+    Xd = pad_rows_bf16(X, DEV, WM)
+    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
+    db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+    loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
+    gscale = 2.0 / (B * F_out)
+    E.hg_train_step(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                    loss, B, F_out, p.act_ids, gscale, True, None, WM, 0, 0,
+                    [int(w) for w in p.widths])
+    torch.cuda.synchronize()
+
+    Xref = Xd[:, :50].float().cpu()
+    loss_ref = torch.nn.functional.mse_loss(ref_net(Xref), Xref)
+    loss_ref.backward()
+    kernel_loss = float(loss.item()) / (B * F_out)
+    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+
+    linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
+    assert len(linears) == 7
+    dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
+    db_sum = db.sum(dim=0).view(L, WM).cpu()
+    for i, lin in enumerate(linears):
+        o, ninp = lin.weight.shape
+        got_w = dw_sum[i, :ninp, :o].T
+        ref_w = lin.weight.grad.detach()
+        scale = ref_w.abs().max().item() or 1.0
+        assert (got_w - ref_w).abs().max().item() < 0.06 * scale + 2e-4, f"layer {i} dW"
+        got_b = db_sum[i, :o]
+        ref_b = lin.bias.grad.detach()
+        bscale = ref_b.abs().max().item() or 1.0
+        assert (got_b - ref_b).abs().max().item() < 0.06 * bscale + 2e-4, f"layer {i} db"
+        assert dw_sum[i, ninp:, :].abs().max().item() == 0.0
+
+
+def test_default_hourglass_l7_trains_and_serves_fused(tmp_path):
+    """End-to-end: default-kind estimator trains on the fused engine AND the
+    server runtime builds the fused anomaly forward for the L=7 model."""
+    import pandas as pd
+
+    from gordo_amd.model.anomaly.diff import DiffBasedAnomalyDetector
+    from gordo_amd.model.models import TorchAutoEncoder
+    from gordo_amd.serializer import dump
+    from gordo_amd.server.model_io import ModelRuntime
+    from sklearn.pipeline import Pipeline
+    from sklearn.preprocessing import MinMaxScaler
+
+    det = DiffBasedAnomalyDetector(
+        base_estimator=Pipeline(
+            [
+                ("scaler", MinMaxScaler()),
+                ("model", TorchAutoEncoder(
+                    kind="feedforward_hourglass", epochs=3, batch_size=4096, seed=0
+                )),  # DEFAULT encoding_layers=3 -> L=7
+            ]
+        )
+    )
+    X = np.random.default_rng(5).uniform(0, 1, size=(20000, 50)).astype(np.float32)
+    det.fit(X)
+    est = det.base_estimator.steps[-1][1]
+    assert est.history["engine"].startswith("fused_hip"), est.history["engine"]
+
+    d = tmp_path / "m"
+    dump(det, str(d), metadata={"dataset": {"tag_list": [{"name": f"t{i}"} for i in range(50)]}})
+    rt = ModelRuntime(str(d))
+    assert rt._fused is not None, "fused serving path must cover the default model"
+    req = pd.DataFrame(X[:500], columns=[f"t{i}" for i in range(50)])
+    frame = rt.anomaly(req, None)
+    assert np.isfinite(frame["total-anomaly-score"].values).all()
