@@ -465,9 +465,15 @@ def test_fused_training_grads_fuzz():
     MT = int(E.MT)
     rng = np.random.default_rng(321)
     acts = ["tanh", "relu", "sigmoid", "linear"]
-    for trial in range(6):
-        L = int(rng.integers(1, 5))
-        widths = [int(rng.integers(3, 257)) for _ in range(L + 1)]
+    for trial in range(8):
+        # trials 0..5: widths up to 256 with 1..4 layers; 6..7: deep (5..8
+        # layers) narrow stacks — the raised W=64/128 layer limit (r2)
+        if trial < 6:
+            L = int(rng.integers(1, 5))
+            widths = [int(rng.integers(3, 257)) for _ in range(L + 1)]
+        else:
+            L = int(rng.integers(5, 9))
+            widths = [int(rng.integers(3, 129)) for _ in range(L + 1)]
         funcs = [str(rng.choice(acts))] * (L - 1)
         torch.manual_seed(500 + trial)
         net = FeedForwardNet(widths[0], widths[-1], widths[1:-1], funcs, "tanh")
@@ -523,9 +529,13 @@ def test_fused_forward_fuzz():
     widths 2..256, 1..5 layers, mixed activations, awkward batch sizes."""
     rng = np.random.default_rng(123)
     acts = ["tanh", "relu", "sigmoid", "linear"]
-    for trial in range(25):
-        L = int(rng.integers(1, 6))
-        widths = [int(rng.integers(2, 257)) for _ in range(L + 1)]
+    for trial in range(28):
+        if trial < 25:
+            L = int(rng.integers(1, 6))
+            widths = [int(rng.integers(2, 257)) for _ in range(L + 1)]
+        else:  # deep narrow stacks (6..8 layers, <=128 wide) — r2 limit
+            L = int(rng.integers(6, 9))
+            widths = [int(rng.integers(2, 129)) for _ in range(L + 1)]
         funcs = [str(rng.choice(acts)) for _ in range(L - 1)]
         out_func = str(rng.choice(acts))
         B = int(rng.integers(1, 200))
