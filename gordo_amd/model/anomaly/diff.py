@@ -87,7 +87,11 @@ class DiffBasedAnomalyDetector(BaseEstimator, AnomalyDetectorBase):
             y = X.values if isinstance(X, pd.DataFrame) else X
         self.scaler.fit(y)
         self.base_estimator.fit(X, y)
+        self.n_features_in_ = np.asarray(X).shape[1]  # sklearn fitted marker
         return self
+
+    def __sklearn_is_fitted__(self) -> bool:
+        return hasattr(self, "n_features_in_")
 
     def predict(self, X):
         return self.base_estimator.predict(X)

@@ -118,6 +118,9 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
             self.model.load_state_dict(payload["state_dict"])
             self.model.eval()
 
+    def __sklearn_is_fitted__(self) -> bool:
+        return self.model is not None
+
     def restore_weights(self, state_dict: Dict[str, torch.Tensor]) -> None:
         """Rebuild the module and load externally-stored tensors (the
         weights.safetensors half of an externalized artifact)."""
@@ -156,6 +159,10 @@ class TorchBaseEstimator(BaseEstimator, GordoBase):
 
         self._n_features = X.shape[1]
         self._n_features_out = y.shape[1]
+        # sklearn's standard fitted marker: without a trailing-underscore
+        # attribute, check_is_fitted() deems the estimator unfitted even
+        # after fit() (a hard error inside Pipeline.predict from sklearn 1.8)
+        self.n_features_in_ = X.shape[1]
         if self.model is None:
             # seed BEFORE the build so weight init is deterministic too (the
             # trainer re-seeds before shuffling; without this the first fit
