@@ -88,19 +88,27 @@ class EndpointStatuses:
 
 
 def build_app(project_name: str, targets: Dict[str, str], refresh_interval: float = 10.0):
+    from contextlib import asynccontextmanager
+
     from fastapi import FastAPI
 
     statuses = EndpointStatuses(targets, refresh_interval)
-    app = FastAPI(title=f"gordo-amd watchman [{project_name}]", version=__version__)
-    app.state.statuses = statuses
 
-    @app.on_event("startup")
-    async def _startup():
+    @asynccontextmanager
+    async def lifespan(app):
+        # lifespan context (the on_event decorators are deprecated)
         statuses.start()
+        try:
+            yield
+        finally:
+            statuses.stop()
 
-    @app.on_event("shutdown")
-    async def _shutdown():
-        statuses.stop()
+    app = FastAPI(
+        title=f"gordo-amd watchman [{project_name}]",
+        version=__version__,
+        lifespan=lifespan,
+    )
+    app.state.statuses = statuses
 
     @app.get("/")
     async def root():
