@@ -45,26 +45,37 @@ def build_app(model_dir: str, device=None):
     async def _extract_frames(request: Request):
         ctype = request.headers.get("content-type", "")
         body = await request.body()
-        if "arrow" in ctype:
-            X = dataframe_from_arrow_bytes(body)
-            y = None
-        elif "octet-stream" in ctype or "parquet" in ctype:
-            X = dataframe_from_parquet_bytes(body)
-            y = None
-        else:
-            import json
+        try:
+            if "arrow" in ctype:
+                X = dataframe_from_arrow_bytes(body)
+                y = None
+            elif "octet-stream" in ctype or "parquet" in ctype:
+                X = dataframe_from_parquet_bytes(body)
+                y = None
+            else:
+                import json
 
-            payload = json.loads(body) if body else {}
-            if "X" not in payload and "data" not in payload:
-                return None, None, JSONResponse(
-                    {"error": "Body must contain 'X'"}, status_code=400
-                )
-            Xp = payload.get("X", payload)
-            X = dataframe_from_dict(Xp if isinstance(Xp, dict) else {"data": Xp})
-            y = None
-            if payload.get("y") is not None:
-                yp = payload["y"]
-                y = dataframe_from_dict(yp if isinstance(yp, dict) else {"data": yp})
+                payload = json.loads(body) if body else {}
+                if not isinstance(payload, dict) or (
+                    "X" not in payload and "data" not in payload
+                ):
+                    return None, None, JSONResponse(
+                        {"error": "Body must contain 'X'"}, status_code=400
+                    )
+                Xp = payload.get("X", payload)
+                X = dataframe_from_dict(Xp if isinstance(Xp, dict) else {"data": Xp})
+                y = None
+                if payload.get("y") is not None:
+                    yp = payload["y"]
+                    y = dataframe_from_dict(yp if isinstance(yp, dict) else {"data": yp})
+        except Exception as e:
+            # undecodable body (corrupt parquet/arrow, invalid JSON, ragged
+            # arrays): a client error, not a server fault — the reference
+            # 400s on bad payloads
+            return None, None, JSONResponse(
+                {"error": f"Could not decode request body as {ctype or 'json'}: {e}"},
+                status_code=400,
+            )
         # column sanity: when metadata lists tags and X has plain columns,
         # reorder to the training tag order. A width match alone is NOT
         # enough for a frame with real column names (a misordered parquet

@@ -268,3 +268,29 @@ def test_positional_value_arrays_still_accepted(client):
     X = _x_frame(6)
     r = client.post("/prediction", json={"X": X.values.tolist()})
     assert r.status_code == 200, r.text
+
+
+def test_corrupt_parquet_body_is_400(client):
+    r = client.post(
+        "/prediction",
+        content=b"\x00\x01 not parquet at all",
+        headers={"Content-Type": "application/octet-stream"},
+    )
+    assert r.status_code == 400
+    assert "decode" in r.json()["error"].lower()
+
+
+def test_invalid_json_body_is_400(client):
+    r = client.post(
+        "/prediction",
+        content=b"{not json",
+        headers={"Content-Type": "application/json"},
+    )
+    assert r.status_code == 400
+
+
+def test_json_non_dict_body_is_400(client):
+    r = client.post(
+        "/prediction", content=b"[1, 2, 3]", headers={"Content-Type": "application/json"}
+    )
+    assert r.status_code == 400
