@@ -155,12 +155,16 @@ class TimeSeriesDataset(GordoBaseDataset):
                 f"(threshold {self.n_samples_threshold})"
             )
 
-        x_names = to_list_of_strings(self.tag_list)
-        X = data[[c for c in data.columns if _col_tag(c) in set(x_names)]]
+        # column order follows the CONFIGURED tag order (not join order):
+        # downstream naming — model outputs, thresholds, anomaly columns —
+        # assumes X matches tag_list and y matches target_tag_list
+        def _ordered(names: List[str]) -> List:
+            return [c for n in names for c in data.columns if _col_tag(c) == n]
+
+        X = data[_ordered(to_list_of_strings(self.tag_list))]
         y = None
         if self.target_tag_list:
-            y_names = set(to_list_of_strings(self.target_tag_list))
-            y = data[[c for c in data.columns if _col_tag(c) in y_names]]
+            y = data[_ordered(to_list_of_strings(self.target_tag_list))]
 
         self._metadata = {
             "tag_list": [t.to_json() for t in self.tag_list],

@@ -177,3 +177,17 @@ def test_multiple_aggregation_methods():
     assert ("tag-0", "mean") in X.columns and ("tag-1", "max") in X.columns
     assert X.shape[1] == 4
     assert (X["tag-0"]["max"] >= X["tag-0"]["mean"] - 1e-9).all()
+
+
+def test_target_tag_order_preserved():
+    """y columns must follow target_tag_list ORDER (not the join order):
+    downstream output naming assumes it."""
+    ds = RandomDataset(
+        train_start_date="2020-01-01T00:00:00Z",
+        train_end_date="2020-02-01T00:00:00Z",
+        tag_list=["tag-a", "tag-b", "tag-c"],
+        target_tag_list=["tag-c", "tag-a"],
+    )
+    X, y = ds.get_data()
+    assert list(X.columns) == ["tag-a", "tag-b", "tag-c"]
+    assert list(y.columns) == ["tag-c", "tag-a"]
