@@ -88,27 +88,46 @@ def _kernels_for(E, W: int):
     return E.lstm_fwd_wide256, E.lstm_bwd_wide256, E.lstm_dw_wide256
 
 
+def _tier(width: int) -> int:
+    """Smallest kernel width tier covering a real width."""
+    if width <= WMAX:
+        return WMAX
+    return WMAX_WIDE if width <= WMAX_WIDE else WMAX_WIDE256
+
+
 class _LSTMPackedParams:
-    """Padded fp32 masters + bf16 kernel weights for the stacked LSTM + head."""
+    """Padded fp32 masters + bf16 kernel weights for the stacked LSTM + head.
+
+    Each cell runs at its OWN width tier (64 register-resident / 128
+    L2-streamed / 256 column-half-split): a 200-tag hourglass stack is
+    [256, 256, 128, 256] instead of uniformly 256, and a 100-tag stack
+    [128, 128, 64, 128] — the narrow tiers are several times cheaper per
+    layer. Tier boundaries are bridged by zero-padded slice copies of the
+    h/dh streams in the engine (~0.2 ms per boundary per step at B=16K,
+    negligible vs the cell kernels)."""
 
     def __init__(self, module, device):
         self.module = module
         self.device = device
         self.Lc = len(module.cells)
-        W = lstm_width(module)
-        self.W = W
-        self.H4 = 4 * W
-        self.KCAT = 2 * W
+        # per-cell tier: must cover both the (padded) input and the hidden
+        self.Wl: List[int] = [
+            _tier(max(cell.input_size, cell.hidden_size)) for cell in module.cells
+        ]
+        self.W = self.Wl[0]        # input padding width (dataset rows)
+        self.H4 = 4 * self.W
+        self.KCAT = 2 * self.W
         self.wcat_master: List[torch.Tensor] = []
         self.bias_master: List[torch.Tensor] = []
         self.wcat_bf16: List[torch.Tensor] = []
         self.wcat_t_bf16: List[torch.Tensor] = []
         self.h_real: List[int] = []
-        for cell in module.cells:
+        for l, cell in enumerate(module.cells):
             H, I = cell.hidden_size, cell.input_size
+            W = self.Wl[l]
             self.h_real.append(H)
-            wcat = torch.zeros(self.H4, self.KCAT, dtype=torch.float32, device=device)
-            bias = torch.zeros(self.H4, dtype=torch.float32, device=device)
+            wcat = torch.zeros(4 * W, 2 * W, dtype=torch.float32, device=device)
+            bias = torch.zeros(4 * W, dtype=torch.float32, device=device)
             w_ih = cell.weight_ih_l0.detach().to(device, torch.float32)  # [4H, I]
             w_hh = cell.weight_hh_l0.detach().to(device, torch.float32)  # [4H, H]
             b = (cell.bias_ih_l0 + cell.bias_hh_l0).detach().to(device, torch.float32)
@@ -120,9 +139,11 @@ class _LSTMPackedParams:
             self.bias_master.append(bias)
             self.wcat_bf16.append(wcat.to(torch.bfloat16).contiguous())
             self.wcat_t_bf16.append(wcat.t().contiguous().to(torch.bfloat16))
-        # dense head: bf16 w [1][W][W] (out,in), wt (in,out), fp32 bias [W]
-        self.head_wm = torch.zeros(W, W, dtype=torch.float32, device=device)
-        self.head_bm = torch.zeros(W, dtype=torch.float32, device=device)
+        # dense head on h_{T-1}: its own tier over (H_last, F_out)
+        W_head = _tier(max(module.cells[-1].hidden_size, module.n_features_out))
+        self.W_head = W_head
+        self.head_wm = torch.zeros(W_head, W_head, dtype=torch.float32, device=device)
+        self.head_bm = torch.zeros(W_head, dtype=torch.float32, device=device)
         out_w, in_w = module.head.weight.shape
         self.head_wm[:out_w, :in_w] = module.head.weight.detach().to(device, torch.float32)
         self.head_bm[:out_w] = module.head.bias.detach().to(device, torch.float32)
@@ -143,9 +164,9 @@ class _LSTMPackedParams:
 
     def sync_to_module(self):
         with torch.no_grad():
-            W = self.W
             for l, cell in enumerate(self.module.cells):
                 H, I = cell.hidden_size, cell.input_size
+                W = self.Wl[l]
                 wcat = self.wcat_master[l]
                 bias = self.bias_master[l]
                 for g in range(4):
@@ -227,51 +248,87 @@ class FusedLSTMTrainer:
         return history
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _dw_buffers(tier: int, device):
+        """Per-tier dW split-K partial buffers (layers of one tier share)."""
+        f32 = torch.float32
+        if tier == WMAX:
+            # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
+            return (
+                torch.zeros(512, 4 * tier * 2 * tier, dtype=f32, device=device),
+                torch.zeros(512, 4 * tier, dtype=f32, device=device),
+            )
+        if tier == WMAX_WIDE:
+            # wide: [4 quadrants][NB][256*128] + dbias [2][NB][256]
+            return (
+                torch.zeros(4, 256, 256 * 128, dtype=f32, device=device),
+                torch.zeros(2, 256, 256, dtype=f32, device=device),
+            )
+        # W=256: [16 quadrants][NB][256*128] + dbias [4][NB][256]
+        return (
+            torch.zeros(16, 64, 256 * 128, dtype=f32, device=device),
+            torch.zeros(4, 64, 256, dtype=f32, device=device),
+        )
+
     def _alloc_step_state(self, Xd, Yd, B, T, n_batches, n_windows, lookahead, F_out):
         device = self.device
         p = self.p
-        Lc, W, WH4, WKC = p.Lc, p.W, p.H4, p.KCAT
+        Lc = p.Lc
         bf16 = torch.bfloat16
         gates_dt = torch.uint8  # u8 fixed-point gate streams (common.h ga_q8_*)
         MT = int(getattr(self.E, "MT", 32))
         n_head_tiles = (B + MT - 1) // MT
+        WH = p.W_head
         st = {
             "Xd": Xd, "Yd": Yd, "B": B, "T": T, "n_batches": n_batches,
             "n_windows": n_windows, "lookahead": lookahead, "F_out": F_out,
             "perm": torch.zeros(n_batches * B, dtype=torch.int32, device=device),
-            "h_seq": [torch.zeros(B, T, W, dtype=bf16, device=device) for _ in range(Lc)],
-            "c_seq": [torch.zeros(B, T, W, dtype=bf16, device=device) for _ in range(Lc)],
+            "h_seq": [torch.zeros(B, T, p.Wl[l], dtype=bf16, device=device) for l in range(Lc)],
+            "c_seq": [torch.zeros(B, T, p.Wl[l], dtype=bf16, device=device) for l in range(Lc)],
             # activated gates stored quantized u8 (common.h ga_q8_*: i,f,o
             # as v*255, g offset) — halves the biggest BPTT stream
-            "gates_seq": [torch.zeros(B, T, WH4, dtype=gates_dt, device=device) for _ in range(Lc)],
-            "x_seq0": torch.zeros(B, T, W, dtype=bf16, device=device),
-            "dgates": torch.zeros(B, T, WH4, dtype=bf16, device=device),
-            "dh_seq": torch.zeros(B, T, W, dtype=bf16, device=device),
-            "gw": [torch.zeros(WH4 * WKC, dtype=torch.float32, device=device) for _ in range(Lc)],
-            "gb": [torch.zeros(WH4, dtype=torch.float32, device=device) for _ in range(Lc)],
-            "h_last": torch.zeros(B, W, dtype=bf16, device=device),
-            "y_head": torch.zeros(B, W, dtype=bf16, device=device),
-            "dx_head": torch.zeros(B, W, dtype=bf16, device=device),
+            "gates_seq": [torch.zeros(B, T, 4 * p.Wl[l], dtype=gates_dt, device=device) for l in range(Lc)],
+            "x_seq0": torch.zeros(B, T, p.Wl[0], dtype=bf16, device=device),
+            "gw": [torch.zeros(4 * p.Wl[l] * 2 * p.Wl[l], dtype=torch.float32, device=device) for l in range(Lc)],
+            "gb": [torch.zeros(4 * p.Wl[l], dtype=torch.float32, device=device) for l in range(Lc)],
+            "h_last": torch.zeros(B, WH, dtype=bf16, device=device),
+            "y_head": torch.zeros(B, WH, dtype=bf16, device=device),
+            "dx_head": torch.zeros(B, WH, dtype=bf16, device=device),
             "counters0": torch.tensor([0, 1, 0], dtype=torch.int32, device=device),
-            "head_dw": torch.zeros(n_head_tiles, 1, W * W, dtype=torch.float32, device=device),
-            "head_db": torch.zeros(n_head_tiles, 1, W, dtype=torch.float32, device=device),
+            "head_dw": torch.zeros(n_head_tiles, 1, WH * WH, dtype=torch.float32, device=device),
+            "head_db": torch.zeros(n_head_tiles, 1, WH, dtype=torch.float32, device=device),
             "loss_acc": torch.zeros(1, dtype=torch.float32, device=device),
             "gscale": 2.0 / (B * F_out),
             "act_out": [ACT_IDS[self.module.out_func]],
         }
-        if W == WMAX:
-            # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
-            st["dw_part"] = torch.zeros(512, WH4 * WKC, dtype=torch.float32, device=device)
-            st["db_part"] = torch.zeros(512, WH4, dtype=torch.float32, device=device)
-        elif W == WMAX_WIDE:
-            # wide: [4 quadrants][NB][256*128] + dbias [2][NB][256]
-            st["dw_part"] = torch.zeros(4, 256, 256 * 128, dtype=torch.float32, device=device)
-            st["db_part"] = torch.zeros(2, 256, 256, dtype=torch.float32, device=device)
-        else:
-            # W=256: [16 quadrants][NB][256*128] + dbias [4][NB][256];
-            # NB=64 -> 16*64 = 1024 workgroups = 4/CU
-            st["dw_part"] = torch.zeros(16, 64, 256 * 128, dtype=torch.float32, device=device)
-            st["db_part"] = torch.zeros(4, 64, 256, dtype=torch.float32, device=device)
+        # tier-bridging staging + shared per-tier streams (mixed-width stacks)
+        st["x_in"] = {
+            l: torch.zeros(B, T, p.Wl[l], dtype=bf16, device=device)
+            for l in range(1, Lc)
+            if p.Wl[l] != p.Wl[l - 1]
+        }
+        st["dh_in"] = {
+            l: torch.zeros(B, T, p.Wl[l], dtype=bf16, device=device)
+            for l in range(Lc - 1)
+            if p.Wl[l + 1] != p.Wl[l]
+        }
+        if WH != p.Wl[-1]:
+            st["dh_head_in"] = torch.zeros(B, p.Wl[-1], dtype=bf16, device=device)
+        bwd_tiers = sorted({p.Wl[l] for l in range(Lc)})
+        dx_tiers = sorted({p.Wl[l] for l in range(1, Lc)})
+        st["dgates"] = {
+            t: torch.zeros(B, T, 4 * t, dtype=bf16, device=device) for t in bwd_tiers
+        }
+        # dh/dx stream per tier: a layer's fused-dx output, read as the next
+        # (lower) layer's dh — aliased in place when both share the tier
+        # (the kernel reads dh[gi][t] before writing dx[gi][t] per block)
+        st["dhdx"] = {
+            t: torch.zeros(B, T, t, dtype=bf16, device=device) for t in dx_tiers
+        }
+        st["dw_part"] = {}
+        st["db_part"] = {}
+        for t in bwd_tiers:
+            st["dw_part"][t], st["db_part"][t] = self._dw_buffers(t, device)
         return st
 
     def _fill_perm(self, st):
@@ -283,18 +340,29 @@ class FusedLSTMTrainer:
 
     # ------------------------------------------------------------------
     def _forward(self, st, win: torch.Tensor):
-        """Run the stacked fused forward for window starts ``win`` [B] int32."""
+        """Run the stacked fused forward for window starts ``win`` [B] int32.
+        Each layer runs at its own width tier; tier boundaries are bridged by
+        zero-padded slice copies of the h stream."""
         E, p = self.E, self.p
         B, T = st["B"], st["T"]
-        k_fwd, _, _ = _kernels_for(E, p.W)
         for l in range(p.Lc):
-            src = st["Xd"] if l == 0 else st["h_seq"][l - 1]
+            k_fwd, _, _ = _kernels_for(E, p.Wl[l])
+            if l == 0:
+                src, w = st["Xd"], win
+            else:
+                src = st["h_seq"][l - 1]
+                if p.Wl[l] != p.Wl[l - 1]:
+                    m = min(p.Wl[l], p.Wl[l - 1])
+                    st["x_in"][l][:, :, :m].copy_(src[:, :, :m])
+                    src = st["x_in"][l]
+                w = None
             k_fwd(
-                src, win if l == 0 else None, p.wcat_bf16[l], p.bias_master[l],
+                src, w, p.wcat_bf16[l], p.bias_master[l],
                 st["h_seq"][l], st["gates_seq"][l], st["c_seq"][l],
                 st["x_seq0"] if l == 0 else None, B, T,
             )
-        st["h_last"].copy_(st["h_seq"][p.Lc - 1][:, T - 1, :])
+        mh = min(p.W_head, p.Wl[-1])
+        st["h_last"][:, :mh].copy_(st["h_seq"][p.Lc - 1][:, T - 1, :mh])
 
     def _step(self, st, s: int, opt):
         win = st["perm"][s * st["B"] : (s + 1) * st["B"]]
@@ -311,52 +379,78 @@ class FusedLSTMTrainer:
 
         # ---- head: dense fwd + MSE + bwd (dx_out = dL/dh_{T-1}) ----------
         tgt = win.long() + (T - 1 + st["lookahead"])
-        torch.index_select(st["Yd"], 0, tgt, out=st["y_head"])
+        if p.W_head == st["Yd"].shape[1]:
+            torch.index_select(st["Yd"], 0, tgt, out=st["y_head"])
+        else:
+            m = min(p.W_head, st["Yd"].shape[1])
+            st["y_head"][:, :m].copy_(st["Yd"].index_select(0, tgt)[:, :m])
         E.hg_train_step(
             st["h_last"], st["y_head"], st["counters0"],
             p.head_w, p.head_wt, p.head_bias, st["head_dw"], st["head_db"],
             st["loss_acc"], B, st["F_out"], st["act_out"], st["gscale"], False,
-            st["dx_head"], p.W, 0, 0,
+            st["dx_head"], p.W_head, 0, 0,
         )
 
         # ---- BPTT (fused dx) + split-K dW kernel, top layer down ---------
-        # dh_seq doubles as the dx output: within a block, dh_out[gi][t] is
-        # read (elementwise phase) before dx_seq[gi][t] is written (GEMM
-        # phase of the same t), and blocks own disjoint gi rows.
-        _, k_bwd, k_dw = _kernels_for(E, p.W)
+        # The per-tier dhdx stream doubles as the dx output: within a block,
+        # dh_out[gi][t] is read (elementwise phase) before dx_seq[gi][t] is
+        # written (GEMM phase of the same t), and blocks own disjoint gi
+        # rows. Tier boundaries go through zero-padded dh_in staging copies.
         for l in reversed(range(p.Lc)):
-            dh = st["dx_head"] if l == p.Lc - 1 else st["dh_seq"]
+            Wt = p.Wl[l]
+            _, k_bwd, k_dw = _kernels_for(E, Wt)
+            last_only = l == p.Lc - 1
+            if last_only:
+                if p.W_head == Wt:
+                    dh = st["dx_head"]
+                else:
+                    m = min(p.W_head, Wt)
+                    st["dh_head_in"][:, :m].copy_(st["dx_head"][:, :m])
+                    dh = st["dh_head_in"]
+            else:
+                up = p.Wl[l + 1]
+                if up == Wt:
+                    dh = st["dhdx"][Wt]
+                else:
+                    m = min(up, Wt)
+                    st["dh_in"][l][:, :, :m].copy_(st["dhdx"][up][:, :, :m])
+                    dh = st["dh_in"][l]
+            dx_out = st["dhdx"][Wt] if l > 0 else None
             k_bwd(
                 dh, st["gates_seq"][l], st["c_seq"][l], p.wcat_t_bf16[l],
-                st["dgates"], st["dh_seq"] if l > 0 else None, B, T,
-                l == p.Lc - 1,
+                st["dgates"][Wt], dx_out, B, T, last_only,
             )
-            x_seq = st["x_seq0"] if l == 0 else st["h_seq"][l - 1]
+            if l == 0:
+                x_seq = st["x_seq0"]
+            elif p.Wl[l] != p.Wl[l - 1]:
+                x_seq = st["x_in"][l]
+            else:
+                x_seq = st["h_seq"][l - 1]
             k_dw(
-                st["dgates"], x_seq, st["h_seq"][l], st["dw_part"], st["db_part"],
-                B * T, T,
+                st["dgates"][Wt], x_seq, st["h_seq"][l],
+                st["dw_part"][Wt], st["db_part"][Wt], B * T, T,
             )
-            if p.W == WMAX:
-                torch.sum(st["dw_part"], dim=0, out=st["gw"][l])
-                torch.sum(st["db_part"], dim=0, out=st["gb"][l])
-                p.wcat_master[l].grad = st["gw"][l].view(p.H4, p.KCAT)
+            if Wt == WMAX:
+                torch.sum(st["dw_part"][Wt], dim=0, out=st["gw"][l])
+                torch.sum(st["db_part"][Wt], dim=0, out=st["gb"][l])
+                p.wcat_master[l].grad = st["gw"][l].view(4 * Wt, 2 * Wt)
                 p.bias_master[l].grad = st["gb"][l]
             else:
                 # quadrant tiles [gh][ih] of [256][128] -> gw [4W][2W]
-                ng = p.H4 // 256
-                ni = p.KCAT // 128
-                q = st["dw_part"].sum(1).view(ng, ni, 256, 128)
-                gw = st["gw"][l].view(p.H4, p.KCAT)
+                ng = 4 * Wt // 256
+                ni = 2 * Wt // 128
+                q = st["dw_part"][Wt].sum(1).view(ng, ni, 256, 128)
+                gw = st["gw"][l].view(4 * Wt, 2 * Wt)
                 for gh in range(ng):
                     for ih in range(ni):
                         gw[gh * 256 : (gh + 1) * 256, ih * 128 : (ih + 1) * 128].copy_(q[gh, ih])
-                st["gb"][l].view(ng, 256).copy_(st["db_part"].sum(1))
+                st["gb"][l].view(ng, 256).copy_(st["db_part"][Wt].sum(1))
                 p.wcat_master[l].grad = gw
                 p.bias_master[l].grad = st["gb"][l]
 
         # ---- head grads from the hg slabs (slab layout is [in][out]) -----
-        p.head_wm.grad = st["head_dw"].sum(0).view(p.W, p.W).t().contiguous()
-        p.head_bm.grad = st["head_db"].sum(0).view(p.W)
+        p.head_wm.grad = st["head_dw"].sum(0).view(p.W_head, p.W_head).t().contiguous()
+        p.head_bm.grad = st["head_db"].sum(0).view(p.W_head)
 
 
 class LSTMStepTrainer:
@@ -487,23 +581,39 @@ class FusedLSTMForward:
         outs = []
         bf16 = torch.bfloat16
         B = min(batch_size, n_windows)
-        h_seq = [torch.zeros(B, T, p.W, dtype=bf16, device=self.device) for _ in range(p.Lc)]
-        c_seq = torch.zeros(B, T, p.W, dtype=bf16, device=self.device)
-        gates = torch.zeros(B, T, p.H4, dtype=torch.uint8, device=self.device)
-        k_fwd, _, _ = _kernels_for(self.E, p.W)
+        h_seq = [torch.zeros(B, T, p.Wl[l], dtype=bf16, device=self.device) for l in range(p.Lc)]
+        c_seq = {t: torch.zeros(B, T, t, dtype=bf16, device=self.device) for t in set(p.Wl)}
+        gates = {t: torch.zeros(B, T, 4 * t, dtype=torch.uint8, device=self.device) for t in set(p.Wl)}
+        x_in = {
+            l: torch.zeros(B, T, p.Wl[l], dtype=bf16, device=self.device)
+            for l in range(1, p.Lc)
+            if p.Wl[l] != p.Wl[l - 1]
+        }
         for s in range(0, n_windows, B):
             n = min(B, n_windows - s)
             win = torch.arange(s, s + n, dtype=torch.int32, device=self.device)
             if n < B:
                 win = torch.cat([win, torch.zeros(B - n, dtype=torch.int32, device=self.device)])
             for l in range(p.Lc):
-                src = Xd if l == 0 else h_seq[l - 1]
+                k_fwd, _, _ = _kernels_for(self.E, p.Wl[l])
+                if l == 0:
+                    src, w = Xd, win
+                else:
+                    src = h_seq[l - 1]
+                    if p.Wl[l] != p.Wl[l - 1]:
+                        m = min(p.Wl[l], p.Wl[l - 1])
+                        x_in[l][:, :, :m].copy_(src[:, :, :m])
+                        src = x_in[l]
+                    w = None
                 k_fwd(
-                    src, win if l == 0 else None, p.wcat_bf16[l], p.bias_master[l],
-                    h_seq[l], gates, c_seq, None, B, T,
+                    src, w, p.wcat_bf16[l], p.bias_master[l],
+                    h_seq[l], gates[p.Wl[l]], c_seq[p.Wl[l]], None, B, T,
                 )
-            h_last = h_seq[p.Lc - 1][:, T - 1, :].float()
-            yhat = torch.addmm(p.head_bm, h_last, p.head_wm.t())[:n, :F_out]
+            H_last = self.module.cells[-1].hidden_size
+            h_last = h_seq[p.Lc - 1][:, T - 1, :H_last].float()
+            yhat = torch.addmm(
+                p.head_bm[:F_out], h_last, p.head_wm[:F_out, :H_last].t()
+            )[:n]
             af = self.module.out_func
             if af and af != "linear":
                 yhat = {"tanh": torch.tanh, "relu": torch.relu, "sigmoid": torch.sigmoid}[af](yhat)
