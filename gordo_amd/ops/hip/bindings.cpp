@@ -27,6 +27,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_fwd_wide", &lstm_fwd_wide_launch, "wide (W=128) persistent LSTM forward");
     m.def("lstm_bwd_wide", &lstm_bwd_wide_launch, "wide reverse-time BPTT cell (+fused dx)");
     m.def("lstm_dw_wide", &lstm_dw_wide_launch, "wide split-K dWcat quadrants + dbias");
+    m.def("lstm_fwd_h32", &lstm_fwd_h32_launch, "narrow (W=32) persistent LSTM forward");
+    m.def("lstm_bwd_h32", &lstm_bwd_h32_launch, "narrow reverse-time BPTT cell (+fused dx)");
+    m.def("lstm_dw_h32", &lstm_dw_h32_launch, "narrow split-K dWcat + dbias");
     m.def("lstm_fwd_wide256", &lstm_fwd_wide256_launch, "W=256 persistent LSTM forward (column-half split)");
     m.def("lstm_bwd_wide256", &lstm_bwd_wide256_launch, "W=256 reverse-time BPTT cell (+fused dx)");
     m.def("lstm_dw_wide256", &lstm_dw_wide256_launch, "W=256 split-K dWcat 4x4 quadrants + dbias");

@@ -93,3 +93,17 @@ void lstm_bwd_wide256_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
 void lstm_dw_wide256_launch(torch::Tensor dgates, torch::Tensor x_seq, torch::Tensor h_seq,
                             torch::Tensor partial, torch::Tensor db_partial,
                             int64_t BT, int64_t T);
+
+void lstm_fwd_h32_launch(torch::Tensor X, c10::optional<torch::Tensor> win,
+                         torch::Tensor wcat, torch::Tensor bias, torch::Tensor h_seq,
+                         torch::Tensor gates_seq, torch::Tensor c_seq,
+                         c10::optional<torch::Tensor> x_seq_out, int64_t B, int64_t T);
+
+void lstm_bwd_h32_launch(torch::Tensor dh_out, torch::Tensor gates_seq,
+                         torch::Tensor c_seq, torch::Tensor wcat_t,
+                         torch::Tensor dgates_seq, c10::optional<torch::Tensor> dx_seq,
+                         int64_t B, int64_t T, bool dh_last_only);
+
+void lstm_dw_h32_launch(torch::Tensor dgates, torch::Tensor x_seq, torch::Tensor h_seq,
+                        torch::Tensor partial, torch::Tensor db_partial,
+                        int64_t BT, int64_t T);

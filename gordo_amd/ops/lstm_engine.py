@@ -45,6 +45,7 @@ H4 = 4 * WMAX  # 256 (narrow variant; wide uses per-instance 4*W)
 KCAT = 2 * WMAX  # 128
 WMAX_WIDE = 128
 WMAX_WIDE256 = 256
+WMIN_NARROW = 32  # narrow cell tier (real widths <= 32; headline 25-wide cells)
 
 
 def lstm_width(module) -> int:
@@ -81,6 +82,8 @@ def lstm_supported(module) -> Optional[str]:
 
 def _kernels_for(E, W: int):
     """(fwd, bwd, dw) kernel triple for a width tier."""
+    if W == WMIN_NARROW:
+        return E.lstm_fwd_h32, E.lstm_bwd_h32, E.lstm_dw_h32
     if W == WMAX:
         return E.lstm_fwd, E.lstm_bwd_cell, E.lstm_dw
     if W == WMAX_WIDE:
@@ -90,6 +93,8 @@ def _kernels_for(E, W: int):
 
 def _tier(width: int) -> int:
     """Smallest kernel width tier covering a real width."""
+    if width <= WMIN_NARROW:
+        return WMIN_NARROW
     if width <= WMAX:
         return WMAX
     return WMAX_WIDE if width <= WMAX_WIDE else WMAX_WIDE256
@@ -139,8 +144,9 @@ class _LSTMPackedParams:
             self.bias_master.append(bias)
             self.wcat_bf16.append(wcat.to(torch.bfloat16).contiguous())
             self.wcat_t_bf16.append(wcat.t().contiguous().to(torch.bfloat16))
-        # dense head on h_{T-1}: its own tier over (H_last, F_out)
-        W_head = _tier(max(module.cells[-1].hidden_size, module.n_features_out))
+        # dense head on h_{T-1}: its own tier over (H_last, F_out) — floor
+        # 64 (the hg head kernel's narrowest instantiation)
+        W_head = max(WMAX, _tier(max(module.cells[-1].hidden_size, module.n_features_out)))
         self.W_head = W_head
         self.head_wm = torch.zeros(W_head, W_head, dtype=torch.float32, device=device)
         self.head_bm = torch.zeros(W_head, dtype=torch.float32, device=device)
@@ -252,6 +258,12 @@ class FusedLSTMTrainer:
     def _dw_buffers(tier: int, device):
         """Per-tier dW split-K partial buffers (layers of one tier share)."""
         f32 = torch.float32
+        if tier == WMIN_NARROW:
+            # narrow split-K: 512 flat slabs like W=64
+            return (
+                torch.zeros(512, 4 * tier * 2 * tier, dtype=f32, device=device),
+                torch.zeros(512, 4 * tier, dtype=f32, device=device),
+            )
         if tier == WMAX:
             # split-K dW partials: 512 blocks = 2 blocks/CU (latency hiding)
             return (
@@ -430,7 +442,7 @@ class FusedLSTMTrainer:
                 st["dgates"][Wt], x_seq, st["h_seq"][l],
                 st["dw_part"][Wt], st["db_part"][Wt], B * T, T,
             )
-            if Wt == WMAX:
+            if Wt in (WMIN_NARROW, WMAX):
                 torch.sum(st["dw_part"][Wt], dim=0, out=st["gw"][l])
                 torch.sum(st["db_part"][Wt], dim=0, out=st["gb"][l])
                 p.wcat_master[l].grad = st["gw"][l].view(4 * Wt, 2 * Wt)

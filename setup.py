@@ -19,6 +19,7 @@ try:
         "gordo_amd/ops/hip/bindings.cpp",
         "gordo_amd/ops/hip/hourglass_kernels.hip",
         "gordo_amd/ops/hip/lstm_kernels.hip",
+        "gordo_amd/ops/hip/lstm_h32_kernels.hip",
         "gordo_amd/ops/hip/lstm_wide_kernels.hip",
         "gordo_amd/ops/hip/lstm_wide256_kernels.hip",
     ]

@@ -74,7 +74,7 @@ def test_lstm_grads_match_autograd():
     tr.p = p
     from gordo_amd.ops.hourglass_engine import pad_rows_bf16
 
-    Xd = pad_rows_bf16(X, DEV)
+    Xd = pad_rows_bf16(X, DEV, p.W)
     st = tr._alloc_step_state(Xd, Xd, B, T, 1, B, 0, F)
     win = torch.arange(B, dtype=torch.int32, device=DEV)
     tr._compute_grads(st, win)
@@ -94,6 +94,7 @@ def test_lstm_grads_match_autograd():
 
     cell = ref_net.cells[0]
     H, I = cell.hidden_size, cell.input_size
+    W = p.Wl[0]  # per-layer tier (r2): a 20->32 cell runs the W=32 kernels
     gw = p.wcat_master[0].grad.cpu()
     gb = p.bias_master[0].grad.cpu()
     for g in range(4):
@@ -102,9 +103,9 @@ def test_lstm_grads_match_autograd():
         # torch's two bias vectors get IDENTICAL grads (both add to the same
         # pre-activation); the fused combined bias's grad equals either one.
         ref_b = cell.bias_ih_l0.grad[g * H : (g + 1) * H]
-        got_ih = gw[g * WMAX : g * WMAX + H, :I]
-        got_hh = gw[g * WMAX : g * WMAX + H, WMAX : WMAX + H]
-        got_b = gb[g * WMAX : g * WMAX + H]
+        got_ih = gw[g * W : g * W + H, :I]
+        got_hh = gw[g * W : g * W + H, W : W + H]
+        got_b = gb[g * W : g * W + H]
         s = ref_ih.abs().max().item() or 1.0
         assert (got_ih - ref_ih).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} dW_ih"
         s = ref_hh.abs().max().item() or 1.0
@@ -112,7 +113,8 @@ def test_lstm_grads_match_autograd():
         s = ref_b.abs().max().item() or 1.0
         assert (got_b - ref_b).abs().max().item() < 0.08 * s + 2e-4, f"gate {g} db"
         # padded rows/cols carry zero grad
-        assert gw[g * WMAX + H : (g + 1) * WMAX].abs().max().item() == 0.0
+        if H < W:
+            assert gw[g * W + H : (g + 1) * W].abs().max().item() == 0.0
     # head grads
     ref_hw = ref_net.head.weight.grad
     got_hw = p.head_wm.grad.cpu()[: ref_hw.shape[0], : ref_hw.shape[1]]
@@ -524,3 +526,104 @@ def test_mixed_tier_forward_matches_torch():
     out = FusedLSTMForward(net, DEV).predict(X)
     assert out.shape == ref.shape
     np.testing.assert_allclose(out, ref, atol=6e-2, rtol=6e-2)
+
+
+# ---------------------------------------------------------------------------
+# W=32 narrow tier (real widths <= 32): register-resident weights at half the
+# lane width; the headline 50-tag stack's 25->25 cell runs here.
+# ---------------------------------------------------------------------------
+def test_h32_lstm_fused_forward_matches_torch():
+    from gordo_amd.ops.lstm_engine import _tier
+
+    T = 16
+    net = make_lstm(n_features=25, dims=(25, 20), T=T, seed=30)
+    assert [_tier(max(c.input_size, c.hidden_size)) for c in net.cells] == [32, 32]
+    X = np.random.default_rng(30).normal(size=(300, 25)).astype(np.float32)
+    ref = ref_forward(net, X, T)
+    out = FusedLSTMForward(net, DEV).predict(X)
+    assert out.shape == ref.shape
+    np.testing.assert_allclose(out, ref, atol=4e-2, rtol=4e-2)
+
+
+def test_h32_lstm_grads_match_autograd():
+    """Full-narrow stack: 25->25->20, all cells at W=32 (head at 64)."""
+    from gordo_amd.ops.hourglass_engine import pad_rows_bf16
+
+    T, B, F = 12, 256, 25
+    net = make_lstm(n_features=F, dims=(25, 20), T=T, seed=31)
+    ref_net = make_lstm(n_features=F, dims=(25, 20), T=T, seed=31)
+    ref_net.load_state_dict(net.state_dict())
+    X = np.random.default_rng(31).normal(size=(B + T + 2, F)).astype(np.float32)
+
+    cfg = TrainConfig(epochs=1, batch_size=B)
+    tr = FusedLSTMTrainer(net, DEV, cfg)
+    pp = _LSTMPackedParams(net, DEV)
+    assert pp.Wl == [32, 32] and pp.W_head == 64
+    tr.p = pp
+    Xd = pad_rows_bf16(X, DEV, pp.W)
+    st = tr._alloc_step_state(Xd, Xd, B, T, 1, B, 0, F)
+    win = torch.arange(B, dtype=torch.int32, device=DEV)
+    tr._compute_grads(st, win)
+    torch.cuda.synchronize()
+
+    Xb = torch.as_tensor(X).to(torch.bfloat16).float()
+    wins = torch.stack([Xb[i : i + T] for i in range(B)])
+    tgt = torch.stack([Xb[i + T - 1] for i in range(B)])
+    loss = torch.nn.functional.mse_loss(ref_net(wins), tgt)
+    loss.backward()
+
+    kern_loss = float(st["loss_acc"].item()) / (B * F)
+    assert abs(kern_loss - float(loss.detach())) / max(float(loss), 1e-9) < 6e-2
+
+    for l, cell in enumerate(ref_net.cells):
+        H, I = cell.hidden_size, cell.input_size
+        W = pp.Wl[l]
+        gw = pp.wcat_master[l].grad.cpu()
+        gb = pp.bias_master[l].grad.cpu()
+        for g in range(4):
+            ref_ih = cell.weight_ih_l0.grad[g * H : (g + 1) * H]
+            ref_hh = cell.weight_hh_l0.grad[g * H : (g + 1) * H]
+            ref_b = cell.bias_ih_l0.grad[g * H : (g + 1) * H]
+            got_ih = gw[g * W : g * W + H, :I]
+            got_hh = gw[g * W : g * W + H, W : W + H]
+            got_b = gb[g * W : g * W + H]
+            s = ref_ih.abs().max().item() or 1.0
+            assert (got_ih - ref_ih).abs().max().item() < 0.08 * s + 2e-4, f"l{l} g{g} dW_ih"
+            s = ref_hh.abs().max().item() or 1.0
+            assert (got_hh - ref_hh).abs().max().item() < 0.08 * s + 2e-4, f"l{l} g{g} dW_hh"
+            s = ref_b.abs().max().item() or 1.0
+            assert (got_b - ref_b).abs().max().item() < 0.08 * s + 2e-4, f"l{l} g{g} db"
+            if H < W:
+                assert gw[g * W + H : (g + 1) * W].abs().max().item() == 0.0
+    ref_hw = ref_net.head.weight.grad
+    got_hw = pp.head_wm.grad.cpu()[: ref_hw.shape[0], : ref_hw.shape[1]]
+    s = ref_hw.abs().max().item() or 1.0
+    assert (got_hw - ref_hw).abs().max().item() < 0.08 * s + 2e-4, "head dW"
+
+
+def test_headline_stack_uses_narrow_tier_and_learns():
+    """The BASELINE config #3 stack (50 tags, dims 38-25-25-38) now runs its
+    25->25 cell at W=32; the estimator must train fused and learn."""
+    from gordo_amd.model.models import TorchLSTMAutoEncoder
+    from gordo_amd.model.factories.lstm_autoencoder import lstm_hourglass
+    from gordo_amd.ops.lstm_engine import _tier
+
+    m = lstm_hourglass(50, 50, lookback_window=16, encoding_layers=2, compression_factor=0.5)
+    assert [_tier(max(c.input_size, c.hidden_size)) for c in m.cells] == [64, 64, 32, 64]
+
+    rng = np.random.default_rng(32)
+    t = np.arange(3000)
+    X = np.stack(
+        [np.sin(t / (8 + i % 9)) + 0.05 * rng.standard_normal(len(t)) for i in range(50)],
+        axis=1,
+    ).astype(np.float32)
+    model = TorchLSTMAutoEncoder(
+        kind="lstm_hourglass", lookback_window=16, encoding_layers=2,
+        compression_factor=0.5, epochs=4, batch_size=1024, lr=3e-3,
+    )
+    model.fit(X)
+    assert model.history["engine"] == "fused_hip_lstm", model.history
+    assert model.history["loss"][-1] < model.history["loss"][0]
+    out = model.predict(X[:200])
+    assert out.shape == (200 - 16 + 1, 50)
+    assert np.isfinite(out).all()
