@@ -19,9 +19,10 @@ model/utils.py::create_keras_timeseriesgenerator). Here each training step is:
                       single-tile, K=B*T shape)
   optimizer:          fp32-master Adam (torch foreach) + bf16 weight re-emit
 
-Weight layout: per layer one packed [4H=256][128] bf16 "wcat" (cols 0..63 =
-W_ih zero-padded, 64..127 = W_hh; rows gate-major i,f,g,o each padded to 64 —
-same gate order as torch.nn.LSTM) plus a fused fp32 bias b_ih+b_hh [256].
+Weight layout: per layer one packed [4W][2W] bf16 "wcat" at the LAYER's own
+width tier W in {32, 64, 128, 256} (cols :W = W_ih zero-padded, W: = W_hh;
+rows gate-major i,f,g,o each padded to W — same gate order as
+torch.nn.LSTM) plus a fused fp32 bias b_ih+b_hh [4W].
 Zero padding is gradient-invariant: padded h stays exactly 0 through the cell
 (weights/bias 0 -> g=tanh(0)=0 -> c=0 -> h=0) and padded grads are 0.
 """
@@ -49,8 +50,9 @@ WMIN_NARROW = 32  # narrow cell tier (real widths <= 32; headline 25-wide cells)
 
 
 def lstm_width(module) -> int:
-    """Padded kernel width tier: 64 (chip-resident weights), 128 (wide,
-    L2-streamed weights) or 256 (column-half split cells)."""
+    """WIDEST padded kernel tier across the stack (64/128/256 — the legacy
+    whole-stack answer; since r2 each layer actually runs at its own tier,
+    including the W=32 narrow one — see _LSTMPackedParams.Wl)."""
     dims = [module.n_features, module.n_features_out]
     for cell in module.cells:
         dims += [cell.hidden_size, cell.input_size]
