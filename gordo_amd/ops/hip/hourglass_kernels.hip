@@ -8,7 +8,7 @@
 // Per training step:
 //   1. hg_train_step<W>: minibatch gather + L-layer dense forward (MFMA bf16
 //      16x16x32, activations in XOR-swizzled LDS tiles) + MSE loss grad +
-//      full backward (dW/db partials per block slab in fp32)
+//      full backward (dW/db partials per block slab in bf16)
 //   2. hg_slab_reduce<W>: stage-A float4 slab reduction (slab layout),
 //      counter bump folded in
 //   3. hg_adam_p<W>: stage-B Adam over G partials (transpose-on-read),
@@ -33,8 +33,11 @@ struct HgTrainArgs {
     const unsigned short* __restrict__ w;   // [L][W][W] bf16 (out,in)
     const unsigned short* __restrict__ wt;  // [L][W][W] bf16 (in,out)
     const float* __restrict__ bias;         // [L][W] fp32
-    float* __restrict__ dw;                 // [n_tiles][L][W*W] fp32 slabs
-    float* __restrict__ db;                 // [n_tiles][L][W] fp32 slabs
+    unsigned short* __restrict__ dw;        // [n_tiles][L][W*W] bf16 slabs
+    unsigned short* __restrict__ db;        // [n_tiles][L][W] bf16 slabs
+                                            // (bf16 halves the dominant
+                                            // HBM stream: 336 MB/step at
+                                            // W=256; reduce accumulates fp32)
     float* __restrict__ loss;               // [1] fp32 accumulator (SSE)
     unsigned short* __restrict__ dx_out;    // [B][W] bf16 dL/dinput, or null
     int B;          // rows per step (per GPU)
@@ -259,7 +262,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
 
         // ---- dW: wave stripes the in dim (W/4 rows), K = 32, 1 K-step ----
         {
-            float* dwl = a.dw + ((size_t)tile * a.L + li) * W * W;
+            unsigned short* dwl = a.dw + ((size_t)tile * a.L + li) * W * W;
 #pragma unroll
             for (int sub = 0; sub < W / 64; ++sub) {
                 int in_base = wid * (W / 4) + sub * 16;
@@ -286,7 +289,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
                         int in = in_base + (lane >> 4) * 4 + r;
-                        dwl[(size_t)in * W + out] = acc[nt][r];
+                        dwl[(size_t)in * W + out] = ga_f32_to_bf16(acc[nt][r]);
                     }
                 }
             }
@@ -310,7 +313,7 @@ __global__ void hg_train_step(HgTrainArgs a) {
                 float tot = 0.f;
 #pragma unroll
                 for (int gg = 0; gg < GR; ++gg) tot += s_red[gg * W + tid];
-                a.db[((size_t)tile * a.L + li) * W + tid] = tot;
+                a.db[((size_t)tile * a.L + li) * W + tid] = ga_f32_to_bf16(tot);
             }
         }
         __syncthreads();
@@ -331,8 +334,8 @@ __global__ void hg_adam(
     float* __restrict__ master,      // [P] fp32 (weights then biases)
     float* __restrict__ m_buf,
     float* __restrict__ v_buf,
-    const float* __restrict__ dw,    // [n_slabs][L][W*W]
-    const float* __restrict__ db,    // [n_slabs][L][W]
+    const unsigned short* __restrict__ dw,  // [n_slabs][L][W*W] bf16
+    const unsigned short* __restrict__ db,  // [n_slabs][L][W] bf16
     unsigned short* __restrict__ w_bf16,   // [L][W][W]
     unsigned short* __restrict__ wt_bf16,  // [L][W][W]
     float* __restrict__ bias,        // [L][W] (fp32 view used by fwd)
@@ -351,10 +354,10 @@ __global__ void hg_adam(
         int rem = idx - l * W * W;
         int out = rem / W, in = rem - out * W;
         size_t sidx = (size_t)l * W * W + (size_t)in * W + out;
-        for (int s = 0; s < n_slabs; ++s) g += dw[(size_t)s * PW + sidx];
+        for (int s = 0; s < n_slabs; ++s) g += ga_bf16_to_f32(dw[(size_t)s * PW + sidx]);
     } else {
         int bidx = idx - PW;
-        for (int s = 0; s < n_slabs; ++s) g += db[(size_t)s * (L * W) + bidx];
+        for (int s = 0; s < n_slabs; ++s) g += ga_bf16_to_f32(db[(size_t)s * (L * W) + bidx]);
     }
 
     int t = counters[1];
@@ -398,41 +401,46 @@ __global__ void hg_bump(int* counters, int n_batches) {
 // ===========================================================================
 template <int W>
 __launch_bounds__(256)
-__global__ void hg_slab_reduce(const float* __restrict__ dw,
-                               const float* __restrict__ db,
+__global__ void hg_slab_reduce(const unsigned short* __restrict__ dw,
+                               const unsigned short* __restrict__ db,
                                float* __restrict__ partial,  // [G][P] SLAB layout
                                int* __restrict__ counters,
                                int n_slabs, int G, int L, int n_batches) {
     const int PW = L * W * W;
     const int P = PW + L * W;
     int g = blockIdx.y;
-    int idx4 = blockIdx.x * blockDim.x + threadIdx.x;
-    if (g == 0 && idx4 == 0 && counters != nullptr) {
+    int idx8 = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g == 0 && idx8 == 0 && counters != nullptr) {
         int nb = (counters[0] + 1) % n_batches;
         counters[0] = nb;
         counters[1] += 1;
         if (nb == 0) counters[2] += 1;  // new epoch -> new shuffle key
     }
-    int idx = idx4 * 4;
+    int idx = idx8 * 8;  // bf16x8 = 16-B loads, fp32 accumulation
     if (idx >= P) return;
     int per = (n_slabs + G - 1) / G;
     int s0 = g * per, s1 = min(n_slabs, s0 + per);
-    float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
     if (idx < PW) {
 #pragma unroll 4
         for (int sl = s0; sl < s1; ++sl) {
-            float4 v = *(const float4*)(dw + (size_t)sl * PW + idx);
-            acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+            ga_u16x8 v = *(const ga_u16x8*)(dw + (size_t)sl * PW + idx);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) acc[e] += ga_bf16_to_f32(v[e]);
         }
     } else {
         int bidx = idx - PW;
 #pragma unroll 4
         for (int sl = s0; sl < s1; ++sl) {
-            float4 v = *(const float4*)(db + (size_t)sl * (L * W) + bidx);
-            acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+            ga_u16x8 v = *(const ga_u16x8*)(db + (size_t)sl * (L * W) + bidx);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) acc[e] += ga_bf16_to_f32(v[e]);
         }
     }
-    *(float4*)(partial + (size_t)g * P + idx) = acc;
+    *(float4*)(partial + (size_t)g * P + idx) = make_float4(acc[0], acc[1], acc[2], acc[3]);
+    *(float4*)(partial + (size_t)g * P + idx + 4) = make_float4(acc[4], acc[5], acc[6], acc[7]);
 }
 
 template <int W>
@@ -615,8 +623,8 @@ __global__ void hg_forward(HgFwdArgs a) {
 // ===========================================================================
 template <int W>
 __launch_bounds__(256)
-__global__ void hg_grad_reduce(const float* __restrict__ dw,
-                               const float* __restrict__ db,
+__global__ void hg_grad_reduce(const unsigned short* __restrict__ dw,
+                               const unsigned short* __restrict__ db,
                                float* __restrict__ g, int n_slabs, int L) {
     const int PW = L * W * W;
     const int P = PW + L * W;
@@ -628,10 +636,10 @@ __global__ void hg_grad_reduce(const float* __restrict__ dw,
         int rem = idx - l * W * W;
         int out = rem / W, in = rem - out * W;
         size_t sidx = (size_t)l * W * W + (size_t)in * W + out;
-        for (int sl = 0; sl < n_slabs; ++sl) s += dw[(size_t)sl * PW + sidx];
+        for (int sl = 0; sl < n_slabs; ++sl) s += ga_bf16_to_f32(dw[(size_t)sl * PW + sidx]);
     } else {
         int bidx = idx - PW;
-        for (int sl = 0; sl < n_slabs; ++sl) s += db[(size_t)sl * (L * W) + bidx];
+        for (int sl = 0; sl < n_slabs; ++sl) s += ga_bf16_to_f32(db[(size_t)sl * (L * W) + bidx]);
     }
     g[idx] = s;
 }
@@ -727,8 +735,10 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
     a.w = (const unsigned short*)w.data_ptr();
     a.wt = (const unsigned short*)wt.data_ptr();
     a.bias = bias.data_ptr<float>();
-    a.dw = dw.data_ptr<float>();
-    a.db = db.data_ptr<float>();
+    TORCH_CHECK(dw.scalar_type() == torch::kBFloat16, "dw slabs must be bf16");
+    TORCH_CHECK(db.scalar_type() == torch::kBFloat16, "db slabs must be bf16");
+    a.dw = (unsigned short*)dw.data_ptr();
+    a.db = (unsigned short*)db.data_ptr();
     a.loss = loss.data_ptr<float>();
     a.dx_out = dx_out.has_value() ? (unsigned short*)dx_out->data_ptr() : nullptr;
     a.B = (int)B;
@@ -780,7 +790,7 @@ void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
     HG_DISPATCH_W(wmax, {
         hipLaunchKernelGGL(hg_adam<W>, dim3(blocks), dim3(256), 0, stream.stream(),
                            master.data_ptr<float>(), m.data_ptr<float>(), v.data_ptr<float>(),
-                           dw.data_ptr<float>(), db.data_ptr<float>(),
+                           (const unsigned short*)dw.data_ptr(), (const unsigned short*)db.data_ptr(),
                            (unsigned short*)w.data_ptr(), (unsigned short*)wt.data_ptr(),
                            bias.data_ptr<float>(), counters.data_ptr<int>(), (int)n_slabs,
                            (int)L, (float)lr, (float)beta1, (float)beta2, (float)eps);
@@ -846,11 +856,14 @@ void hg_slab_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor par
                            c10::optional<torch::Tensor> counters, int64_t n_slabs,
                            int64_t G, int64_t L, int64_t n_batches, int64_t wmax) {
     int P = (int)(L * wmax * wmax + L * wmax);
-    dim3 grid((P / 4 + 255) / 256, (int)G);
+    TORCH_CHECK(dw.scalar_type() == torch::kBFloat16 && db.scalar_type() == torch::kBFloat16,
+                "slab buffers must be bf16");
+    dim3 grid((P / 8 + 255) / 256, (int)G);
     auto stream = at::hip::getCurrentHIPStream();
     HG_DISPATCH_W(wmax, {
         hipLaunchKernelGGL(hg_slab_reduce<W>, grid, dim3(256), 0, stream.stream(),
-                           dw.data_ptr<float>(), db.data_ptr<float>(),
+                           (const unsigned short*)dw.data_ptr(),
+                           (const unsigned short*)db.data_ptr(),
                            partial.data_ptr<float>(),
                            counters.has_value() ? counters->data_ptr<int>() : nullptr,
                            (int)n_slabs, (int)G, (int)L, (int)n_batches);
@@ -893,7 +906,8 @@ void hg_grad_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,
     auto stream = at::hip::getCurrentHIPStream();
     HG_DISPATCH_W(wmax, {
         hipLaunchKernelGGL(hg_grad_reduce<W>, dim3(blocks), dim3(256), 0, stream.stream(),
-                           dw.data_ptr<float>(), db.data_ptr<float>(), g.data_ptr<float>(),
+                           (const unsigned short*)dw.data_ptr(),
+                           (const unsigned short*)db.data_ptr(), g.data_ptr<float>(),
                            (int)n_slabs, (int)L);
     });
 }

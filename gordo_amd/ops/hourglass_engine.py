@@ -151,8 +151,10 @@ class FusedHourglassTrainer:
         n_tiles = (B + MT - 1) // MT
         n_batches = max(1, math.ceil(N / B))
 
-        dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=device)
-        db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=device)
+        # bf16 slabs: halves the dominant dW-slab HBM stream (fp32 masters
+        # still accumulate exactly; per-tile partials round at ~bf16 eps)
+        dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.bfloat16, device=device)
+        db = torch.zeros(n_tiles, L, WM, dtype=torch.bfloat16, device=device)
         loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
         # counters [batch idx, adam t, epoch]: hg_slab_reduce pre-bumps
         counters = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)

@@ -309,8 +309,8 @@ class FusedLSTMTrainer:
             "y_head": torch.zeros(B, WH, dtype=bf16, device=device),
             "dx_head": torch.zeros(B, WH, dtype=bf16, device=device),
             "counters0": torch.tensor([0, 1, 0], dtype=torch.int32, device=device),
-            "head_dw": torch.zeros(n_head_tiles, 1, WH * WH, dtype=torch.float32, device=device),
-            "head_db": torch.zeros(n_head_tiles, 1, WH, dtype=torch.float32, device=device),
+            "head_dw": torch.zeros(n_head_tiles, 1, WH * WH, dtype=torch.bfloat16, device=device),
+            "head_db": torch.zeros(n_head_tiles, 1, WH, dtype=torch.bfloat16, device=device),
             "loss_acc": torch.zeros(1, dtype=torch.float32, device=device),
             "gscale": 2.0 / (B * F_out),
             "act_out": [ACT_IDS[self.module.out_func]],
@@ -463,8 +463,10 @@ class FusedLSTMTrainer:
                 p.bias_master[l].grad = st["gb"][l]
 
         # ---- head grads from the hg slabs (slab layout is [in][out]) -----
-        p.head_wm.grad = st["head_dw"].sum(0).view(p.W_head, p.W_head).t().contiguous()
-        p.head_bm.grad = st["head_db"].sum(0).view(p.W_head)
+        p.head_wm.grad = (
+            st["head_dw"].sum(0, dtype=torch.float32).view(p.W_head, p.W_head).t().contiguous()
+        )
+        p.head_bm.grad = st["head_db"].sum(0, dtype=torch.float32).view(p.W_head)
 
 
 class LSTMStepTrainer:

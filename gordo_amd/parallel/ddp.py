@@ -126,8 +126,8 @@ class HourglassStepTrainer:
         self.N = N
         MT = int(getattr(self.E, "MT", 32))
         self.n_tiles = (self.B + MT - 1) // MT
-        self.dw = torch.zeros(self.n_tiles, p.L, WM * WM, dtype=torch.float32, device=device)
-        self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.float32, device=device)
+        self.dw = torch.zeros(self.n_tiles, p.L, WM * WM, dtype=torch.bfloat16, device=device)
+        self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.bfloat16, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
         # counters [batch idx, adam t, epoch]: hg_slab_reduce pre-bumps

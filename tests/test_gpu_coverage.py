@@ -115,8 +115,8 @@ def test_fused_training_wide_matches_autograd():
     MT = int(E.MT)
     n_tiles = (B + MT - 1) // MT
     Xd = pad_rows_bf16(X, DEV, WM)
-    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
-    db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.bfloat16, device=DEV)
+    db = torch.zeros(n_tiles, L, WM, dtype=torch.bfloat16, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
     counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
@@ -132,8 +132,8 @@ def test_fused_training_wide_matches_autograd():
     assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
-    dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
-    db_sum = db.sum(dim=0).view(L, WM).cpu()
+    dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WM, WM).cpu()
+    db_sum = db.sum(dim=0, dtype=torch.float32).view(L, WM).cpu()
     for i, lin in enumerate(linears):
         o, ninp = lin.weight.shape
         got_w = dw_sum[i, :ninp, :o].T  # kernel stores [in][out]
@@ -181,8 +181,8 @@ def test_fused_training_y_not_x():
     n_tiles = (B + MT - 1) // MT
     Xd = pad_rows_bf16(X, DEV)
     Yd = pad_rows_bf16(Y, DEV)
-    dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=DEV)
-    db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=DEV)
+    dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.bfloat16, device=DEV)
+    db = torch.zeros(n_tiles, L, WMAX, dtype=torch.bfloat16, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
     counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
@@ -198,7 +198,7 @@ def test_fused_training_y_not_x():
     kernel_loss = float(loss.item()) / (B * F_out)
     assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
     lin0 = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)][0]
-    got = dw.sum(0).view(L, WMAX, WMAX)[0, :30, :20].T.cpu()
+    got = dw.sum(0, dtype=torch.float32).view(L, WMAX, WMAX)[0, :30, :20].T.cpu()
     ref = lin0.weight.grad.detach()
     s = ref.abs().max().item() or 1.0
     assert (got - ref).abs().max().item() < 0.05 * s + 1e-4
@@ -374,8 +374,8 @@ def test_fused_training_x256_matches_autograd():
     MT = int(E.MT)
     n_tiles = (B + MT - 1) // MT
     Xd = pad_rows_bf16(X, DEV, WM)
-    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
-    db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.bfloat16, device=DEV)
+    db = torch.zeros(n_tiles, L, WM, dtype=torch.bfloat16, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
     counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
@@ -391,8 +391,8 @@ def test_fused_training_x256_matches_autograd():
     assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
-    dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
-    db_sum = db.sum(dim=0).view(L, WM).cpu()
+    dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WM, WM).cpu()
+    db_sum = db.sum(dim=0, dtype=torch.float32).view(L, WM).cpu()
     for i, lin in enumerate(linears):
         o, ninp = lin.weight.shape
         got_w = dw_sum[i, :ninp, :o].T  # kernel stores [in][out]
@@ -489,8 +489,8 @@ def test_fused_training_grads_fuzz():
         n_tiles = (B + MT - 1) // MT
         Xd = pad_rows_bf16(X, DEV, WM)
         Yd = Xd if y_is_x else pad_rows_bf16(Y, DEV, WM)
-        dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
-        db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+        dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.bfloat16, device=DEV)
+        db = torch.zeros(n_tiles, L, WM, dtype=torch.bfloat16, device=DEV)
         loss = torch.zeros(1, dtype=torch.float32, device=DEV)
         counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
         gscale = 2.0 / (B * F_out)
@@ -508,8 +508,8 @@ def test_fused_training_grads_fuzz():
         assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 6e-2, msg
 
         linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
-        dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
-        db_sum = db.sum(dim=0).view(L, WM).cpu()
+        dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WM, WM).cpu()
+        db_sum = db.sum(dim=0, dtype=torch.float32).view(L, WM).cpu()
         for i, lin in enumerate(linears):
             o, ninp = lin.weight.shape
             got_w = dw_sum[i, :ninp, :o].T
@@ -579,8 +579,8 @@ def test_default_hourglass_l7_fused_grads_match_autograd():
     MT = int(E.MT)
     n_tiles = (B + MT - 1) // MT
     Xd = pad_rows_bf16(X, DEV, WM)
-    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.float32, device=DEV)
-    db = torch.zeros(n_tiles, L, WM, dtype=torch.float32, device=DEV)
+    dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.bfloat16, device=DEV)
+    db = torch.zeros(n_tiles, L, WM, dtype=torch.bfloat16, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
     counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
@@ -597,8 +597,8 @@ def test_default_hourglass_l7_fused_grads_match_autograd():
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
     assert len(linears) == 7
-    dw_sum = dw.sum(dim=0).view(L, WM, WM).cpu()
-    db_sum = db.sum(dim=0).view(L, WM).cpu()
+    dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WM, WM).cpu()
+    db_sum = db.sum(dim=0, dtype=torch.float32).view(L, WM).cpu()
     for i, lin in enumerate(linears):
         o, ninp = lin.weight.shape
         got_w = dw_sum[i, :ninp, :o].T

@@ -102,8 +102,8 @@ def test_train_step_gradients_match_autograd():
     B = N
     MT = int(E.MT)
     n_tiles = (B + MT - 1) // MT
-    dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.float32, device=DEV)
-    db = torch.zeros(n_tiles, L, WMAX, dtype=torch.float32, device=DEV)
+    dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.bfloat16, device=DEV)
+    db = torch.zeros(n_tiles, L, WMAX, dtype=torch.bfloat16, device=DEV)
     loss = torch.zeros(1, dtype=torch.float32, device=DEV)
     counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
     gscale = 2.0 / (B * F_out)
@@ -122,8 +122,8 @@ def test_train_step_gradients_match_autograd():
     assert abs(kernel_loss - float(loss_ref)) / max(float(loss_ref), 1e-9) < 5e-2
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
-    dw_sum = dw.sum(dim=0).view(L, WMAX, WMAX).cpu()
-    db_sum = db.sum(dim=0).view(L, WMAX).cpu()
+    dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WMAX, WMAX).cpu()
+    db_sum = db.sum(dim=0, dtype=torch.float32).view(L, WMAX).cpu()
     for i, lin in enumerate(linears):
         o, ninp = lin.weight.shape
         got_w = dw_sum[i, :ninp, :o].T  # kernel stores [in][out]
@@ -150,8 +150,8 @@ def test_train_step_gradients_match_autograd():
     # manual reference on the kernel's own summed grads (layout-adjusted)
     g = torch.zeros_like(master_before)
     PW = L * WMAX * WMAX
-    g[:PW] = dw.sum(dim=0).view(L, WMAX, WMAX).transpose(1, 2).reshape(-1)
-    g[PW:] = db.sum(dim=0).reshape(-1)
+    g[:PW] = dw.sum(dim=0, dtype=torch.float32).view(L, WMAX, WMAX).transpose(1, 2).reshape(-1)
+    g[PW:] = db.sum(dim=0, dtype=torch.float32).reshape(-1)
     m_ref = (1 - b1) * g
     v_ref = (1 - b2) * g * g
     upd = master_before - lr * (m_ref / (1 - b1)) / ((v_ref / (1 - b2)).sqrt() + eps)
