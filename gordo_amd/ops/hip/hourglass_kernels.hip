@@ -493,19 +493,21 @@ template <int W>
 __launch_bounds__(256)
 __global__ void hg_partial_flat(const float* __restrict__ partial,
                                 float* __restrict__ g_out, int G, int L) {
-    // partial is SLAB layout; emit MASTER layout for the DP bucket
+    // partial is SLAB layout; emit MASTER layout for the DP bucket.
+    // Iterate in slab order so the G reads per element coalesce; only the
+    // single g_out write scatters (same flip as hg_adam_p, +5% headline).
     const int PW = L * W * W;
     const int P = PW + L * W;
-    int idx = blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= P) return;
-    size_t sidx;
-    if (idx < PW) {
-        int l = idx / (W * W);
-        int rem = idx - l * W * W;
-        int out = rem / W, in = rem - out * W;
-        sidx = (size_t)l * W * W + (size_t)in * W + out;
+    int sidx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (sidx >= P) return;
+    size_t idx;
+    if (sidx < PW) {
+        int l = sidx / (W * W);
+        int rem = sidx - l * W * W;
+        int in = rem / W, out = rem - in * W;
+        idx = (size_t)l * W * W + (size_t)out * W + in;
     } else {
-        sidx = (size_t)idx;
+        idx = (size_t)sidx;
     }
     float sum = 0.f;
 #pragma unroll 4
