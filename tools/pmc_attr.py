@@ -18,7 +18,7 @@ def agg(path):
 
 def main():
     for arg in sys.argv[1:]:
-        label, _, path = arg.partition("=")
+        path_sep = arg.rpartition("="); label, path = path_sep[0], path_sep[2]
         print("##", label)
         for k, c in sorted(agg(path).items()):
             wc = c.get("SQ_WAVE_CYCLES", 1)
