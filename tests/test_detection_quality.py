@@ -24,8 +24,7 @@ def _normal_data(n, noise_seed=0, t0=0):
     return X.astype(np.float32)
 
 
-@pytest.mark.timeout(300)
-def test_autoencoder_detects_injected_faults():
+def _run_quality(require_fused: bool = False):
     from sklearn.pipeline import Pipeline
     from sklearn.preprocessing import MinMaxScaler
 
@@ -48,6 +47,11 @@ def test_autoencoder_detects_injected_faults():
     det.cross_validate(X=X_train, y=X_train)
     det.fit(X_train)
     assert det.aggregate_threshold_ is not None
+    if require_fused:
+        # on a GPU box the SAME statistical bar must hold with training run
+        # by the fused bf16 HIP engine, not the eager fp32 fallback
+        engine = det.base_estimator.steps[-1][1].history["engine"]
+        assert engine.startswith("fused_hip"), engine
 
     # test stream: normal continuation + faults injected at known rows
     X_test = _normal_data(600, noise_seed=2, t0=4000)
@@ -82,3 +86,20 @@ def test_autoencoder_detects_injected_faults():
     for r, c in zip(fault_rows, faulty_tags):
         window = tag_scores[r : r + 5]
         assert (window.argmax(axis=1) == c).mean() >= 0.8, (r, c, window.argmax(axis=1))
+
+
+@pytest.mark.timeout(300)
+def test_autoencoder_detects_injected_faults():
+    _run_quality()
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_autoencoder_detects_injected_faults_fused_gpu():
+    """Same statistical quality bar, but trained by the fused bf16 HIP
+    engine on MI355X — proves the kernel numerics don't degrade detection."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs MI355X")
+    _run_quality(require_fused=True)
