@@ -129,7 +129,8 @@ def test_fused_training_wide_matches_autograd():
     loss_ref = torch.nn.functional.mse_loss(out, Xref)
     loss_ref.backward()
     kernel_loss = float(loss.item()) / (B * F_out)
-    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+    ref_val = float(loss_ref.detach())
+    assert abs(kernel_loss - ref_val) / max(ref_val, 1e-9) < 5e-2
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
     dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WM, WM).cpu()
@@ -196,7 +197,8 @@ def test_fused_training_y_not_x():
     loss_ref = torch.nn.functional.mse_loss(out, Yr)
     loss_ref.backward()
     kernel_loss = float(loss.item()) / (B * F_out)
-    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+    ref_val = float(loss_ref.detach())
+    assert abs(kernel_loss - ref_val) / max(ref_val, 1e-9) < 5e-2
     lin0 = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)][0]
     got = dw.sum(0, dtype=torch.float32).view(L, WMAX, WMAX)[0, :30, :20].T.cpu()
     ref = lin0.weight.grad.detach()
@@ -388,7 +390,8 @@ def test_fused_training_x256_matches_autograd():
     loss_ref = torch.nn.functional.mse_loss(out, Xref)
     loss_ref.backward()
     kernel_loss = float(loss.item()) / (B * F_out)
-    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+    ref_val = float(loss_ref.detach())
+    assert abs(kernel_loss - ref_val) / max(ref_val, 1e-9) < 5e-2
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
     dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WM, WM).cpu()
@@ -593,7 +596,8 @@ def test_default_hourglass_l7_fused_grads_match_autograd():
     loss_ref = torch.nn.functional.mse_loss(ref_net(Xref), Xref)
     loss_ref.backward()
     kernel_loss = float(loss.item()) / (B * F_out)
-    assert abs(kernel_loss - float(loss_ref.detach())) / max(float(loss_ref), 1e-9) < 5e-2
+    ref_val = float(loss_ref.detach())
+    assert abs(kernel_loss - ref_val) / max(ref_val, 1e-9) < 5e-2
 
     linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
     assert len(linears) == 7
