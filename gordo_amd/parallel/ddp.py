@@ -83,8 +83,6 @@ class HourglassStepTrainer:
     """
 
     def __init__(self, module, X: np.ndarray, batch_size: int, cfg, dp: Optional[DPContext] = None):
-        from ..runtime.trainer import TrainConfig  # noqa: F401 (type only)
-
         self.dp = dp or DPContext()
         self.device = self.dp.device
         self.cfg = cfg
