@@ -52,6 +52,12 @@ def test_download_model(client):
 
     model = loads(r.content)
     assert hasattr(model, "anomaly")
+    # the server's runtime was loaded from an EXTERNALIZED artifact
+    # (weights in safetensors); the download bytes must be self-contained —
+    # the restored model has to actually predict, not just unpickle
+    X = _x_frame()
+    out = model.predict(X.values)
+    assert out.shape == (len(X), len(TAGS)) and np.isfinite(out).all()
 
 
 def test_prediction_json(client):
