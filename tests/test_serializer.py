@@ -101,6 +101,29 @@ def test_dump_load_predict_equality(tmp_path):
     assert (tmp_path / "definition.yaml").exists()
 
 
+def test_dump_load_dump_load_chain(tmp_path):
+    """A model restored from an externalized artifact must re-dump with its
+    weights intact (re-deploy / artifact-migration path)."""
+    rng = np.random.default_rng(3)
+    X = rng.normal(size=(80, 6)).astype(np.float32)
+    pipe = Pipeline(
+        [
+            ("scaler", MinMaxScaler()),
+            ("model", TorchAutoEncoder(kind="feedforward_hourglass", epochs=2, batch_size=32, seed=2)),
+        ]
+    )
+    pipe.fit(X)
+    expected = pipe.predict(X)
+
+    d1, d2 = tmp_path / "a", tmp_path / "b"
+    serializer.dump(pipe, str(d1))
+    first = serializer.load(str(d1))
+    serializer.dump(first, str(d2))
+    assert (d2 / "weights.safetensors").exists()
+    second = serializer.load(str(d2))
+    np.testing.assert_allclose(expected, second.predict(X), rtol=1e-5, atol=1e-6)
+
+
 def test_dumps_loads_bytes():
     model = TorchAutoEncoder(kind="feedforward_hourglass", epochs=1)
     X = np.random.default_rng(1).normal(size=(50, 5)).astype(np.float32)
