@@ -49,6 +49,55 @@ def build_fleet_app(models_dir: str, project_name: str = "project",
     async def healthcheck():
         return {"gordo-amd-version": __version__, "models": len(mounted)}
 
+    # Prometheus scrape endpoint: request counts/latency per machine path and
+    # a resident-model gauge. Registry is per-app so repeated build_fleet_app
+    # calls (tests, reloads) don't collide in the process-global registry.
+    try:
+        from prometheus_client import (
+            CollectorRegistry,
+            Counter,
+            Gauge,
+            Histogram,
+            generate_latest,
+            CONTENT_TYPE_LATEST,
+        )
+
+        registry = CollectorRegistry()
+        req_count = Counter(
+            "gordo_requests_total", "HTTP requests served",
+            ["machine", "status"], registry=registry,
+        )
+        req_latency = Histogram(
+            "gordo_request_seconds", "HTTP request latency",
+            ["machine"], registry=registry,
+            buckets=(0.001, 0.0025, 0.005, 0.01, 0.025, 0.05, 0.1, 0.25, 1.0, 5.0),
+        )
+        Gauge(
+            "gordo_resident_models", "models mounted in this fleet server",
+            registry=registry,
+        ).set(len(mounted))
+
+        @app.middleware("http")
+        async def _observe(request, call_next):
+            import time
+
+            parts = request.url.path.split("/")
+            # /gordo/v0/<project>/<machine>/... -> label by machine
+            machine = parts[4] if len(parts) > 4 and parts[1] == "gordo" else "_fleet"
+            t0 = time.perf_counter()
+            response = await call_next(request)
+            req_latency.labels(machine).observe(time.perf_counter() - t0)
+            req_count.labels(machine, str(response.status_code)).inc()
+            return response
+
+        @app.get("/metrics")
+        async def metrics():
+            from fastapi import Response
+
+            return Response(generate_latest(registry), media_type=CONTENT_TYPE_LATEST)
+    except ImportError:  # pragma: no cover - prometheus_client not installed
+        logger.info("prometheus_client unavailable; /metrics disabled")
+
     @app.get("/")
     async def root():
         return {

@@ -120,3 +120,11 @@ def test_fleet_app(tmp_path):
         r = c.get(eps["machine-a"] + "/metadata")
         assert r.status_code == 200
         assert r.json()["metadata"]["name"] == "machine-a"
+
+        # Prometheus scrape reflects the traffic above, labeled by machine
+        r = c.get("/metrics")
+        assert r.status_code == 200
+        body = r.text
+        assert "gordo_resident_models 2.0" in body
+        assert 'gordo_requests_total{machine="machine-a",status="200"}' in body
+        assert "gordo_request_seconds_bucket" in body
