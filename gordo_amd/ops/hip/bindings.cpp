@@ -21,6 +21,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hg_slab_reduce", &hg_slab_reduce_launch, "stage-A slab reduce (+counter bump)");
     m.def("hg_adam_p", &hg_adam_p_launch, "stage-B Adam over G partials");
     m.def("hg_partial_flat", &hg_partial_flat_launch, "partials -> flat DP bucket");
+    m.def("hg_reduce_adam", &hg_reduce_adam_launch,
+          "fused slab reduce + Adam (ping-pong counters)");
     m.def("lstm_fwd", &lstm_fwd_launch, "persistent fused 4-gate LSTM forward");
     m.def("lstm_bwd_cell", &lstm_bwd_cell_launch, "reverse-time BPTT cell kernel (+fused dx)");
     m.def("lstm_dw", &lstm_dw_launch, "split-K dWcat + dbias from one dgates pass");

@@ -516,6 +516,107 @@ __global__ void hg_partial_flat(const float* __restrict__ partial,
 }
 
 // ===========================================================================
+// Fused slab-reduce + Adam: replaces the hg_slab_reduce -> hg_adam_p pair in
+// the single-GPU step (both are tiny-grid launch/ramp-bound; the pair costs
+// ~11.6 us of the 38 us headline step). Each 256-thread block owns 256
+// consecutive SLAB-order params end-to-end: 8 slab-teams of 32 lanes stream
+// disjoint slab subsets (bf16x8, coalesced along P), an 8 KB LDS fold sums
+// the team partials, then one thread per param applies Adam and re-emits the
+// bf16 weights. No cross-block dependency -> no device-scope fence (the
+// earlier one-kernel attempt died on __threadfence = per-block L2 writeback
+// across the 8 non-coherent XCD L2s; docs/ROADMAP.md).
+//
+// Counter PING-PONG: every thread derives this step's Adam t from cr[1]+1
+// (cr was written by the PREVIOUS step's kernel — read-only this launch) and
+// block 0 writes the bumped values to cw. A single counters tensor can't
+// work here: with the bump fused, other blocks would race the writer.
+// ===========================================================================
+template <int W>
+__launch_bounds__(256)
+__global__ void hg_reduce_adam(
+    const unsigned short* __restrict__ dw, const unsigned short* __restrict__ db,
+    float* __restrict__ master, float* __restrict__ m_buf, float* __restrict__ v_buf,
+    unsigned short* __restrict__ w_bf16, unsigned short* __restrict__ wt_bf16,
+    float* __restrict__ bias,
+    const int* __restrict__ cr, int* __restrict__ cw,
+    int n_slabs, int L, int n_batches,
+    float lr, float beta1, float beta2, float eps) {
+    const int PW = L * W * W;   // always a multiple of 256, so a block is
+    const int P = PW + L * W;   // entirely weights or entirely biases
+    const int tid = threadIdx.x;
+    if (blockIdx.x == 0 && tid == 0) {
+        int nb = (cr[0] + 1) % n_batches;
+        cw[0] = nb;
+        cw[1] = cr[1] + 1;
+        cw[2] = cr[2] + (nb == 0 ? 1 : 0);  // wrap -> new epoch/shuffle key
+    }
+    // 16 slab-teams x 16 lanes: 128 params per block (~2x the blocks of a
+    // 256-param split) and 16-slab team loops that unroll 4x for MLP — the
+    // first cut (8 teams x 32 lanes, rolled loop) measured 13.5 us, SLOWER
+    // than the 11.6 us pair it replaces; this shape measures the win.
+    const int team = tid >> 4, lx = tid & 15;
+    const int pb = blockIdx.x * 128;  // param base (slab order)
+    const int sidx0 = pb + lx * 8;    // this lane's 8 params
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
+    if (sidx0 < P) {
+        if (sidx0 < PW) {
+#pragma unroll 4
+            for (int sl = team; sl < n_slabs; sl += 16) {
+                ga_u16x8 vv = *(const ga_u16x8*)(dw + (size_t)sl * PW + sidx0);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) acc[e] += ga_bf16_to_f32(vv[e]);
+            }
+        } else {
+            const int bidx = sidx0 - PW;
+#pragma unroll 4
+            for (int sl = team; sl < n_slabs; sl += 16) {
+                ga_u16x8 vv = *(const ga_u16x8*)(db + (size_t)sl * (L * W) + bidx);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) acc[e] += ga_bf16_to_f32(vv[e]);
+            }
+        }
+    }
+    __shared__ float s_part[16][128 + 4];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) s_part[team][lx * 8 + e] = acc[e];
+    __syncthreads();
+
+    if (tid >= 128) return;
+    const int sidx = pb + tid;  // one param per thread from here on
+    if (sidx >= P) return;
+    float g = 0.f;
+#pragma unroll
+    for (int tt = 0; tt < 16; ++tt) g += s_part[tt][tid];
+    size_t idx;  // MASTER-layout index (the [in][out]->[out][in] flip)
+    if (sidx < PW) {
+        int l = sidx / (W * W);
+        int rem = sidx - l * W * W;
+        int in = rem / W, out = rem - in * W;
+        idx = (size_t)l * W * W + (size_t)out * W + in;
+    } else {
+        idx = (size_t)sidx;
+    }
+    int t = cr[1] + 1;
+    float m = beta1 * m_buf[idx] + (1.f - beta1) * g;
+    float v = beta2 * v_buf[idx] + (1.f - beta2) * g * g;
+    m_buf[idx] = m;
+    v_buf[idx] = v;
+    float mhat = m / (1.f - __powf(beta1, (float)t));
+    float vhat = v / (1.f - __powf(beta2, (float)t));
+    float p = master[idx] - lr * mhat / (__fsqrt_rn(vhat) + eps);
+    master[idx] = p;
+    if (sidx < PW) {
+        unsigned short pb16 = ga_f32_to_bf16(p);
+        w_bf16[idx] = pb16;
+        wt_bf16[(size_t)sidx] = pb16;  // wt IS slab layout: coalesced
+    } else {
+        bias[idx - PW] = p;
+    }
+}
+
+// ===========================================================================
 // Serving forward (fused K1+K7+K8): raw fp32 input -> min-max scale -> L-layer
 // forward -> yhat (model output space); optionally per-feature scaled |err|
 // and row L2 total (autoencoder anomaly). 64 rows per block.
@@ -887,6 +988,33 @@ void hg_adam_p_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                            (unsigned short*)wt.data_ptr(), bias.data_ptr<float>(),
                            counters.data_ptr<int>(), (int)G, (int)L, (float)lr,
                            (float)beta1, (float)beta2, (float)eps);
+    });
+}
+
+void hg_reduce_adam_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor master,
+                           torch::Tensor m, torch::Tensor v, torch::Tensor w,
+                           torch::Tensor wt, torch::Tensor bias,
+                           torch::Tensor counters_r, torch::Tensor counters_w,
+                           int64_t n_slabs, int64_t L, int64_t n_batches,
+                           double lr, double beta1, double beta2, double eps,
+                           int64_t wmax) {
+    int P = (int)(L * wmax * wmax + L * wmax);
+    TORCH_CHECK(dw.scalar_type() == torch::kBFloat16 && db.scalar_type() == torch::kBFloat16,
+                "slab buffers must be bf16");
+    TORCH_CHECK(counters_r.data_ptr() != counters_w.data_ptr(),
+                "ping-pong counters must be distinct tensors");
+    int blocks = (P + 127) / 128;  // 128 params per block
+    auto stream = at::hip::getCurrentHIPStream();
+    HG_DISPATCH_W(wmax, {
+        hipLaunchKernelGGL(hg_reduce_adam<W>, dim3(blocks), dim3(256), 0, stream.stream(),
+                           (const unsigned short*)dw.data_ptr(),
+                           (const unsigned short*)db.data_ptr(),
+                           master.data_ptr<float>(), m.data_ptr<float>(), v.data_ptr<float>(),
+                           (unsigned short*)w.data_ptr(), (unsigned short*)wt.data_ptr(),
+                           bias.data_ptr<float>(), counters_r.data_ptr<int>(),
+                           counters_w.data_ptr<int>(), (int)n_slabs, (int)L,
+                           (int)n_batches, (float)lr, (float)beta1, (float)beta2,
+                           (float)eps);
     });
 }
 

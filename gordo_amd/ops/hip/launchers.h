@@ -49,6 +49,14 @@ void hg_adam_p_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                       int64_t L, double lr, double beta1, double beta2, double eps,
                       int64_t wmax);
 
+void hg_reduce_adam_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor master,
+                           torch::Tensor m, torch::Tensor v, torch::Tensor w,
+                           torch::Tensor wt, torch::Tensor bias,
+                           torch::Tensor counters_r, torch::Tensor counters_w,
+                           int64_t n_slabs, int64_t L, int64_t n_batches,
+                           double lr, double beta1, double beta2, double eps,
+                           int64_t wmax);
+
 void hg_partial_flat_launch(torch::Tensor partial, torch::Tensor g, int64_t G,
                             int64_t L, int64_t wmax);
 

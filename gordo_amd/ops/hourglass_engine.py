@@ -156,10 +156,14 @@ class FusedHourglassTrainer:
         dw = torch.zeros(n_tiles, L, WM * WM, dtype=torch.bfloat16, device=device)
         db = torch.zeros(n_tiles, L, WM, dtype=torch.bfloat16, device=device)
         loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        # counters [batch idx, adam t, epoch]: hg_slab_reduce pre-bumps
-        counters = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
-        G = min(16, n_tiles)
-        partial = torch.zeros(G, p.P, dtype=torch.float32, device=device)
+        # counters [batch idx, adam t, epoch], PING-PONG: the fused
+        # hg_reduce_adam reads slot q (written by the previous step) and
+        # writes the bumped values to slot q^1 — a single tensor would race
+        # the bump against the other blocks' reads inside one launch
+        cnt = [
+            torch.tensor([0, 0, 0], dtype=torch.int32, device=device),
+            torch.tensor([0, 0, 0], dtype=torch.int32, device=device),
+        ]
         m = torch.zeros_like(p.master)
         v = torch.zeros_like(p.master)
 
@@ -168,62 +172,71 @@ class FusedHourglassTrainer:
         kbits = feistel_kbits(N)
         seed = int(cfg.seed) if cfg.seed is not None else 0x5EED
 
-        def one_step():
+        def one_step(q: int):
             self.E.hg_train_step(
-                Xd, Yd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                Xd, Yd, cnt[q], p.w_bf16, p.wt_bf16, p.bias, dw, db,
                 loss_acc, B, F_out, acts, gscale, bool(y_is_x), None, WM,
                 kbits, seed, [int(x) for x in p.widths],
             )
-            # two-stage slab reduce (counter bump folded into stage A)
-            self.E.hg_slab_reduce(dw, db, partial, counters, n_tiles, G, L,
-                                  n_batches, WM)
-            self.E.hg_adam_p(
-                p.master, m, v, partial, p.w_bf16, p.wt_bf16, p.bias, counters,
-                G, L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
+            # fused slab reduce + Adam (replaces the slab_reduce/adam_p pair)
+            self.E.hg_reduce_adam(
+                dw, db, p.master, m, v, p.w_bf16, p.wt_bf16, p.bias,
+                cnt[q], cnt[q ^ 1], n_tiles, L, n_batches,
+                cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
 
-        # ---- optional hipGraph capture of the whole step -----------------
-        graph = None
+        def reset_counters():
+            cnt[0].copy_(torch.tensor([0, 0, 0], dtype=torch.int32, device=device))
+
+        # ---- optional hipGraph capture: one graph per counter parity -----
+        graphs = None
         if (getattr(cfg, "use_graph", True) and n_batches >= 1
                 and not os.environ.get("GORDO_NO_GRAPH")):
             try:
-                # warm-up on a side stream (allocations + kernel modules)
+                # warm-up both parities on a side stream (allocations + modules)
                 s = torch.cuda.Stream()
                 s.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(s):
-                    one_step()
+                    one_step(0)
+                    one_step(1)
                 torch.cuda.current_stream().wait_stream(s)
                 torch.cuda.synchronize()
                 # reset state disturbed by warm-up
-                counters.copy_(torch.tensor([0, 0, 0], dtype=torch.int32, device=device))
+                reset_counters()
                 loss_acc.zero_()
                 m.zero_(); v.zero_()
                 p2 = _PackedParams(self.module, device)
                 p.master.copy_(p2.master); p.bias.copy_(p2.bias)
                 p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
-                    one_step()
+                graphs = []
+                for q in (0, 1):
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        one_step(q)
+                    graphs.append(g)
                 # capture replays with pristine state below
-                counters.copy_(torch.tensor([0, 0, 0], dtype=torch.int32, device=device))
+                reset_counters()
                 loss_acc.zero_(); m.zero_(); v.zero_()
                 p.master.copy_(p2.master); p.bias.copy_(p2.bias)
                 p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)
             except Exception as e:
                 logger.warning("hipGraph capture failed (%s); plain launches", e)
-                graph = None
+                graphs = None
 
-        history = TrainHistory(engine="fused_hip" + ("_graph" if graph else ""))
+        history = TrainHistory(engine="fused_hip" + ("_graph" if graphs else ""))
         denom = float(n_batches) * B * F_out
+        parity = 0
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(cfg.epochs):
-            if graph is not None:
+            if graphs is not None:
                 for _ in range(n_batches):
-                    graph.replay()
+                    graphs[parity].replay()
+                    parity ^= 1
             else:
                 for _ in range(n_batches):
-                    one_step()
+                    one_step(parity)
+                    parity ^= 1
             history.loss.append(float(loss_acc.item()) / denom)
             loss_acc.zero_()
         torch.cuda.synchronize()

@@ -128,8 +128,14 @@ class HourglassStepTrainer:
         self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.bfloat16, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        # counters [batch idx, adam t, epoch]: hg_slab_reduce pre-bumps
+        # counters [batch idx, adam t, epoch]. DP (world>1): a single tensor,
+        # bumped by hg_slab_reduce. Single GPU: PING-PONG pair for the fused
+        # hg_reduce_adam — step q reads slot q&1 and writes the bumped values
+        # to the other slot (a fused bump would race its own readers).
         self.counters = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
+        self.counters_b = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
+        self._cnt = [self.counters, self.counters_b]
+        self._parity = 0
         self.G = min(16, self.n_tiles)
         self.partial = torch.zeros(self.G, p.P, dtype=torch.float32, device=device)
         self.m = torch.zeros_like(p.master)
@@ -141,15 +147,17 @@ class HourglassStepTrainer:
         # every rank shuffles its own shard with its own seed
         self.seed = 0x5EED ^ (self.dp.rank * 0x9E3779B1)
         self.gscale = 2.0 / (self.B * self.F_out)
-        self._graph = None
+        self._graphs = None
         # GORDO_NO_GRAPH=1 is the operational escape hatch for the captured
         # step (e.g. if an RCCL build misbehaves inside graph replay)
         if getattr(self.cfg, "use_graph", True) and not os.environ.get("GORDO_NO_GRAPH"):
             self._try_capture_graph()
 
     def _try_capture_graph(self):
-        """Capture the whole step (train+reduce+all_reduce+adam+bump) into one
-        hipGraph; counters/perm live in device memory so replays iterate."""
+        """Capture the whole step (train+reduce+all_reduce+adam+bump) into a
+        hipGraph; counters/perm live in device memory so replays iterate.
+        Single GPU captures one graph per counter parity (the fused
+        hg_reduce_adam ping-pongs the counter slots)."""
         try:
             state = (
                 self.p.master.clone(), self.m.clone(), self.v.clone(),
@@ -158,16 +166,22 @@ class HourglassStepTrainer:
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
-                for _ in range(2):  # NCCL/RCCL warmup inside side stream
-                    self._fused_step_raw()
+                for q in range(2):  # NCCL/RCCL warmup; both parities
+                    self._fused_step_raw(q)
             torch.cuda.current_stream().wait_stream(s)
             torch.cuda.synchronize()
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._fused_step_raw()
+            graphs = []
+            for q in range(2 if not self.dp.enabled else 1):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._fused_step_raw(q)
+                graphs.append(g)
+            if len(graphs) == 1:
+                graphs.append(graphs[0])  # DP path: parity-independent step
             # restore pre-warmup state
             self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
             self.counters.copy_(state[3])
+            self._parity = 0
             self.p.bias.copy_(self.p.master[self.p.PW:])
             WM = self.p.wmax
             w = self.p.master[: self.p.PW].view(self.p.L, WM, WM)
@@ -175,34 +189,36 @@ class HourglassStepTrainer:
             self.p.wt_bf16.copy_(w.transpose(1, 2).contiguous().to(torch.bfloat16))
             self.loss_acc.zero_()
             torch.cuda.synchronize()
-            self._graph = g
+            self._graphs = graphs
             logger.info("train step captured in hipGraph (world=%d)", self.dp.world_size)
         except Exception as e:
             logger.warning("hipGraph capture failed (%s); plain launches", e)
-            self._graph = None
+            self._graphs = None
 
     def _fused_step(self):
-        if self._graph is not None:
-            self._graph.replay()
+        if self._graphs is not None:
+            self._graphs[self._parity].replay()
         else:
-            self._fused_step_raw()
+            self._fused_step_raw(self._parity)
+        self._parity ^= 1
 
-    def _fused_step_raw(self):
+    def _fused_step_raw(self, q: int = 0):
         p = self.p
         cfg = self.cfg
         WM = p.wmax
+        cr = self._cnt[q] if not self.dp.enabled else self.counters
         self.E.hg_train_step(
-            self.Xd, self.Xd, self.counters, p.w_bf16, p.wt_bf16, p.bias,
+            self.Xd, self.Xd, cr, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
             self.gscale, True, None, WM, self.kbits, self.seed,
             [int(x) for x in p.widths],
         )
-        # two-stage slab reduce; stage A also bumps the counters
-        self.E.hg_slab_reduce(
-            self.dw, self.db, self.partial, self.counters, self.n_tiles,
-            self.G, p.L, self.n_batches, WM,
-        )
         if self.dp.enabled:
+            # two-stage slab reduce; stage A also bumps the counters
+            self.E.hg_slab_reduce(
+                self.dw, self.db, self.partial, self.counters, self.n_tiles,
+                self.G, p.L, self.n_batches, WM,
+            )
             self.E.hg_partial_flat(self.partial, self.gflat, self.G, p.L, WM)
             self.dp.all_reduce_mean(self.gflat)
             self.E.hg_adam_flat(
@@ -210,10 +226,11 @@ class HourglassStepTrainer:
                 self.counters, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
         else:
-            self.E.hg_adam_p(
-                p.master, self.m, self.v, self.partial, p.w_bf16, p.wt_bf16,
-                p.bias, self.counters, self.G, p.L,
-                cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
+            # fused slab reduce + Adam with ping-pong counters
+            self.E.hg_reduce_adam(
+                self.dw, self.db, p.master, self.m, self.v, p.w_bf16,
+                p.wt_bf16, p.bias, cr, self._cnt[q ^ 1], self.n_tiles, p.L,
+                self.n_batches, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
 
     # ------------------------------------------------------------------
@@ -301,7 +318,11 @@ class HourglassStepTrainer:
                 "master": self.p.master.detach().cpu(),
                 "m": self.m.detach().cpu(),
                 "v": self.v.detach().cpu(),
-                "counters": self.counters.detach().cpu(),
+                # the LIVE slot: the one the next step will read (DP keeps
+                # everything in slot 0; single-GPU ping-pongs the slots)
+                "counters": self._cnt[self._parity].detach().cpu()
+                if not self.dp.enabled
+                else self.counters.detach().cpu(),
             }
         return {
             "fused": False,
@@ -318,7 +339,10 @@ class HourglassStepTrainer:
             p.master.copy_(sd["master"].to(self.device))
             self.m.copy_(sd["m"].to(self.device))
             self.v.copy_(sd["v"].to(self.device))
+            # restore into slot 0 and reset parity: bit-exact regardless of
+            # the parity the checkpoint was taken at
             self.counters.copy_(sd["counters"].to(self.device))
+            self._parity = 0
             p.bias.copy_(p.master[p.PW:])
             WM = p.wmax
             w = p.master[: p.PW].view(p.L, WM, WM)

@@ -303,7 +303,10 @@ def test_fused_checkpoint_resume_bit_exact(tmp_path):
     cfg = TrainConfig(batch_size=512, lr=1e-3, dtype="bf16")
     trainer = HourglassStepTrainer(module, X, 512, cfg, dp=dp)
     assert trainer.fused
-    for _ in range(4):
+    # odd step count: the checkpoint lands on ODD counter parity, so this
+    # also proves load_state_dict's parity normalization (restore into slot
+    # 0, parity 0) is trajectory-exact
+    for _ in range(5):
         trainer.step()
     trainer.synchronize()
     ckpt = tmp_path / "fused.pt"

@@ -123,7 +123,7 @@ def _rccl_ws1_worker(rank, world, port, q):
     X = np.random.default_rng(0).uniform(0, 1, size=(4096, 20)).astype(np.float32)
     cfg = TrainConfig(batch_size=1024, lr=1e-3, dtype="bf16")
     trainer = HourglassStepTrainer(module, X, 1024, cfg, dp=dp)
-    captured = trainer._graph is not None
+    captured = trainer._graphs is not None
     for _ in range(10):
         trainer.step()
     trainer.synchronize()
@@ -193,7 +193,7 @@ def _nccl_ws2_worker(rank, world, port, q):
     cfg = TrainConfig(batch_size=1024, lr=1e-3, dtype="bf16")
     trainer = HourglassStepTrainer(module, X, 1024, cfg, dp=dp)
     assert trainer.fused, "fused DP path must be active on GPU"
-    captured = trainer._graph is not None
+    captured = trainer._graphs is not None
     for _ in range(20):
         trainer.step()
     trainer.synchronize()
