@@ -23,6 +23,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hg_partial_flat", &hg_partial_flat_launch, "partials -> flat DP bucket");
     m.def("hg_reduce_adam", &hg_reduce_adam_launch,
           "fused slab reduce + Adam (ping-pong counters)");
+    m.def("hg_flat_reduce", &hg_flat_reduce_launch,
+          "fused slabs -> flat DP bucket (ping-pong counters)");
     m.def("lstm_fwd", &lstm_fwd_launch, "persistent fused 4-gate LSTM forward");
     m.def("lstm_bwd_cell", &lstm_bwd_cell_launch, "reverse-time BPTT cell kernel (+fused dx)");
     m.def("lstm_dw", &lstm_dw_launch, "split-K dWcat + dbias from one dgates pass");

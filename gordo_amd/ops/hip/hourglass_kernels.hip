@@ -617,6 +617,75 @@ __global__ void hg_reduce_adam(
 }
 
 // ===========================================================================
+// DP variant of the fused reduce: slabs -> MASTER-order flat bucket (the
+// RCCL all-reduce payload) in ONE kernel, with the same ping-pong counter
+// bump. Replaces the hg_slab_reduce -> hg_partial_flat pair in the world>1
+// step; hg_adam_flat then reads the bumped slot (a later kernel — no race).
+// ===========================================================================
+template <int W>
+__launch_bounds__(256)
+__global__ void hg_flat_reduce(
+    const unsigned short* __restrict__ dw, const unsigned short* __restrict__ db,
+    float* __restrict__ g_out,
+    const int* __restrict__ cr, int* __restrict__ cw,
+    int n_slabs, int L, int n_batches) {
+    const int PW = L * W * W;
+    const int P = PW + L * W;
+    const int tid = threadIdx.x;
+    if (blockIdx.x == 0 && tid == 0) {
+        int nb = (cr[0] + 1) % n_batches;
+        cw[0] = nb;
+        cw[1] = cr[1] + 1;
+        cw[2] = cr[2] + (nb == 0 ? 1 : 0);
+    }
+    const int team = tid >> 4, lx = tid & 15;
+    const int pb = blockIdx.x * 128;
+    const int sidx0 = pb + lx * 8;
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
+    if (sidx0 < P) {
+        if (sidx0 < PW) {
+#pragma unroll 4
+            for (int sl = team; sl < n_slabs; sl += 16) {
+                ga_u16x8 vv = *(const ga_u16x8*)(dw + (size_t)sl * PW + sidx0);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) acc[e] += ga_bf16_to_f32(vv[e]);
+            }
+        } else {
+            const int bidx = sidx0 - PW;
+#pragma unroll 4
+            for (int sl = team; sl < n_slabs; sl += 16) {
+                ga_u16x8 vv = *(const ga_u16x8*)(db + (size_t)sl * (L * W) + bidx);
+#pragma unroll
+                for (int e = 0; e < 8; ++e) acc[e] += ga_bf16_to_f32(vv[e]);
+            }
+        }
+    }
+    __shared__ float s_part[16][128 + 4];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) s_part[team][lx * 8 + e] = acc[e];
+    __syncthreads();
+
+    if (tid >= 128) return;
+    const int sidx = pb + tid;
+    if (sidx >= P) return;
+    float g = 0.f;
+#pragma unroll
+    for (int tt = 0; tt < 16; ++tt) g += s_part[tt][tid];
+    size_t idx;  // MASTER layout for the DP bucket (matches hg_adam_flat)
+    if (sidx < PW) {
+        int l = sidx / (W * W);
+        int rem = sidx - l * W * W;
+        int in = rem / W, out = rem - in * W;
+        idx = (size_t)l * W * W + (size_t)out * W + in;
+    } else {
+        idx = (size_t)sidx;
+    }
+    g_out[idx] = g;
+}
+
+// ===========================================================================
 // Serving forward (fused K1+K7+K8): raw fp32 input -> min-max scale -> L-layer
 // forward -> yhat (model output space); optionally per-feature scaled |err|
 // and row L2 total (autoencoder anomaly). 64 rows per block.
@@ -1015,6 +1084,27 @@ void hg_reduce_adam_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor mas
                            counters_w.data_ptr<int>(), (int)n_slabs, (int)L,
                            (int)n_batches, (float)lr, (float)beta1, (float)beta2,
                            (float)eps);
+    });
+}
+
+void hg_flat_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,
+                           torch::Tensor counters_r, torch::Tensor counters_w,
+                           int64_t n_slabs, int64_t L, int64_t n_batches,
+                           int64_t wmax) {
+    int P = (int)(L * wmax * wmax + L * wmax);
+    TORCH_CHECK(dw.scalar_type() == torch::kBFloat16 && db.scalar_type() == torch::kBFloat16,
+                "slab buffers must be bf16");
+    TORCH_CHECK(counters_r.data_ptr() != counters_w.data_ptr(),
+                "ping-pong counters must be distinct tensors");
+    int blocks = (P + 127) / 128;
+    auto stream = at::hip::getCurrentHIPStream();
+    HG_DISPATCH_W(wmax, {
+        hipLaunchKernelGGL(hg_flat_reduce<W>, dim3(blocks), dim3(256), 0, stream.stream(),
+                           (const unsigned short*)dw.data_ptr(),
+                           (const unsigned short*)db.data_ptr(),
+                           g.data_ptr<float>(), counters_r.data_ptr<int>(),
+                           counters_w.data_ptr<int>(), (int)n_slabs, (int)L,
+                           (int)n_batches);
     });
 }
 

@@ -57,6 +57,11 @@ void hg_reduce_adam_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor mas
                            double lr, double beta1, double beta2, double eps,
                            int64_t wmax);
 
+void hg_flat_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,
+                           torch::Tensor counters_r, torch::Tensor counters_w,
+                           int64_t n_slabs, int64_t L, int64_t n_batches,
+                           int64_t wmax);
+
 void hg_partial_flat_launch(torch::Tensor partial, torch::Tensor g, int64_t G,
                             int64_t L, int64_t wmax);
 

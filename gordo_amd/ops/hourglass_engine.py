@@ -214,6 +214,12 @@ class FusedHourglassTrainer:
                     with torch.cuda.graph(g):
                         one_step(q)
                     graphs.append(g)
+                # even-parity PAIR graph: two steps per replay — halves the
+                # per-step Python/replay overhead in the epoch loop
+                pair_graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(pair_graph):
+                    one_step(0)
+                    one_step(1)
                 # capture replays with pristine state below
                 reset_counters()
                 loss_acc.zero_(); m.zero_(); v.zero_()
@@ -222,6 +228,7 @@ class FusedHourglassTrainer:
             except Exception as e:
                 logger.warning("hipGraph capture failed (%s); plain launches", e)
                 graphs = None
+                pair_graph = None
 
         history = TrainHistory(engine="fused_hip" + ("_graph" if graphs else ""))
         denom = float(n_batches) * B * F_out
@@ -229,14 +236,21 @@ class FusedHourglassTrainer:
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(cfg.epochs):
+            r = n_batches
             if graphs is not None:
-                for _ in range(n_batches):
-                    graphs[parity].replay()
-                    parity ^= 1
+                while r:
+                    if parity == 0 and r >= 2:
+                        pair_graph.replay()  # parity 0 -> 0
+                        r -= 2
+                    else:
+                        graphs[parity].replay()
+                        parity ^= 1
+                        r -= 1
             else:
-                for _ in range(n_batches):
+                while r:
                     one_step(parity)
                     parity ^= 1
+                    r -= 1
             history.loss.append(float(loss_acc.item()) / denom)
             loss_acc.zero_()
         torch.cuda.synchronize()

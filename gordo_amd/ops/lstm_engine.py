@@ -512,6 +512,12 @@ class LSTMStepTrainer:
                 sum(t.numel() for t in p.masters()), dtype=torch.float32, device=self.device
             )
 
+    def steps(self, n: int):
+        # ms-scale steps: per-step Python overhead is negligible here (the
+        # hourglass trainer's pair-graph trick buys nothing at 10 ms/step)
+        for _ in range(n):
+            self.step()
+
     def step(self):
         st = self.st
         win = st["perm"][self._s * self.B : (self._s + 1) * self.B]

@@ -148,6 +148,7 @@ class HourglassStepTrainer:
         self.seed = 0x5EED ^ (self.dp.rank * 0x9E3779B1)
         self.gscale = 2.0 / (self.B * self.F_out)
         self._graphs = None
+        self._pair_graph = None
         # GORDO_NO_GRAPH=1 is the operational escape hatch for the captured
         # step (e.g. if an RCCL build misbehaves inside graph replay)
         if getattr(self.cfg, "use_graph", True) and not os.environ.get("GORDO_NO_GRAPH"):
@@ -171,13 +172,19 @@ class HourglassStepTrainer:
             torch.cuda.current_stream().wait_stream(s)
             torch.cuda.synchronize()
             graphs = []
-            for q in range(2 if not self.dp.enabled else 1):
+            for q in range(2):  # one graph per counter parity
                 g = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g):
                     self._fused_step_raw(q)
                 graphs.append(g)
-            if len(graphs) == 1:
-                graphs.append(graphs[0])  # DP path: parity-independent step
+            # even-parity PAIR graph: two whole steps per replay halves the
+            # per-step Python/launch overhead (visible once the kernels
+            # shrank to ~30 us/step)
+            gp = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gp):
+                self._fused_step_raw(0)
+                self._fused_step_raw(1)
+            self._pair_graph = gp
             # restore pre-warmup state
             self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
             self.counters.copy_(state[3])
@@ -194,6 +201,7 @@ class HourglassStepTrainer:
         except Exception as e:
             logger.warning("hipGraph capture failed (%s); plain launches", e)
             self._graphs = None
+            self._pair_graph = None
 
     def _fused_step(self):
         if self._graphs is not None:
@@ -202,11 +210,28 @@ class HourglassStepTrainer:
             self._fused_step_raw(self._parity)
         self._parity ^= 1
 
+    def steps(self, n: int):
+        """Run n whole training steps, using the 2-step pair graph where
+        parity allows (same work as n step() calls, fewer replays)."""
+        if not self.fused or self._graphs is None:
+            for _ in range(n):
+                self.step()
+            return
+        i = 0
+        pair = getattr(self, "_pair_graph", None)
+        while i < n:
+            if pair is not None and self._parity == 0 and n - i >= 2:
+                pair.replay()  # parity 0 -> 0
+                i += 2
+            else:
+                self._fused_step()
+                i += 1
+
     def _fused_step_raw(self, q: int = 0):
         p = self.p
         cfg = self.cfg
         WM = p.wmax
-        cr = self._cnt[q] if not self.dp.enabled else self.counters
+        cr, cw = self._cnt[q], self._cnt[q ^ 1]
         self.E.hg_train_step(
             self.Xd, self.Xd, cr, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
@@ -214,22 +239,22 @@ class HourglassStepTrainer:
             [int(x) for x in p.widths],
         )
         if self.dp.enabled:
-            # two-stage slab reduce; stage A also bumps the counters
-            self.E.hg_slab_reduce(
-                self.dw, self.db, self.partial, self.counters, self.n_tiles,
-                self.G, p.L, self.n_batches, WM,
+            # fused slabs -> flat bucket (+ ping-pong counter bump); the
+            # Adam kernel runs after the collective and reads the bumped slot
+            self.E.hg_flat_reduce(
+                self.dw, self.db, self.gflat, cr, cw, self.n_tiles, p.L,
+                self.n_batches, WM,
             )
-            self.E.hg_partial_flat(self.partial, self.gflat, self.G, p.L, WM)
             self.dp.all_reduce_mean(self.gflat)
             self.E.hg_adam_flat(
                 p.master, self.m, self.v, self.gflat, p.w_bf16, p.wt_bf16, p.bias,
-                self.counters, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
+                cw, p.L, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
         else:
             # fused slab reduce + Adam with ping-pong counters
             self.E.hg_reduce_adam(
                 self.dw, self.db, p.master, self.m, self.v, p.w_bf16,
-                p.wt_bf16, p.bias, cr, self._cnt[q ^ 1], self.n_tiles, p.L,
+                p.wt_bf16, p.bias, cr, cw, self.n_tiles, p.L,
                 self.n_batches, cfg.lr, cfg.betas[0], cfg.betas[1], cfg.eps, WM,
             )
 
