@@ -214,12 +214,16 @@ class FusedHourglassTrainer:
                     with torch.cuda.graph(g):
                         one_step(q)
                     graphs.append(g)
-                # even-parity PAIR graph: two steps per replay — halves the
-                # per-step Python/replay overhead in the epoch loop
+                # even-parity PAIR/QUAD graphs: 2/4 steps per replay — cut
+                # the per-step Python/replay overhead in the epoch loop
                 pair_graph = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(pair_graph):
                     one_step(0)
                     one_step(1)
+                quad_graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(quad_graph):
+                    for q in (0, 1, 0, 1):
+                        one_step(q)
                 # capture replays with pristine state below
                 reset_counters()
                 loss_acc.zero_(); m.zero_(); v.zero_()
@@ -229,6 +233,7 @@ class FusedHourglassTrainer:
                 logger.warning("hipGraph capture failed (%s); plain launches", e)
                 graphs = None
                 pair_graph = None
+                quad_graph = None
 
         history = TrainHistory(engine="fused_hip" + ("_graph" if graphs else ""))
         denom = float(n_batches) * B * F_out
@@ -239,7 +244,10 @@ class FusedHourglassTrainer:
             r = n_batches
             if graphs is not None:
                 while r:
-                    if parity == 0 and r >= 2:
+                    if parity == 0 and r >= 4:
+                        quad_graph.replay()  # parity 0 -> 0
+                        r -= 4
+                    elif parity == 0 and r >= 2:
                         pair_graph.replay()  # parity 0 -> 0
                         r -= 2
                     else:

@@ -149,6 +149,7 @@ class HourglassStepTrainer:
         self.gscale = 2.0 / (self.B * self.F_out)
         self._graphs = None
         self._pair_graph = None
+        self._quad_graph = None
         # GORDO_NO_GRAPH=1 is the operational escape hatch for the captured
         # step (e.g. if an RCCL build misbehaves inside graph replay)
         if getattr(self.cfg, "use_graph", True) and not os.environ.get("GORDO_NO_GRAPH"):
@@ -177,14 +178,19 @@ class HourglassStepTrainer:
                 with torch.cuda.graph(g):
                     self._fused_step_raw(q)
                 graphs.append(g)
-            # even-parity PAIR graph: two whole steps per replay halves the
-            # per-step Python/launch overhead (visible once the kernels
+            # even-parity PAIR/QUAD graphs: 2/4 whole steps per replay cut
+            # the per-step Python/launch overhead (visible once the kernels
             # shrank to ~30 us/step)
             gp = torch.cuda.CUDAGraph()
             with torch.cuda.graph(gp):
                 self._fused_step_raw(0)
                 self._fused_step_raw(1)
             self._pair_graph = gp
+            gq = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gq):
+                for q in (0, 1, 0, 1):
+                    self._fused_step_raw(q)
+            self._quad_graph = gq
             # restore pre-warmup state
             self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
             self.counters.copy_(state[3])
@@ -202,6 +208,7 @@ class HourglassStepTrainer:
             logger.warning("hipGraph capture failed (%s); plain launches", e)
             self._graphs = None
             self._pair_graph = None
+            self._quad_graph = None
 
     def _fused_step(self):
         if self._graphs is not None:
@@ -219,8 +226,12 @@ class HourglassStepTrainer:
             return
         i = 0
         pair = getattr(self, "_pair_graph", None)
+        quad = getattr(self, "_quad_graph", None)
         while i < n:
-            if pair is not None and self._parity == 0 and n - i >= 2:
+            if quad is not None and self._parity == 0 and n - i >= 4:
+                quad.replay()  # parity 0 -> 0
+                i += 4
+            elif pair is not None and self._parity == 0 and n - i >= 2:
                 pair.replay()  # parity 0 -> 0
                 i += 2
             else:
