@@ -215,15 +215,20 @@ class FusedHourglassTrainer:
                         one_step(q)
                     graphs.append(g)
                 # even-parity PAIR/QUAD graphs: 2/4 steps per replay — cut
-                # the per-step Python/replay overhead in the epoch loop
-                pair_graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(pair_graph):
-                    one_step(0)
-                    one_step(1)
-                quad_graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(quad_graph):
-                    for q in (0, 1, 0, 1):
-                        one_step(q)
+                # the per-step Python/replay overhead in the epoch loop.
+                # Only worth the extra captures on jobs with enough steps
+                # (fleet builds run fit() per model; capture time is per-
+                # model overhead there)
+                pair_graph = quad_graph = None
+                if cfg.epochs * n_batches >= 64:
+                    pair_graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(pair_graph):
+                        one_step(0)
+                        one_step(1)
+                    quad_graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(quad_graph):
+                        for q in (0, 1, 0, 1):
+                            one_step(q)
                 # capture replays with pristine state below
                 reset_counters()
                 loss_acc.zero_(); m.zero_(); v.zero_()
@@ -244,10 +249,10 @@ class FusedHourglassTrainer:
             r = n_batches
             if graphs is not None:
                 while r:
-                    if parity == 0 and r >= 4:
+                    if quad_graph is not None and parity == 0 and r >= 4:
                         quad_graph.replay()  # parity 0 -> 0
                         r -= 4
-                    elif parity == 0 and r >= 2:
+                    elif pair_graph is not None and parity == 0 and r >= 2:
                         pair_graph.replay()  # parity 0 -> 0
                         r -= 2
                     else:
