@@ -1,4 +1,4 @@
-"""GPU DP tests: the FUSED data-parallel step (hg_partial_flat → all_reduce →
+"""GPU DP tests: the FUSED data-parallel step (hg_flat_reduce → all_reduce →
 hg_adam_flat, and the LSTM flat-bucket variant) with real collectives on CUDA
 tensors. A 1-GPU box can't run 2 nccl ranks (duplicate-device), so both ranks
 share cuda:0 over gloo — same code path as the driver's 8-rank RCCL run except
@@ -115,7 +115,7 @@ def _rccl_ws1_worker(rank, world, port, q):
     dist.init_process_group("nccl", rank=0, world_size=1)
     dp = DPContext(device=torch.device("cuda:0"))
     # world_size=1 disables DP by default; force the comm branch so the step
-    # runs hg_partial_flat -> dist.all_reduce (RCCL) -> hg_adam_flat and the
+    # runs hg_flat_reduce -> dist.all_reduce (RCCL) -> hg_adam_flat and the
     # hipGraph capture includes the collective — the 8-rank code path.
     dp.enabled = True
     dp.backend = "nccl"
