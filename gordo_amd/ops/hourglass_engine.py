@@ -216,11 +216,13 @@ class FusedHourglassTrainer:
                     graphs.append(g)
                 # even-parity PAIR/QUAD graphs: 2/4 steps per replay — cut
                 # the per-step Python/replay overhead in the epoch loop.
-                # Only worth the extra captures on jobs with enough steps
-                # (fleet builds run fit() per model; capture time is per-
-                # model overhead there)
+                # Only worth the extra captures on jobs with enough steps:
+                # two captures + two warmup steps cost ~10 ms per fit()
+                # while saving ~2.5 us/step — breakeven ~4k steps. Fleet
+                # builds (~300 steps/model) measured 10.1 -> 13.0 s per 256
+                # models with these captured unconditionally.
                 pair_graph = quad_graph = None
-                if cfg.epochs * n_batches >= 64:
+                if cfg.epochs * n_batches >= 4096:
                     pair_graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(pair_graph):
                         one_step(0)
