@@ -654,3 +654,32 @@ def test_default_hourglass_l7_trains_and_serves_fused(tmp_path):
     req = pd.DataFrame(X[:500], columns=[f"t{i}" for i in range(50)])
     frame = rt.anomaly(req, None)
     assert np.isfinite(frame["total-anomaly-score"].values).all()
+
+
+def test_ping_pong_counters_track_steps():
+    """The fused hg_reduce_adam bumps [batch, adam-t, epoch] across two
+    ping-pong counter slots; after n steps the LIVE slot (the one the next
+    step reads) must hold exactly n steps' worth of state."""
+    from gordo_amd.model.factories import feedforward_hourglass
+    from gordo_amd.parallel.ddp import DPContext, HourglassStepTrainer
+
+    torch.manual_seed(3)
+    module = feedforward_hourglass(50, 50, encoding_layers=2, compression_factor=0.5)
+    X = np.random.default_rng(3).uniform(0, 1, size=(4096, 50)).astype(np.float32)
+    cfg = TrainConfig(batch_size=1024, lr=1e-3, dtype="bf16")
+    trainer = HourglassStepTrainer(module, X, 1024, cfg, dp=DPContext(device=DEV))
+    assert trainer.fused
+    n_batches = trainer.n_batches  # 4096/1024 = 4
+
+    for n in (1, 2, 3, 7, 10):  # odd and even parities, epoch wraps
+        while int(trainer._cnt[trainer._parity][1].item()) < n:
+            trainer.step()
+        trainer.synchronize()
+        live = trainer._cnt[trainer._parity].cpu().tolist()
+        assert live == [n % n_batches, n, n // n_batches], (n, live)
+
+    # steps() (pair/quad graphs) advances the same counter trajectory
+    trainer.steps(9)
+    trainer.synchronize()
+    live = trainer._cnt[trainer._parity].cpu().tolist()
+    assert live == [19 % n_batches, 19, 19 // n_batches], live
