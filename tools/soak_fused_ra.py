@@ -5,8 +5,11 @@ is below the first (still learning / not diverged).
 
 Run on a GPU box:  python tools/soak_fused_ra.py [n_steps]
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import numpy as np
 import torch
