@@ -241,6 +241,14 @@ class FusedHourglassTrainer:
                 graphs = None
                 pair_graph = None
                 quad_graph = None
+                # warm-up may have stepped the real state before the failure:
+                # restore so the raw-launch run still starts pristine
+                torch.cuda.synchronize()
+                reset_counters()
+                loss_acc.zero_(); m.zero_(); v.zero_()
+                p2 = _PackedParams(self.module, device)
+                p.master.copy_(p2.master); p.bias.copy_(p2.bias)
+                p.w_bf16.copy_(p2.w_bf16); p.wt_bf16.copy_(p2.wt_bf16)
 
         history = TrainHistory(engine="fused_hip" + ("_graph" if graphs else ""))
         denom = float(n_batches) * B * F_out

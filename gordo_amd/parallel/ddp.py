@@ -160,11 +160,11 @@ class HourglassStepTrainer:
         hipGraph; counters/perm live in device memory so replays iterate.
         Single GPU captures one graph per counter parity (the fused
         hg_reduce_adam ping-pongs the counter slots)."""
+        state = (
+            self.p.master.clone(), self.m.clone(), self.v.clone(),
+            self.counters.clone(),
+        )
         try:
-            state = (
-                self.p.master.clone(), self.m.clone(), self.v.clone(),
-                self.counters.clone(),
-            )
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
@@ -209,6 +209,18 @@ class HourglassStepTrainer:
             self._graphs = None
             self._pair_graph = None
             self._quad_graph = None
+            # warm-up may have stepped the real state before the failure:
+            # restore so the raw-launch run still starts pristine
+            torch.cuda.synchronize()
+            self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
+            self.counters.copy_(state[3])
+            self._parity = 0
+            self.p.bias.copy_(self.p.master[self.p.PW:])
+            WM = self.p.wmax
+            w = self.p.master[: self.p.PW].view(self.p.L, WM, WM)
+            self.p.w_bf16.copy_(w.to(torch.bfloat16))
+            self.p.wt_bf16.copy_(w.transpose(1, 2).contiguous().to(torch.bfloat16))
+            self.loss_acc.zero_()
 
     def _fused_step(self):
         if self._graphs is not None:
