@@ -7,10 +7,13 @@ so the all-reduce is latency-bound — a single fused bucket is the right shape
 for xGMI, not many small ones).
 
 Step pipeline per rank (fused GPU path):
-    hg_train_step (gather+fwd+loss+bwd, per-block slabs)
-    hg_grad_reduce (slabs -> flat bucket)
+    hg_train_step (gather+fwd+loss+bwd, per-block bf16 slabs)
+    hg_flat_reduce (slabs -> flat bucket + ping-pong counter bump)
     all_reduce(bucket, AVG) over RCCL        <- the only comm
-    hg_adam_flat + hg_bump
+    hg_adam_flat (reads the bumped counter slot)
+Single GPU: hg_train_step + hg_reduce_adam (the whole tail in one kernel);
+both variants replay from per-parity hipGraphs plus 2-/4-step pair/quad
+graphs.
 
 CPU path (gloo, used by the world_size=2 CPU tests here): eager torch autograd
 with the same single-bucket all-reduce semantics.
@@ -128,10 +131,10 @@ class HourglassStepTrainer:
         self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.bfloat16, device=device)
         self.gflat = torch.zeros(p.P, dtype=torch.float32, device=device)
         self.loss_acc = torch.zeros(1, dtype=torch.float32, device=device)
-        # counters [batch idx, adam t, epoch]. DP (world>1): a single tensor,
-        # bumped by hg_slab_reduce. Single GPU: PING-PONG pair for the fused
-        # hg_reduce_adam — step q reads slot q&1 and writes the bumped values
-        # to the other slot (a fused bump would race its own readers).
+        # counters [batch idx, adam t, epoch]: PING-PONG pair — step q reads
+        # slot q&1 and writes the bumped values to the other slot (bumping
+        # inside hg_reduce_adam/hg_flat_reduce would race their own readers
+        # on a single tensor). self.counters IS slot 0 (checkpoint name).
         self.counters = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
         self.counters_b = torch.tensor([0, 0, 0], dtype=torch.int32, device=device)
         self._cnt = [self.counters, self.counters_b]
