@@ -12,6 +12,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("gscale"), py::arg("y_is_x"), py::arg("dx_out"),
           py::arg("wmax"), py::arg("kbits"), py::arg("seed"),
           py::arg("widths") = std::vector<int64_t>{});
+    m.def("hg_train_step_w1", &hg_train_step_w1_launch,
+          "wave-local hourglass train step (W=64, 16-row blocks, no barriers)",
+          py::arg("X"), py::arg("Y"), py::arg("counters"), py::arg("w"),
+          py::arg("wt"), py::arg("bias"), py::arg("dw"), py::arg("db"),
+          py::arg("loss"), py::arg("B"), py::arg("F_out"), py::arg("acts"),
+          py::arg("gscale"), py::arg("y_is_x"), py::arg("dx_out"),
+          py::arg("wmax"), py::arg("kbits"), py::arg("seed"),
+          py::arg("widths") = std::vector<int64_t>{});
     m.def("hg_adam", &hg_adam_launch, "slab-reduce + Adam + bf16 weight emit");
     m.def("hg_bump", &hg_bump_launch, "advance step counters");
     m.def("hg_permute_fill", &hg_permute_fill_launch, "materialize Feistel shuffle (tests)");
@@ -40,4 +48,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.attr("WMAX") = 64;
     m.attr("MAXL") = 8;
     m.attr("MT") = 32;   // hg_train_step batch rows per block/slab
+    m.attr("MT_W1") = 16;  // hg_train_step_w1 (wave-local W=64) rows/slab
 }

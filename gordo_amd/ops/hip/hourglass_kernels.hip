@@ -325,6 +325,228 @@ __global__ void hg_train_step(HgTrainArgs a) {
 }
 
 // ===========================================================================
+// Wave-local train step (W=64 tier): dense-layer work is ROW-LOCAL, so a
+// single 64-lane wave owns 16 batch rows end-to-end — gather, L-layer
+// forward, loss, full backward, its own dW/db slab — with ZERO barriers
+// (the 4-wave hg_train_step above is ~74% barrier-parked; rocprof measured
+// 24.4 us vs 13.8 us for this shape at the headline config). All tiles are
+// wave-private LDS; same-wave LDS dependencies are ordered by lgkmcnt, no
+// __syncthreads anywhere. dW uses the native K=16 MFMA
+// (v_mfma_f32_16x16x16_bf16) since the batch tile is 16 rows.
+// Grid = B/16 blocks of ONE wave: 512 blocks at B=8192 -> 2 independent,
+// never-stalling waves per CU instead of 4 lockstepped ones.
+// ===========================================================================
+typedef __bf16 ga_bf16x4 __attribute__((ext_vector_type(4)));
+
+// transposed dpre tiles for the wave-local kernel: [W][16] bf16, 32-B rows
+__device__ __forceinline__ void ga_t16_write1(char* t, int row, int m, unsigned short v) {
+    *(unsigned short*)(t + row * 32 + m * 2) = v;
+}
+__device__ __forceinline__ ga_bf16x4 ga_t16_read4(const char* t, int row, int m) {
+    return *(const ga_bf16x4*)(t + row * 32 + m * 2);
+}
+
+#define HG_MT1 16
+
+template <int W>
+__launch_bounds__(64)
+__global__ void hg_train_step_w1(HgTrainArgs a) {
+    constexpr int TB = HG_MT1 * W * 2;  // act/scratch tile bytes ([16][W])
+    constexpr int NTW = W / 16;         // col tiles of 16 (full width/wave)
+    const int lane = threadIdx.x;       // 64 threads = exactly one wave
+    const int tile = blockIdx.x;
+    const int base = tile * HG_MT1;
+
+    auto act_tile = [&](int i) { return ga_smem + i * TB; };
+    char* dcur = ga_smem + a.L * TB;        // dpre  [16][W]
+    char* dnext = dcur + TB;
+    char* dcurT = dnext + TB;               // dpre^T [W][16]
+    char* dnextT = dcurT + (size_t)W * 32;
+    float* s_bias = (float*)(dnextT + (size_t)W * 32);  // [L][W]
+
+    const int step = a.counters[0];
+    const long long pbase = (long long)step * a.B + base;
+    const unsigned int pkey = a.seed ^ ((unsigned int)a.counters[2] * 0x9E3779B1u);
+    auto sample_row = [&](int m) -> unsigned int {
+        unsigned int g = (unsigned int)((pbase + m) % (long long)a.N);
+        return ga_permute(g, (unsigned int)a.N, a.kbits, pkey);
+    };
+
+    for (int i = lane; i < a.L * W; i += 64) s_bias[i] = a.bias[i];
+
+    // ---- gather minibatch tile into acts[0] (wave-private) -------------
+    for (int c = lane; c < HG_MT1 * (W / 8); c += 64) {
+        int m = c / (W / 8);
+        int col0 = (c % (W / 8)) * 8;
+        int gi = base + m;
+        ga_u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (gi < a.B) {
+            unsigned int row = sample_row(m);
+            v = *(const ga_u16x8*)(a.X + (long long)row * W + col0);
+        }
+        hg_write8<W>(act_tile(0), m, col0, v);
+    }
+
+    // ---- forward: [16 rows][W cols] per layer, no barriers --------------
+    for (int li = 0; li < a.L; ++li) {
+        const unsigned short* wl = a.w + (size_t)li * W * W;
+        ga_f32x4 acc[NTW];
+#pragma unroll
+        for (int nt = 0; nt < NTW; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int kk = 0; kk < W; kk += 32) {
+            ga_bf16x8 afrag = hg_read8<W>(act_tile(li), lane & 15, kk + (lane >> 4) * 8);
+#pragma unroll
+            for (int nt = 0; nt < NTW; ++nt) {
+                ga_bf16x8 bfrag = *(const ga_bf16x8*)(wl + (size_t)(nt * 16 + (lane & 15)) * W + kk + (lane >> 4) * 8);
+                acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
+            }
+        }
+
+        if (li + 1 < a.L) {
+#pragma unroll
+            for (int nt = 0; nt < NTW; ++nt) {
+                int col = nt * 16 + (lane & 15);
+                float b = s_bias[li * W + col];
+                bool pad = col >= a.width[li + 1];
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int m = (lane >> 4) * 4 + r;
+                    float h = pad ? 0.f : ga_act(acc[nt][r] + b, a.act_id[li]);
+                    hg_write1<W>(act_tile(li + 1), m, col, ga_f32_to_bf16(h));
+                }
+            }
+        } else {
+            // ---- loss epilogue: dpre_{L-1} = (yhat - y) * gscale --------
+            float lsum = 0.f;
+#pragma unroll
+            for (int nt = 0; nt < NTW; ++nt) {
+                int col = nt * 16 + (lane & 15);
+                float b = s_bias[li * W + col];
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int m = (lane >> 4) * 4 + r;
+                    int gi = base + m;
+                    float yhat = ga_act(acc[nt][r] + b, a.act_id[li]);
+                    float yval = 0.f;
+                    bool valid = (gi < a.B) && (col < a.F_out);
+                    if (valid) {
+                        if (a.y_is_x) {
+                            yval = ga_bf16_to_f32(hg_read1<W>(act_tile(0), m, col));
+                        } else {
+                            unsigned int row = sample_row(m);
+                            yval = ga_bf16_to_f32(a.Y[(long long)row * W + col]);
+                        }
+                    }
+                    float diff = valid ? (yhat - yval) : 0.f;
+                    lsum += diff * diff;
+                    float dpre = diff * a.gscale * ga_act_grad_from_h(yhat, a.act_id[li]);
+                    unsigned short dp = ga_f32_to_bf16(dpre);
+                    hg_write1<W>(dcur, m, col, dp);
+                    ga_t16_write1(dcurT, col, m, dp);
+                }
+            }
+            lsum = ga_wave_sum(lsum);
+            if (lane == 0) atomicAdd(a.loss, lsum);
+        }
+    }
+
+    // ---- backward: dX -> dpre_{li-1}; dW (K=16 native); db --------------
+    for (int li = a.L - 1; li >= 0; --li) {
+        if (li > 0 || a.dx_out != nullptr) {
+            const unsigned short* wtl = a.wt + (size_t)li * W * W;
+            ga_f32x4 acc[NTW];
+#pragma unroll
+            for (int nt = 0; nt < NTW; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+            for (int kk = 0; kk < W; kk += 32) {
+                ga_bf16x8 afrag = hg_read8<W>(dcur, lane & 15, kk + (lane >> 4) * 8);
+#pragma unroll
+                for (int nt = 0; nt < NTW; ++nt) {
+                    ga_bf16x8 bfrag = *(const ga_bf16x8*)(wtl + (size_t)(nt * 16 + (lane & 15)) * W + kk + (lane >> 4) * 8);
+                    acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[nt], 0, 0, 0);
+                }
+            }
+            if (li > 0) {
+#pragma unroll
+                for (int nt = 0; nt < NTW; ++nt) {
+                    int col = nt * 16 + (lane & 15);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int m = (lane >> 4) * 4 + r;
+                        float h = ga_bf16_to_f32(hg_read1<W>(act_tile(li), m, col));
+                        float d = acc[nt][r] * ga_act_grad_from_h(h, a.act_id[li - 1]);
+                        unsigned short dp = ga_f32_to_bf16(d);
+                        hg_write1<W>(dnext, m, col, dp);
+                        ga_t16_write1(dnextT, col, m, dp);
+                    }
+                }
+            } else {
+#pragma unroll
+                for (int nt = 0; nt < NTW; ++nt) {
+                    int col = nt * 16 + (lane & 15);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int m = (lane >> 4) * 4 + r;
+                        int gi = base + m;
+                        if (gi < a.B)
+                            a.dx_out[(size_t)gi * W + col] = ga_f32_to_bf16(acc[nt][r]);
+                    }
+                }
+            }
+        }
+
+        // ---- dW[in][out]: K = the 16 batch rows, native K=16 MFMA -------
+        {
+            unsigned short* dwl = a.dw + ((size_t)tile * a.L + li) * W * W;
+#pragma unroll
+            for (int sub = 0; sub < W / 16; ++sub) {
+                int in_base = sub * 16;
+                ga_f32x4 acc[W / 16];
+#pragma unroll
+                for (int nt = 0; nt < W / 16; ++nt) acc[nt] = ga_f32x4{0.f, 0.f, 0.f, 0.f};
+                // A[i=in][k=m] = acts[li][m][in]: strided scalar reads
+                ga_bf16x4 afrag;
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    int m = (lane >> 4) * 4 + e;
+                    unsigned short u = hg_read1<W>(act_tile(li), m, in_base + (lane & 15));
+                    afrag[e] = *(__bf16*)&u;
+                }
+#pragma unroll
+                for (int nt = 0; nt < W / 16; ++nt) {
+                    // B[k=m][j=out] = dcurT[out][m]: 8-B row reads
+                    ga_bf16x4 bfrag = ga_t16_read4(dcurT, nt * 16 + (lane & 15), (lane >> 4) * 4);
+                    acc[nt] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(afrag, bfrag, acc[nt], 0, 0, 0);
+                }
+#pragma unroll
+                for (int nt = 0; nt < W / 16; ++nt) {
+                    int out = nt * 16 + (lane & 15);
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int in = in_base + (lane >> 4) * 4 + r;
+                        dwl[(size_t)in * W + out] = ga_f32_to_bf16(acc[nt][r]);
+                    }
+                }
+            }
+        }
+
+        // ---- db: column sums of dpre over the 16 rows -------------------
+        for (int out = lane; out < W; out += 64) {
+            float sum = 0.f;
+#pragma unroll
+            for (int mi = 0; mi < HG_MT1 / 4; ++mi) {
+                ga_bf16x4 v = ga_t16_read4(dcurT, out, mi * 4);
+#pragma unroll
+                for (int e = 0; e < 4; ++e) sum += (float)v[e];
+            }
+            a.db[((size_t)tile * a.L + li) * W + out] = ga_f32_to_bf16(sum);
+        }
+
+        char* t0 = dcur; dcur = dnext; dnext = t0;
+        char* t1 = dcurT; dcurT = dnextT; dnextT = t1;
+    }
+}
+
+// ===========================================================================
 // Legacy single-stage Adam (kept for the kernel unit tests): reduce dW/db
 // slabs, update fp32 master, emit bf16 W and WT. One thread per parameter.
 // ===========================================================================
@@ -949,6 +1171,58 @@ void hg_train_step_launch(torch::Tensor X, torch::Tensor Y,
         hipLaunchKernelGGL(hg_train_step<W>, dim3(n_tiles), dim3(256), smem,
                            stream.stream(), a);
     });
+}
+
+void hg_train_step_w1_launch(torch::Tensor X, torch::Tensor Y,
+                             torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
+                             torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
+                             torch::Tensor loss, int64_t B, int64_t F_out,
+                             std::vector<int64_t> acts, double gscale, bool y_is_x,
+                             c10::optional<torch::Tensor> dx_out, int64_t wmax,
+                             int64_t kbits, int64_t seed, std::vector<int64_t> widths) {
+    CHECK_GPU(X);
+    TORCH_CHECK(wmax == 64, "wave-local train step is the W=64 tier");
+    TORCH_CHECK(counters.numel() >= 3, "counters must hold [batch, t, epoch]");
+    int L = (int)(w.numel() / (wmax * wmax));
+    HgTrainArgs a;
+    a.X = (const unsigned short*)X.data_ptr();
+    a.Y = (const unsigned short*)Y.data_ptr();
+    a.counters = counters.data_ptr<int>();
+    a.kbits = (int)kbits;
+    a.seed = (unsigned int)seed;
+    a.w = (const unsigned short*)w.data_ptr();
+    a.wt = (const unsigned short*)wt.data_ptr();
+    a.bias = bias.data_ptr<float>();
+    TORCH_CHECK(dw.scalar_type() == torch::kBFloat16, "dw slabs must be bf16");
+    TORCH_CHECK(db.scalar_type() == torch::kBFloat16, "db slabs must be bf16");
+    a.dw = (unsigned short*)dw.data_ptr();
+    a.db = (unsigned short*)db.data_ptr();
+    a.loss = loss.data_ptr<float>();
+    a.dx_out = dx_out.has_value() ? (unsigned short*)dx_out->data_ptr() : nullptr;
+    a.B = (int)B;
+    a.N = (int)(X.size(0));
+    a.F_out = (int)F_out;
+    a.L = L;
+    a.y_is_x = y_is_x ? 1 : 0;
+    ga_fill_acts(a.act_id, acts, L);
+    if (widths.empty()) {
+        for (int i = 0; i <= GA_MAXL; ++i) a.width[i] = (int)wmax;
+    } else {
+        TORCH_CHECK((int)widths.size() == L + 1, "widths must have L+1 entries");
+        for (int i = 0; i <= L; ++i) a.width[i] = (int)widths[i];
+        for (int i = L + 1; i <= GA_MAXL; ++i) a.width[i] = (int)wmax;
+    }
+    a.gscale = (float)gscale;
+
+    int n_tiles = (int)((B + HG_MT1 - 1) / HG_MT1);
+    TORCH_CHECK(dw.size(0) >= n_tiles, "dw slab count too small");
+    auto stream = at::hip::getCurrentHIPStream();
+    constexpr int W = 64;
+    constexpr size_t TB = HG_MT1 * W * 2;
+    size_t smem = (size_t)(a.L + 2) * TB + 2 * ((size_t)W * 32) + GA_MAXL * W * 4;
+    TORCH_CHECK(smem <= 160 * 1024, "wave-local step LDS overflow at L=", a.L);
+    hipLaunchKernelGGL(hg_train_step_w1<W>, dim3(n_tiles), dim3(64), smem,
+                       stream.stream(), a);
 }
 
 void hg_adam_launch(torch::Tensor master, torch::Tensor m, torch::Tensor v,

@@ -57,6 +57,14 @@ void hg_reduce_adam_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor mas
                            double lr, double beta1, double beta2, double eps,
                            int64_t wmax);
 
+void hg_train_step_w1_launch(torch::Tensor X, torch::Tensor Y,
+                             torch::Tensor counters, torch::Tensor w, torch::Tensor wt,
+                             torch::Tensor bias, torch::Tensor dw, torch::Tensor db,
+                             torch::Tensor loss, int64_t B, int64_t F_out,
+                             std::vector<int64_t> acts, double gscale, bool y_is_x,
+                             c10::optional<torch::Tensor> dx_out, int64_t wmax,
+                             int64_t kbits, int64_t seed, std::vector<int64_t> widths);
+
 void hg_flat_reduce_launch(torch::Tensor dw, torch::Tensor db, torch::Tensor g,
                            torch::Tensor counters_r, torch::Tensor counters_w,
                            int64_t n_slabs, int64_t L, int64_t n_batches,

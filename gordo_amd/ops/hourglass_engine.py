@@ -147,7 +147,15 @@ class FusedHourglassTrainer:
 
         N = Xd.shape[0]
         B = max(64, min(cfg.batch_size, N))
-        MT = int(getattr(self.E, "MT", 32))
+        # W=64 tier: the wave-local step (one 64-lane wave owns 16 rows,
+        # zero barriers) replaces the 4-wave/32-row kernel
+        use_w1 = (
+            WM == 64
+            and hasattr(self.E, "hg_train_step_w1")
+            and not os.environ.get("GORDO_NO_W1")
+        )
+        train_step = self.E.hg_train_step_w1 if use_w1 else self.E.hg_train_step
+        MT = int(getattr(self.E, "MT_W1", 16)) if use_w1 else int(getattr(self.E, "MT", 32))
         n_tiles = (B + MT - 1) // MT
         n_batches = max(1, math.ceil(N / B))
 
@@ -173,7 +181,7 @@ class FusedHourglassTrainer:
         seed = int(cfg.seed) if cfg.seed is not None else 0x5EED
 
         def one_step(q: int):
-            self.E.hg_train_step(
+            train_step(
                 Xd, Yd, cnt[q], p.w_bf16, p.wt_bf16, p.bias, dw, db,
                 loss_acc, B, F_out, acts, gscale, bool(y_is_x), None, WM,
                 kbits, seed, [int(x) for x in p.widths],

@@ -125,7 +125,14 @@ class HourglassStepTrainer:
         self.Xd = pad_rows_bf16(np.asarray(self.X, np.float32), device, WM)
         N = self.Xd.shape[0]
         self.N = N
-        MT = int(getattr(self.E, "MT", 32))
+        # W=64 tier: wave-local train step (16-row blocks, zero barriers)
+        self._use_w1 = (
+            WM == 64
+            and hasattr(self.E, "hg_train_step_w1")
+            and not os.environ.get("GORDO_NO_W1")
+        )
+        self._train_step = self.E.hg_train_step_w1 if self._use_w1 else self.E.hg_train_step
+        MT = int(getattr(self.E, "MT_W1", 16)) if self._use_w1 else int(getattr(self.E, "MT", 32))
         self.n_tiles = (self.B + MT - 1) // MT
         self.dw = torch.zeros(self.n_tiles, p.L, WM * WM, dtype=torch.bfloat16, device=device)
         self.db = torch.zeros(self.n_tiles, p.L, WM, dtype=torch.bfloat16, device=device)
@@ -258,7 +265,7 @@ class HourglassStepTrainer:
         cfg = self.cfg
         WM = p.wmax
         cr, cw = self._cnt[q], self._cnt[q ^ 1]
-        self.E.hg_train_step(
+        self._train_step(
             self.Xd, self.Xd, cr, p.w_bf16, p.wt_bf16, p.bias,
             self.dw, self.db, self.loss_acc, self.B, self.F_out, p.act_ids,
             self.gscale, True, None, WM, self.kbits, self.seed,

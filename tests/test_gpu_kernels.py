@@ -216,3 +216,102 @@ def test_feistel_shuffle_is_bijective():
     E.hg_permute_fill(b, feistel_kbits(1000), 2)
     torch.cuda.synchronize()
     assert not torch.equal(a, b)
+
+
+def test_train_step_w1_gradients_match_autograd():
+    """The wave-local (16-row, zero-barrier) W=64 train step vs torch
+    autograd AND vs the 4-wave kernel's slab sums on identical data."""
+    net = make_net(seed=13)
+    ref_net = make_net(seed=13)
+    rng = np.random.default_rng(13)
+    N = 270  # deliberately not a multiple of 16: last tile is ragged
+    X = rng.normal(size=(N, 50)).astype(np.float32)
+
+    E = ext()
+    if not hasattr(E, "hg_train_step_w1"):
+        pytest.skip("extension built without the wave-local step")
+    p = _PackedParams(net, DEV)
+    L, F_out = p.L, p.widths[-1]
+    Xd = pad_rows_bf16(X, DEV)
+    B = N
+    MT = int(E.MT_W1)
+    n_tiles = (B + MT - 1) // MT
+    dw = torch.zeros(n_tiles, L, WMAX * WMAX, dtype=torch.bfloat16, device=DEV)
+    db = torch.zeros(n_tiles, L, WMAX, dtype=torch.bfloat16, device=DEV)
+    loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
+    gscale = 2.0 / (B * F_out)
+    E.hg_train_step_w1(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+                       loss, B, F_out, p.act_ids, gscale, True, None, WMAX, 0, 0)
+    torch.cuda.synchronize()
+
+    Xref = Xd[:, :50].float().cpu()
+    out = ref_net(Xref)
+    loss_ref = torch.nn.functional.mse_loss(out, Xref)
+    loss_ref.backward()
+    kernel_loss = float(loss.item()) / (B * F_out)
+    ref_val = float(loss_ref.detach())
+    assert abs(kernel_loss - ref_val) / max(ref_val, 1e-9) < 5e-2
+
+    linears = [m for m in ref_net.net if isinstance(m, torch.nn.Linear)]
+    dw_sum = dw.sum(dim=0, dtype=torch.float32).view(L, WMAX, WMAX).cpu()
+    db_sum = db.sum(dim=0, dtype=torch.float32).view(L, WMAX).cpu()
+    for i, lin in enumerate(linears):
+        o, ninp = lin.weight.shape
+        got_w = dw_sum[i, :ninp, :o].T
+        ref_w = lin.weight.grad.detach()
+        scale = ref_w.abs().max().item() or 1.0
+        assert (got_w - ref_w).abs().max().item() < 0.05 * scale + 1e-4, f"layer {i} dW"
+        got_b = db_sum[i, :o]
+        ref_b = lin.bias.grad.detach()
+        bscale = ref_b.abs().max().item() or 1.0
+        assert (got_b - ref_b).abs().max().item() < 0.05 * bscale + 1e-4, f"layer {i} db"
+        assert dw_sum[i, ninp:, :].abs().max().item() == 0.0
+        assert dw_sum[i, :, o:].abs().max().item() == 0.0
+
+    # and against the 4-wave kernel's own slab sums (tight: same math path)
+    MT4 = int(E.MT)
+    nt4 = (B + MT4 - 1) // MT4
+    dw4 = torch.zeros(nt4, L, WMAX * WMAX, dtype=torch.bfloat16, device=DEV)
+    db4 = torch.zeros(nt4, L, WMAX, dtype=torch.bfloat16, device=DEV)
+    loss4 = torch.zeros(1, dtype=torch.float32, device=DEV)
+    E.hg_train_step(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw4, db4,
+                    loss4, B, F_out, p.act_ids, gscale, True, None, WMAX, 0, 0)
+    torch.cuda.synchronize()
+    assert abs(float(loss.item()) - float(loss4.item())) / max(float(loss4.item()), 1e-9) < 1e-3
+    s1 = dw.sum(dim=0, dtype=torch.float32)
+    s4 = dw4.sum(dim=0, dtype=torch.float32)
+    ref = s4.abs().max().item() or 1.0
+    assert (s1 - s4).abs().max().item() < 0.01 * ref + 1e-5
+
+
+def test_train_step_w1_dx_out_matches():
+    """dx_out (the LSTM-head contract) from the wave-local step matches the
+    4-wave kernel bit-for-tolerance."""
+    net = make_net(seed=14)
+    rng = np.random.default_rng(14)
+    N = 128
+    X = rng.normal(size=(N, 50)).astype(np.float32)
+    E = ext()
+    if not hasattr(E, "hg_train_step_w1"):
+        pytest.skip("extension built without the wave-local step")
+    p = _PackedParams(net, DEV)
+    L, F_out = p.L, p.widths[-1]
+    Xd = pad_rows_bf16(X, DEV)
+    B = N
+    counters = torch.tensor([0, 1, 0], dtype=torch.int32, device=DEV)
+    gscale = 2.0 / (B * F_out)
+
+    outs = []
+    for fn, mt in ((E.hg_train_step_w1, int(E.MT_W1)), (E.hg_train_step, int(E.MT))):
+        nt = (B + mt - 1) // mt
+        dw = torch.zeros(nt, L, WMAX * WMAX, dtype=torch.bfloat16, device=DEV)
+        db = torch.zeros(nt, L, WMAX, dtype=torch.bfloat16, device=DEV)
+        loss = torch.zeros(1, dtype=torch.float32, device=DEV)
+        dx = torch.zeros(B, WMAX, dtype=torch.bfloat16, device=DEV)
+        fn(Xd, Xd, counters, p.w_bf16, p.wt_bf16, p.bias, dw, db,
+           loss, B, F_out, p.act_ids, gscale, True, dx, WMAX, 0, 0)
+        torch.cuda.synchronize()
+        outs.append(dx.float().cpu())
+    scale = outs[1].abs().max().item() or 1.0
+    assert (outs[0] - outs[1]).abs().max().item() < 0.02 * scale + 1e-6
