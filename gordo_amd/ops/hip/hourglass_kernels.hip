@@ -325,16 +325,18 @@ __global__ void hg_train_step(HgTrainArgs a) {
 }
 
 // ===========================================================================
-// Wave-local train step (W=64 tier): dense-layer work is ROW-LOCAL, so a
-// single 64-lane wave owns 16 batch rows end-to-end — gather, L-layer
-// forward, loss, full backward, its own dW/db slab — with ZERO barriers
-// (the 4-wave hg_train_step above is ~74% barrier-parked; rocprof measured
-// 24.4 us vs 13.8 us for this shape at the headline config). All tiles are
-// wave-private LDS; same-wave LDS dependencies are ordered by lgkmcnt, no
-// __syncthreads anywhere. dW uses the native K=16 MFMA
+// Wave-local train step (W=64 tier, OPT-IN via GORDO_W1=1): dense-layer
+// work is ROW-LOCAL, so a single 64-lane wave owns 16 batch rows end-to-end
+// — gather, L-layer forward, loss, full backward, its own dW/db slab — with
+// ZERO barriers. All tiles are wave-private LDS (same-wave dependencies
+// ordered by lgkmcnt); dW uses the native K=16 MFMA
 // (v_mfma_f32_16x16x16_bf16) since the batch tile is 16 rows.
-// Grid = B/16 blocks of ONE wave: 512 blocks at B=8192 -> 2 independent,
-// never-stalling waves per CU instead of 4 lockstepped ones.
+// MEASURED SLOWER than the 4-wave kernel at the headline config (184.7M vs
+// 263.5M samples/s): B=8192 gives only 512 single-wave blocks, so half the
+// SIMDs idle and the solo wave eats its own latencies — the 4-wave kernel's
+// barrier parking is the batch's parallelism ceiling, not waste. Kept
+// correct + tested (grad parity, dx_out, cross-kernel slab agreement) for
+// larger-batch experiments; see docs/ROADMAP.md.
 // ===========================================================================
 typedef __bf16 ga_bf16x4 __attribute__((ext_vector_type(4)));
 

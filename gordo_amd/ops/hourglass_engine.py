@@ -147,12 +147,17 @@ class FusedHourglassTrainer:
 
         N = Xd.shape[0]
         B = max(64, min(cfg.batch_size, N))
-        # W=64 tier: the wave-local step (one 64-lane wave owns 16 rows,
-        # zero barriers) replaces the 4-wave/32-row kernel
+        # GORDO_W1=1 opts into the wave-local W=64 step (one 64-lane wave
+        # owns 16 rows, zero barriers). Measured SLOWER at the headline
+        # config (184.7M vs 263.5M samples/s): B=8192 yields only 512
+        # single-wave blocks, so half the SIMDs idle and the solo wave eats
+        # its LDS/global latencies — the 4-wave kernel's barrier parking is
+        # the parallelism ceiling, not waste (docs/ROADMAP.md). Kept
+        # correct + tested for larger-batch experiments.
         use_w1 = (
             WM == 64
             and hasattr(self.E, "hg_train_step_w1")
-            and not os.environ.get("GORDO_NO_W1")
+            and os.environ.get("GORDO_W1") == "1"
         )
         train_step = self.E.hg_train_step_w1 if use_w1 else self.E.hg_train_step
         MT = int(getattr(self.E, "MT_W1", 16)) if use_w1 else int(getattr(self.E, "MT", 32))

@@ -125,11 +125,12 @@ class HourglassStepTrainer:
         self.Xd = pad_rows_bf16(np.asarray(self.X, np.float32), device, WM)
         N = self.Xd.shape[0]
         self.N = N
-        # W=64 tier: wave-local train step (16-row blocks, zero barriers)
+        # GORDO_W1=1 opts into the wave-local W=64 step — measured slower
+        # at the headline batch (see hourglass_engine.py / docs/ROADMAP.md)
         self._use_w1 = (
             WM == 64
             and hasattr(self.E, "hg_train_step_w1")
-            and not os.environ.get("GORDO_NO_W1")
+            and os.environ.get("GORDO_W1") == "1"
         )
         self._train_step = self.E.hg_train_step_w1 if self._use_w1 else self.E.hg_train_step
         MT = int(getattr(self.E, "MT_W1", 16)) if self._use_w1 else int(getattr(self.E, "MT", 32))
