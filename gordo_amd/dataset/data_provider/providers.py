@@ -15,6 +15,7 @@ ncs_reader.py, iroc_reader.py):
 from __future__ import annotations
 
 import os
+import zlib
 from datetime import datetime
 from typing import Iterable, List, Optional
 
@@ -47,7 +48,9 @@ class RandomDataProvider(GordoBaseDataProvider):
             raise ValueError(f"from_ts {from_ts} must be before to_ts {to_ts}")
         index = pd.date_range(start=from_ts, end=to_ts, freq=self.resolution)
         for tag in tag_list:
-            rng = np.random.default_rng(abs(hash(tag.name)) % (2**32))
+            # zlib.crc32 is stable across processes; hash() is salted per
+            # interpreter and would make "deterministic" data run-dependent.
+            rng = np.random.default_rng(zlib.crc32(tag.name.encode()))
             n = min(len(index), int(rng.integers(self.min_size, self.max_size + 1)))
             # walkish signal so scaling / anomaly scoring has structure
             vals = np.cumsum(rng.standard_normal(n) * 0.1) + rng.normal(0, 1)
