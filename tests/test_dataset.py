@@ -1,5 +1,7 @@
 """Dataset & provider tests (parity: tests/.../test_dataset.py)."""
+import os
 from datetime import datetime, timezone
+from pathlib import Path
 
 import numpy as np
 import pandas as pd
@@ -23,6 +25,8 @@ from gordo_amd.machine.sensor_tag import (
     normalize_sensor_tag,
     normalize_sensor_tags,
 )
+
+_REPO_ROOT = Path(__file__).resolve().parent.parent
 
 TAGS = [f"tag-{i}" for i in range(5)]
 START, END = "2020-01-01T00:00:00Z", "2020-02-01T00:00:00Z"
@@ -55,6 +59,29 @@ def test_dataset_from_config_roundtrip():
     X1, _ = ds.get_data()
     X2, _ = ds2.get_data()
     pd.testing.assert_frame_equal(X1, X2)
+
+
+def test_random_provider_deterministic_across_processes():
+    """Seeds must not depend on Python's per-process hash salt: a fresh
+    interpreter (different PYTHONHASHSEED) must produce identical data."""
+    import subprocess
+    import sys
+
+    script = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "from gordo_amd.dataset import RandomDataset\n"
+        "X, _ = RandomDataset(%r, %r, %r).get_data()\n"
+        "print(X.to_numpy().sum(), X.shape)\n" % (str(_REPO_ROOT), START, END, TAGS)
+    )
+    outs = [
+        subprocess.run(
+            [sys.executable, "-c", script],
+            capture_output=True, text=True, check=True,
+            env={**os.environ, "PYTHONHASHSEED": seed},
+        ).stdout
+        for seed in ("1", "2")
+    ]
+    assert outs[0] == outs[1] and outs[0].strip()
 
 
 def test_invalid_date_order_raises():
