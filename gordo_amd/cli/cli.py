@@ -164,7 +164,8 @@ def workflow():
 
 @workflow.command("generate")
 @click.option("--machine-config", required=True, type=click.Path(exists=True))
-@click.option("--project-name", default="project")
+@click.option("--project-name", default=None,
+              help="Overrides the YAML's top-level project_name (default: 'project')")
 @click.option("--output-dir", default="./models")
 @click.option("--model-register-dir", default=None)
 @click.option("--n-gpus", default=8, type=int)
@@ -178,7 +179,7 @@ def workflow_generate(machine_config, project_name, output_dir, model_register_d
         config = NormalizedConfig(yaml.safe_load(f), project_name=project_name)
     text = generate_yaml(
         config,
-        project_name=project_name,
+        project_name=config.project_name,
         output_dir=output_dir,
         registry_dir=model_register_dir,
         n_gpus=n_gpus,
@@ -210,7 +211,8 @@ def fleet():
 
 @fleet.command("build")
 @click.option("--machine-config", required=True, type=click.Path(exists=True))
-@click.option("--project-name", default="project")
+@click.option("--project-name", default=None,
+              help="Overrides the YAML's top-level project_name (default: 'project')")
 @click.option("--output-dir", required=True)
 @click.option("--model-register-dir", default=None)
 @click.option("--workers", default=None, type=int)

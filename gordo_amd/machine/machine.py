@@ -154,7 +154,10 @@ class NormalizedConfig:
     ``globals:`` defaults merged under each entry of ``machines:``.
     """
 
-    def __init__(self, config: Dict[str, Any], project_name: str = "project"):
+    def __init__(self, config: Dict[str, Any], project_name: Optional[str] = None):
+        # Explicit argument wins; otherwise a top-level ``project_name:`` key
+        # in the YAML itself; otherwise the reference's "project" default.
+        project_name = project_name or config.get("project_name") or "project"
         self.project_name = project_name
         self.globals: Dict[str, Any] = config.get("globals") or {}
         machines = config.get("machines")
@@ -170,5 +173,5 @@ class NormalizedConfig:
         ]
 
     @classmethod
-    def from_yaml(cls, text: str, project_name: str = "project") -> "NormalizedConfig":
+    def from_yaml(cls, text: str, project_name: Optional[str] = None) -> "NormalizedConfig":
         return cls(yaml.safe_load(text), project_name=project_name)

@@ -37,6 +37,14 @@ machines:
 """
 
 
+def test_project_name_resolution():
+    """YAML top-level project_name is honored; explicit argument wins."""
+    yml = "project_name: from-yaml\n" + PROJECT_YAML
+    assert NormalizedConfig.from_yaml(yml).project_name == "from-yaml"
+    assert NormalizedConfig.from_yaml(yml, project_name="cli").project_name == "cli"
+    assert NormalizedConfig.from_yaml(PROJECT_YAML).project_name == "project"
+
+
 def test_normalized_config_merging():
     cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="proj")
     assert len(cfg.machines) == 2
