@@ -111,15 +111,26 @@ def client():
 @click.option("--target", default=None)
 @click.option("--output-dir", default=None, help="write parquet per machine")
 @click.option("--parallelism", default=10, type=int)
-def client_predict(start, end, watchman_url, endpoints, target, output_dir, parallelism):
+@click.option("--data-provider", default=None,
+              help='YAML/JSON provider config, e.g. \'{"type": "RandomDataProvider"}\' '
+                   "(parity: the reference CLI's --data-provider)")
+def client_predict(start, end, watchman_url, endpoints, target, output_dir,
+                   parallelism, data_provider):
     from ..client import Client, ForwardPredictionsToDisk
+    from ..dataset.data_provider import GordoBaseDataProvider
 
     forwarder = ForwardPredictionsToDisk(output_dir) if output_dir else None
+    provider = (
+        GordoBaseDataProvider.from_dict(yaml.safe_load(data_provider))
+        if data_provider
+        else None
+    )
     c = Client(
         watchman_url=watchman_url,
         endpoints=list(endpoints) or None,
         target=target,
         parallelism=parallelism,
+        data_provider=provider,
         prediction_forwarder=forwarder,
     )
     results = c.predict(start, end)

@@ -143,6 +143,38 @@ def test_version_and_help_surfaces():
         assert r.exit_code == 0, (args, r.output)
 
 
+def test_client_predict_data_provider_option(monkeypatch):
+    """--data-provider YAML/JSON reaches Client as a constructed provider
+    (reference CLI parity: predictions can be pulled without library code)."""
+    import gordo_amd.client as client_mod
+    from gordo_amd.dataset.data_provider import RandomDataProvider
+
+    captured = {}
+
+    class FakeClient:
+        def __init__(self, **kwargs):
+            captured.update(kwargs)
+
+        def predict(self, start, end):
+            return []
+
+    monkeypatch.setattr(client_mod, "Client", FakeClient)
+    runner = CliRunner()
+    result = runner.invoke(
+        gordo,
+        [
+            "client", "predict",
+            "--endpoint", "http://127.0.0.1:1",
+            "--data-provider", '{"type": "RandomDataProvider", "min_size": 5, "max_size": 7}',
+            "2020-01-01T00:00:00Z", "2020-01-02T00:00:00Z",
+        ],
+    )
+    assert result.exit_code == 0, result.output
+    provider = captured["data_provider"]
+    assert isinstance(provider, RandomDataProvider)
+    assert (provider.min_size, provider.max_size) == (5, 7)
+
+
 def test_build_command_bad_config_fails_with_message(tmp_path):
     """Builder-pod error reporting (reference: the build entrypoint surfaces
     config errors with a non-zero exit instead of a silent crash)."""
