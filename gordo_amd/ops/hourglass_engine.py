@@ -227,14 +227,14 @@ class FusedHourglassTrainer:
                     with torch.cuda.graph(g):
                         one_step(q)
                     graphs.append(g)
-                # even-parity PAIR/QUAD graphs: 2/4 steps per replay — cut
-                # the per-step Python/replay overhead in the epoch loop.
+                # even-parity PAIR/QUAD/OCT graphs: 2/4/8 steps per replay —
+                # cut the per-step Python/replay overhead in the epoch loop.
                 # Only worth the extra captures on jobs with enough steps:
                 # two captures + two warmup steps cost ~10 ms per fit()
                 # while saving ~2.5 us/step — breakeven ~4k steps. Fleet
                 # builds (~300 steps/model) measured 10.1 -> 13.0 s per 256
                 # models with these captured unconditionally.
-                pair_graph = quad_graph = None
+                pair_graph = quad_graph = oct_graph = None
                 if cfg.epochs * n_batches >= 4096:
                     pair_graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(pair_graph):
@@ -243,6 +243,10 @@ class FusedHourglassTrainer:
                     quad_graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(quad_graph):
                         for q in (0, 1, 0, 1):
+                            one_step(q)
+                    oct_graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(oct_graph):
+                        for q in (0, 1, 0, 1, 0, 1, 0, 1):
                             one_step(q)
                 # capture replays with pristine state below
                 reset_counters()
@@ -254,6 +258,7 @@ class FusedHourglassTrainer:
                 graphs = None
                 pair_graph = None
                 quad_graph = None
+                oct_graph = None
                 # warm-up may have stepped the real state before the failure:
                 # restore so the raw-launch run still starts pristine
                 torch.cuda.synchronize()
@@ -272,7 +277,10 @@ class FusedHourglassTrainer:
             r = n_batches
             if graphs is not None:
                 while r:
-                    if quad_graph is not None and parity == 0 and r >= 4:
+                    if oct_graph is not None and parity == 0 and r >= 8:
+                        oct_graph.replay()  # parity 0 -> 0
+                        r -= 8
+                    elif quad_graph is not None and parity == 0 and r >= 4:
                         quad_graph.replay()  # parity 0 -> 0
                         r -= 4
                     elif pair_graph is not None and parity == 0 and r >= 2:

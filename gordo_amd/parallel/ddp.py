@@ -189,9 +189,9 @@ class HourglassStepTrainer:
                 with torch.cuda.graph(g):
                     self._fused_step_raw(q)
                 graphs.append(g)
-            # even-parity PAIR/QUAD graphs: 2/4 whole steps per replay cut
-            # the per-step Python/launch overhead (visible once the kernels
-            # shrank to ~30 us/step)
+            # even-parity PAIR/QUAD/OCT graphs: 2/4/8 whole steps per replay
+            # cut the per-step Python/launch overhead (visible once the
+            # kernels shrank to ~30 us/step)
             gp = torch.cuda.CUDAGraph()
             with torch.cuda.graph(gp):
                 self._fused_step_raw(0)
@@ -202,6 +202,11 @@ class HourglassStepTrainer:
                 for q in (0, 1, 0, 1):
                     self._fused_step_raw(q)
             self._quad_graph = gq
+            go = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(go):
+                for q in (0, 1, 0, 1, 0, 1, 0, 1):
+                    self._fused_step_raw(q)
+            self._oct_graph = go
             # restore pre-warmup state
             self.p.master.copy_(state[0]); self.m.copy_(state[1]); self.v.copy_(state[2])
             self.counters.copy_(state[3])
@@ -220,6 +225,7 @@ class HourglassStepTrainer:
             self._graphs = None
             self._pair_graph = None
             self._quad_graph = None
+            self._oct_graph = None
             # warm-up may have stepped the real state before the failure:
             # restore so the raw-launch run still starts pristine
             torch.cuda.synchronize()
@@ -241,8 +247,9 @@ class HourglassStepTrainer:
         self._parity ^= 1
 
     def steps(self, n: int):
-        """Run n whole training steps, using the 2-step pair graph where
-        parity allows (same work as n step() calls, fewer replays)."""
+        """Run n whole training steps, using the widest multi-step graph
+        (8/4/2 steps) parity allows — same work as n step() calls, fewer
+        replays."""
         if not self.fused or self._graphs is None:
             for _ in range(n):
                 self.step()
@@ -250,8 +257,12 @@ class HourglassStepTrainer:
         i = 0
         pair = getattr(self, "_pair_graph", None)
         quad = getattr(self, "_quad_graph", None)
+        oct_ = getattr(self, "_oct_graph", None)
         while i < n:
-            if quad is not None and self._parity == 0 and n - i >= 4:
+            if oct_ is not None and self._parity == 0 and n - i >= 8:
+                oct_.replay()  # parity 0 -> 0
+                i += 8
+            elif quad is not None and self._parity == 0 and n - i >= 4:
                 quad.replay()  # parity 0 -> 0
                 i += 4
             elif pair is not None and self._parity == 0 and n - i >= 2:
