@@ -248,7 +248,9 @@ def fleet_build(machine_config, project_name, output_dir, model_register_dir,
 
 @fleet.command("serve")
 @click.option("--models-dir", required=True, type=click.Path(exists=True))
-@click.option("--project-name", default="project")
+@click.option("--project-name", default=None,
+              help="Defaults to the project name recorded by `fleet build` "
+                   "in <models-dir>/project.json, else 'project'")
 @click.option("--host", default="0.0.0.0")
 @click.option("--port", default=5557, type=int)
 def fleet_serve(models_dir, project_name, host, port):

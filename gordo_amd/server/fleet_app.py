@@ -17,10 +17,24 @@ from .app import build_app
 logger = logging.getLogger(__name__)
 
 
-def build_fleet_app(models_dir: str, project_name: str = "project",
+def _recorded_project_name(models_dir: str) -> Optional[str]:
+    """Project name the fleet builder recorded alongside the artifacts."""
+    import json
+
+    path = os.path.join(models_dir, "project.json")
+    try:
+        with open(path) as f:
+            return json.load(f).get("project-name")
+    except (OSError, ValueError):
+        return None
+
+
+def build_fleet_app(models_dir: str, project_name: Optional[str] = None,
                     machines: Optional[List[str]] = None):
     from fastapi import FastAPI
 
+    if project_name is None:
+        project_name = _recorded_project_name(models_dir) or "project"
     if machines is None:
         machines = sorted(
             d
@@ -111,7 +125,7 @@ def build_fleet_app(models_dir: str, project_name: str = "project",
     return app
 
 
-def run_fleet_server(models_dir: str, project_name: str = "project",
+def run_fleet_server(models_dir: str, project_name: Optional[str] = None,
                      host: str = "0.0.0.0", port: int = 5557):
     import uvicorn
 

@@ -81,6 +81,19 @@ class FleetBuilder:
 
     def build_all(self) -> List[Dict[str, Any]]:
         os.makedirs(self.output_dir, exist_ok=True)
+        # record the project identity with the artifacts so `fleet serve`
+        # mounts the same /gordo/v0/<project>/ prefix the plan/clients expect
+        import json
+
+        with open(os.path.join(self.output_dir, "project.json"), "w") as f:
+            json.dump(
+                {
+                    "project-name": self.config.project_name,
+                    "machines": [m.name for m in self.config.machines],
+                },
+                f,
+                indent=1,
+            )
         ctx = mp.get_context("spawn")
         results: List[Dict[str, Any]] = []
         t0 = time.time()

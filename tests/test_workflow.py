@@ -106,6 +106,33 @@ def test_fleet_builder_builds_all(tmp_path):
     assert all(r["status"] == "succeeded" for r in results2)
 
 
+def test_fleet_serve_uses_recorded_project_name(tmp_path):
+    """fleet build records project.json; build_fleet_app mounts under that
+    project's prefix when no explicit name is given (build/serve coherence)."""
+    import json
+
+    cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="recorded-proj")
+    for m in cfg.machines:
+        m.evaluation["cv_mode"] = "full_build"
+    models = str(tmp_path / "models")
+    results = FleetBuilder(cfg, models, n_workers=1, n_gpus=0).build_all()
+    assert all(r["status"] == "succeeded" for r in results), results
+    with open(tmp_path / "models" / "project.json") as f:
+        assert json.load(f)["project-name"] == "recorded-proj"
+
+    from fastapi.testclient import TestClient
+
+    from gordo_amd.server.fleet_app import build_fleet_app
+
+    app = build_fleet_app(models)  # no explicit project_name
+    with TestClient(app) as c:
+        r = c.get("/")
+        assert r.json()["project-name"] == "recorded-proj"
+        ep = r.json()["endpoints"][0]["endpoint"]
+        assert ep.startswith("/gordo/v0/recorded-proj/")
+        assert c.get(ep + "/healthcheck").status_code == 200
+
+
 def test_fleet_app(tmp_path):
     """Multi-model server mounts each built machine under its project path."""
     cfg = NormalizedConfig.from_yaml(PROJECT_YAML, project_name="proj")
