@@ -227,27 +227,22 @@ class FusedHourglassTrainer:
                     with torch.cuda.graph(g):
                         one_step(q)
                     graphs.append(g)
-                # even-parity PAIR/QUAD/OCT graphs: 2/4/8 steps per replay —
+                # even-parity multi-step graphs: 2/4/8/16 steps per replay —
                 # cut the per-step Python/replay overhead in the epoch loop.
                 # Only worth the extra captures on jobs with enough steps:
-                # two captures + two warmup steps cost ~10 ms per fit()
-                # while saving ~2.5 us/step — breakeven ~4k steps. Fleet
-                # builds (~300 steps/model) measured 10.1 -> 13.0 s per 256
-                # models with these captured unconditionally.
-                pair_graph = quad_graph = oct_graph = None
+                # each capture costs ~5 ms per fit() while the tiers save
+                # ~3.5 us/step — breakeven ~4k steps. Fleet builds (~300
+                # steps/model) measured 10.1 -> 13.0 s per 256 models with
+                # these captured unconditionally.
+                multi_graphs = []
                 if cfg.epochs * n_batches >= 4096:
-                    pair_graph = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(pair_graph):
-                        one_step(0)
-                        one_step(1)
-                    quad_graph = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(quad_graph):
-                        for q in (0, 1, 0, 1):
-                            one_step(q)
-                    oct_graph = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(oct_graph):
-                        for q in (0, 1, 0, 1, 0, 1, 0, 1):
-                            one_step(q)
+                    for size in (2, 4, 8, 16):
+                        g = torch.cuda.CUDAGraph()
+                        with torch.cuda.graph(g):
+                            for i in range(size):
+                                one_step(i & 1)
+                        multi_graphs.append((size, g))
+                    multi_graphs.reverse()  # widest first for greedy replay
                 # capture replays with pristine state below
                 reset_counters()
                 loss_acc.zero_(); m.zero_(); v.zero_()
@@ -256,9 +251,7 @@ class FusedHourglassTrainer:
             except Exception as e:
                 logger.warning("hipGraph capture failed (%s); plain launches", e)
                 graphs = None
-                pair_graph = None
-                quad_graph = None
-                oct_graph = None
+                multi_graphs = []
                 # warm-up may have stepped the real state before the failure:
                 # restore so the raw-launch run still starts pristine
                 torch.cuda.synchronize()
@@ -277,18 +270,19 @@ class FusedHourglassTrainer:
             r = n_batches
             if graphs is not None:
                 while r:
-                    if oct_graph is not None and parity == 0 and r >= 8:
-                        oct_graph.replay()  # parity 0 -> 0
-                        r -= 8
-                    elif quad_graph is not None and parity == 0 and r >= 4:
-                        quad_graph.replay()  # parity 0 -> 0
-                        r -= 4
-                    elif pair_graph is not None and parity == 0 and r >= 2:
-                        pair_graph.replay()  # parity 0 -> 0
-                        r -= 2
+                    if parity == 0:
+                        for size, g in multi_graphs:  # widest first; 0 -> 0
+                            if r >= size:
+                                g.replay()
+                                r -= size
+                                break
+                        else:
+                            graphs[0].replay()
+                            parity = 1
+                            r -= 1
                     else:
-                        graphs[parity].replay()
-                        parity ^= 1
+                        graphs[1].replay()
+                        parity = 0
                         r -= 1
             else:
                 while r:
