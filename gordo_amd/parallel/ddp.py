@@ -12,8 +12,8 @@ Step pipeline per rank (fused GPU path):
     all_reduce(bucket, AVG) over RCCL        <- the only comm
     hg_adam_flat (reads the bumped counter slot)
 Single GPU: hg_train_step + hg_reduce_adam (the whole tail in one kernel);
-both variants replay from per-parity hipGraphs plus 2-/4-step pair/quad
-graphs.
+both variants replay from per-parity hipGraphs plus 2/4/8/16-step
+multi-step graphs (widest tier parity allows per replay).
 
 CPU path (gloo, used by the world_size=2 CPU tests here): eager torch autograd
 with the same single-bucket all-reduce semantics.
