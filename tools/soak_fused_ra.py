@@ -1,5 +1,5 @@
-"""Soak the fused reduce+Adam step path: 100K steps through the quad/pair
-graphs, then verify (a) the ping-pong counters hold exactly N steps of
+"""Soak the fused reduce+Adam step path: 100K steps through the multi-step
+replay tiers, then verify (a) the ping-pong counters hold exactly N steps of
 state, (b) parameters are finite, (c) the loss over the last epoch-window
 is below the first (still learning / not diverged).
 
@@ -29,7 +29,7 @@ def main(n_steps: int = 100_000):
     module = feedforward_hourglass(50, 50, encoding_layers=3, compression_factor=0.5)
     cfg = TrainConfig(batch_size=8192, lr=1e-3, dtype="bf16")
     tr = HourglassStepTrainer(module, X, 8192, cfg, dp=DPContext(device=torch.device("cuda:0")))
-    assert tr.fused and tr._graphs is not None and tr._quad_graph is not None
+    assert tr.fused and tr._graphs is not None and tr._multi_graphs
 
     tr.steps(64)
     tr.synchronize()
