@@ -64,7 +64,8 @@ def project_config(n_machines: int, n_tags: int):
 
 def measure_http(out_dir: str, args, X):
     """End-to-end HTTP: uvicorn fleet server in a subprocess; measures both
-    wire formats (parquet for reference parity, arrow IPC = the fast path)."""
+    wire formats (parquet for reference parity, arrow IPC, raw float32 = the
+    fast paths)."""
     import socket
     import subprocess
     import httpx
@@ -97,26 +98,31 @@ def measure_http(out_dir: str, args, X):
                 time.sleep(0.5)
             out = {}
             abody = dataframe_into_arrow_bytes(X)
-            for fmt, payload, ctype in (
-                ("parquet", body, "application/octet-stream"),
-                ("arrow", abody, "application/vnd.apache.arrow.stream"),
+            rarr = np.ascontiguousarray(X.to_numpy(np.float32))
+            rheaders = {"content-type": "application/x-gordo-raw",
+                        "X-Gordo-Shape": f"{rarr.shape[0]},{rarr.shape[1]}"}
+            for fmt, payload, headers in (
+                ("parquet", body, {"content-type": "application/octet-stream"}),
+                ("arrow", abody,
+                 {"content-type": "application/vnd.apache.arrow.stream"}),
+                ("raw", rarr.tobytes(), rheaders),
             ):
                 urls = [
                     f"{base}/gordo/v0/bench-fleet/machine-{i:03d}/anomaly/prediction?format={fmt}"
                     for i in range(min(8, args.machines))
                 ]
                 for u in urls:  # warm
-                    r = client.post(u, content=payload, headers={"content-type": ctype})
+                    r = client.post(u, content=payload, headers=headers)
                     assert r.status_code == 200, r.text[:200]
                 lat = []
                 for i in range(min(args.requests, 200)):
                     t = time.perf_counter()
                     r = client.post(urls[i % len(urls)], content=payload,
-                                    headers={"content-type": ctype})
+                                    headers=headers)
                     assert r.status_code == 200
                     lat.append(time.perf_counter() - t)
                 lat_ms = np.array(lat) * 1e3
-                key = "http_" + ("" if fmt == "parquet" else "arrow_")
+                key = "http_" + ("" if fmt == "parquet" else fmt + "_")
                 out[key + "p50_ms"] = float(np.percentile(lat_ms, 50))
                 out[key + "p99_ms"] = float(np.percentile(lat_ms, 99))
         return out

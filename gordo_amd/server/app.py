@@ -11,7 +11,7 @@ Parity: gordo_components/server/server.py + views — one server per model dir:
 MI355X-native: FastAPI/uvicorn (async) instead of Flask/gunicorn; the model is
 loaded once; inference goes through ModelRuntime's fused HIP path on GPU.
 Request/response formats match the reference's capability: JSON records,
-nested dict, or parquet (?format=parquet or parquet content-type).
+nested dict, parquet, arrow IPC, or raw float32 (?format=raw + X-Gordo-Shape header).
 """
 import logging
 
@@ -46,7 +46,16 @@ def build_app(model_dir: str, device=None):
         ctype = request.headers.get("content-type", "")
         body = await request.body()
         try:
-            if "arrow" in ctype:
+            if "x-gordo-raw" in ctype:
+                # zero-copy fast path: body is row-major float32 LE, shape
+                # in the X-Gordo-Shape header ("N,F"); columns are
+                # positional by construction (training tag order)
+                shape = request.headers.get("x-gordo-shape", "")
+                n, f = (int(v) for v in shape.split(","))
+                arr = np.frombuffer(body, dtype="<f4").reshape(n, f)
+                X = pd.DataFrame(arr)
+                y = None
+            elif "arrow" in ctype:
                 X = dataframe_from_arrow_bytes(body)
                 y = None
             elif "octet-stream" in ctype or "parquet" in ctype:
@@ -124,6 +133,22 @@ def build_app(model_dir: str, device=None):
     def _respond(df: pd.DataFrame, request: Request) -> Response:
         fmt = request.query_params.get("format", "json")
         accept = request.headers.get("accept", "")
+        if fmt == "raw" or "x-gordo-raw" in accept:
+            # row-major float32 LE values; shape in X-Gordo-Shape, column
+            # layout on request (?columns=1 — it is fixed per model, so
+            # clients normally fetch it once)
+            headers = {"X-Gordo-Shape": f"{df.shape[0]},{df.shape[1]}"}
+            if request.query_params.get("columns"):
+                import json as _json
+
+                headers["X-Gordo-Columns"] = _json.dumps(
+                    [list(c) if isinstance(c, tuple) else [c, ""] for c in df.columns]
+                )
+            return Response(
+                content=np.ascontiguousarray(df.to_numpy(np.float32)).tobytes(),
+                media_type="application/x-gordo-raw",
+                headers=headers,
+            )
         if fmt == "arrow" or "arrow" in accept:
             return Response(
                 content=dataframe_into_arrow_bytes(df),

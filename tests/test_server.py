@@ -188,6 +188,45 @@ def test_anomaly_prediction_arrow(client):
     assert len(df) == len(X)
 
 
+def test_anomaly_prediction_raw_format(client):
+    """Raw float32 wire format: positional body + X-Gordo-Shape in, values +
+    shape (+ column layout on ?columns=1) out; agrees with the JSON path."""
+    import json
+
+    X = _x_frame()
+    arr = np.ascontiguousarray(X.to_numpy(np.float32))
+    r = client.post(
+        "/anomaly/prediction?format=raw&columns=1",
+        content=arr.tobytes(),
+        headers={
+            "Content-Type": "application/x-gordo-raw",
+            "X-Gordo-Shape": f"{arr.shape[0]},{arr.shape[1]}",
+        },
+    )
+    assert r.status_code == 200, r.text
+    n, c = (int(v) for v in r.headers["X-Gordo-Shape"].split(","))
+    assert n == len(X)
+    out = np.frombuffer(r.content, dtype="<f4").reshape(n, c)
+    cols = [tuple(v) for v in json.loads(r.headers["X-Gordo-Columns"])]
+    assert len(cols) == c
+    total_i = [i for i, cc in enumerate(cols) if cc[0] == "total-anomaly-score"]
+    assert len(total_i) == 1
+
+    rj = client.post("/anomaly/prediction", json={"X": arr.tolist()})
+    assert rj.status_code == 200
+    total_json = rj.json()["data"]["total-anomaly-score"]
+    np.testing.assert_allclose(out[:, total_i[0]], total_json, rtol=1e-5)
+
+    # bad shape header -> 400, not a 500
+    r = client.post(
+        "/anomaly/prediction",
+        content=arr.tobytes(),
+        headers={"Content-Type": "application/x-gordo-raw",
+                 "X-Gordo-Shape": "nonsense"},
+    )
+    assert r.status_code == 400
+
+
 def test_lstm_model_serving_offsets(tmp_path):
     """An LSTM artifact serves /prediction with lookback-aligned rows."""
     import numpy as np
