@@ -136,7 +136,9 @@ def build_app(model_dir: str, device=None):
         if fmt == "raw" or "x-gordo-raw" in accept:
             # row-major float32 LE values; shape in X-Gordo-Shape, column
             # layout on request (?columns=1 — it is fixed per model, so
-            # clients normally fetch it once)
+            # clients normally fetch it once). Values-only: the caller keeps
+            # its own time index; lookback models emit fewer rows than sent
+            # (rows align to the TAIL of the input, like every other format)
             headers = {"X-Gordo-Shape": f"{df.shape[0]},{df.shape[1]}"}
             if request.query_params.get("columns"):
                 import json as _json
